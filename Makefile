@@ -1,0 +1,47 @@
+# gloo_amd build: host C++ with g++, HIP device code with hipcc
+# (gfx950 only). Produces the in-tree python extension gloo_amd/_C*.so
+# and (later) the benchmark binary.
+
+PYTHON ?= python3
+HIPCC ?= /opt/rocm/bin/hipcc
+CXX ?= g++
+ROCM ?= /opt/rocm
+
+EXT_SUFFIX := $(shell $(PYTHON) -c "import sysconfig;print(sysconfig.get_config_var('EXT_SUFFIX'))")
+PY_INC := $(shell $(PYTHON) -c "import sysconfig;print(sysconfig.get_paths()['include'])")
+PYBIND_INC := $(shell $(PYTHON) -c "import pybind11;print(pybind11.get_include())")
+
+TARGET := gloo_amd/_C$(EXT_SUFFIX)
+BENCH := bin/gloo_amd_bench
+
+CXXFLAGS := -O3 -g -std=c++17 -fPIC -Wall -Wextra -Wno-unused-parameter \
+  -pthread -Icsrc -I$(ROCM)/include -D__HIP_PLATFORM_AMD__=1
+HIPCCFLAGS := -O3 -std=c++17 -fPIC --offload-arch=gfx950 -Icsrc
+
+CC_SRCS := $(shell find csrc -name '*.cc' ! -path 'csrc/bindings/*' ! -path 'csrc/bench/*')
+HIP_SRCS := $(shell find csrc -name '*.hip' 2>/dev/null)
+BIND_SRCS := $(shell find csrc/bindings -name '*.cc')
+
+CC_OBJS := $(patsubst csrc/%.cc,build/%.o,$(CC_SRCS))
+HIP_OBJS := $(patsubst csrc/%.hip,build/%.hip.o,$(HIP_SRCS))
+BIND_OBJS := $(patsubst csrc/%.cc,build/%.o,$(BIND_SRCS))
+
+LDFLAGS := -L$(ROCM)/lib -lamdhip64 -pthread
+
+all: $(TARGET)
+
+build/%.o: csrc/%.cc
+	@mkdir -p $(dir $@)
+	$(CXX) $(CXXFLAGS) -I$(PY_INC) -I$(PYBIND_INC) -c $< -o $@
+
+build/%.hip.o: csrc/%.hip
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(HIPCCFLAGS) -c $< -o $@
+
+$(TARGET): $(CC_OBJS) $(HIP_OBJS) $(BIND_OBJS)
+	$(CXX) -shared -o $@ $^ $(LDFLAGS)
+
+clean:
+	rm -rf build $(TARGET) $(BENCH)
+
+.PHONY: all clean

@@ -1,0 +1,571 @@
+// pybind11 bindings for gloo_amd.
+//
+// Exposes devices, stores, contexts, the v2 collectives, unbound buffers
+// (tagged send/recv incl. recv-from-any) and bound buffers, so the Python
+// test-suite and the torch.distributed ProcessGroup wrapper can drive the
+// C++ core. Buffers are passed as raw addresses (tensor.data_ptr()); all
+// blocking entry points release the GIL.
+#include <pybind11/chrono.h>
+#include <pybind11/functional.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "collectives/collectives.h"
+#include "collectives/reduce_fns.h"
+#include "common/store.h"
+#include "context.h"
+#include "rendezvous/stores.h"
+#include "transport/tcp/context.h"
+#include "transport/tcp/device.h"
+
+namespace py = pybind11;
+using namespace glooamd;
+
+namespace {
+
+std::chrono::milliseconds ms(long v) {
+  return std::chrono::milliseconds(v);
+}
+
+AllreduceOptions::Algorithm algoFromString(const std::string& s) {
+  if (s == "ring") {
+    return AllreduceOptions::Algorithm::RING;
+  }
+  if (s == "bcube") {
+    return AllreduceOptions::Algorithm::BCUBE;
+  }
+  GA_THROW("unknown allreduce algorithm: ", s);
+}
+
+} // namespace
+
+PYBIND11_MODULE(_C, m) {
+  m.doc() = "gloo_amd: MI355X-native collective communications";
+
+  // --- exceptions -----------------------------------------------------------
+  static py::exception<EnforceNotMet> excEnforce(m, "EnforceNotMet");
+  static py::exception<TimeoutException> excTimeout(m, "TimeoutError");
+  static py::exception<IoException> excIo(m, "IoError");
+  static py::exception<Exception> excBase(m, "GlooAmdError");
+  py::register_exception_translator([](std::exception_ptr p) {
+    try {
+      if (p) {
+        std::rethrow_exception(p);
+      }
+    } catch (const TimeoutException& e) {
+      py::set_error(excTimeout, e.what());
+    } catch (const IoException& e) {
+      py::set_error(excIo, e.what());
+    } catch (const EnforceNotMet& e) {
+      py::set_error(excEnforce, e.what());
+    } catch (const Exception& e) {
+      py::set_error(excBase, e.what());
+    }
+  });
+
+  // --- enums ----------------------------------------------------------------
+  py::enum_<DType>(m, "DType")
+      .value("f32", DType::F32)
+      .value("f64", DType::F64)
+      .value("f16", DType::F16)
+      .value("bf16", DType::BF16)
+      .value("i8", DType::I8)
+      .value("u8", DType::U8)
+      .value("i32", DType::I32)
+      .value("i64", DType::I64)
+      .value("u64", DType::U64);
+
+  py::enum_<ReduceOp>(m, "ReduceOp")
+      .value("sum", ReduceOp::SUM)
+      .value("product", ReduceOp::PRODUCT)
+      .value("min", ReduceOp::MIN)
+      .value("max", ReduceOp::MAX);
+
+  m.def("dtype_size", &dtypeSize);
+
+  // --- stores ---------------------------------------------------------------
+  py::class_<IStore, std::shared_ptr<IStore>>(m, "Store")
+      .def(
+          "set",
+          [](IStore& s, const std::string& key, py::bytes data) {
+            std::string d = data;
+            s.set(key, std::vector<char>(d.begin(), d.end()));
+          })
+      .def(
+          "get",
+          [](IStore& s, const std::string& key) {
+            std::vector<char> v;
+            {
+              py::gil_scoped_release rel;
+              v = s.get(key);
+            }
+            return py::bytes(v.data(), v.size());
+          })
+      .def(
+          "wait",
+          [](IStore& s, const std::vector<std::string>& keys, long timeoutMs) {
+            py::gil_scoped_release rel;
+            s.wait(keys, ms(timeoutMs));
+          },
+          py::arg("keys"),
+          py::arg("timeout_ms") = 30000);
+
+  py::class_<HashStore, IStore, std::shared_ptr<HashStore>>(m, "HashStore")
+      .def(py::init<>());
+  py::class_<FileStore, IStore, std::shared_ptr<FileStore>>(m, "FileStore")
+      .def(py::init<const std::string&>());
+  py::class_<PrefixStore, IStore, std::shared_ptr<PrefixStore>>(
+      m, "PrefixStore")
+      .def(py::init<const std::string&, std::shared_ptr<IStore>>());
+  py::class_<TcpStore, IStore, std::shared_ptr<TcpStore>>(m, "TcpStore")
+      .def(
+          py::init([](const std::string& host, int port, bool isServer,
+                      long timeoutMs) {
+            py::gil_scoped_release rel;
+            return std::make_shared<TcpStore>(host, port, isServer,
+                                              ms(timeoutMs));
+          }),
+          py::arg("host"),
+          py::arg("port"),
+          py::arg("is_server"),
+          py::arg("timeout_ms") = 60000);
+
+  // --- device ---------------------------------------------------------------
+  py::class_<transport::Device, std::shared_ptr<transport::Device>>(
+      m, "Device")
+      .def("__str__", &transport::Device::str);
+
+  m.def(
+      "create_tcp_device",
+      [](const std::string& hostname) {
+        tcp::TcpAttr attr;
+        attr.hostname = hostname;
+        return std::static_pointer_cast<transport::Device>(
+            tcp::createTcpDevice(attr));
+      },
+      py::arg("hostname") = std::string());
+
+  // --- unbound + bound buffers ---------------------------------------------
+  py::class_<transport::UnboundBuffer>(m, "UnboundBuffer")
+      .def(
+          "send",
+          [](transport::UnboundBuffer& b, int dst, uint64_t slot, size_t off,
+             long nbytes) {
+            py::gil_scoped_release rel;
+            b.send(dst, slot, off,
+                   nbytes < 0 ? transport::kUnspecified : size_t(nbytes));
+          },
+          py::arg("dst"),
+          py::arg("slot"),
+          py::arg("offset") = 0,
+          py::arg("nbytes") = -1)
+      .def(
+          "recv",
+          [](transport::UnboundBuffer& b, int src, uint64_t slot, size_t off,
+             long nbytes) {
+            py::gil_scoped_release rel;
+            b.recv(src, slot, off,
+                   nbytes < 0 ? transport::kUnspecified : size_t(nbytes));
+          },
+          py::arg("src"),
+          py::arg("slot"),
+          py::arg("offset") = 0,
+          py::arg("nbytes") = -1)
+      .def(
+          "recv_any",
+          [](transport::UnboundBuffer& b, const std::vector<int>& srcs,
+             uint64_t slot, size_t off, long nbytes) {
+            py::gil_scoped_release rel;
+            b.recv(srcs, slot, off,
+                   nbytes < 0 ? transport::kUnspecified : size_t(nbytes));
+          },
+          py::arg("srcs"),
+          py::arg("slot"),
+          py::arg("offset") = 0,
+          py::arg("nbytes") = -1)
+      .def(
+          "wait_recv",
+          [](transport::UnboundBuffer& b, long timeoutMs) {
+            int src = -1;
+            bool ok;
+            {
+              py::gil_scoped_release rel;
+              ok = b.waitRecv(&src, ms(timeoutMs));
+            }
+            return py::make_tuple(ok, src);
+          },
+          py::arg("timeout_ms") = -1)
+      .def(
+          "wait_send",
+          [](transport::UnboundBuffer& b, long timeoutMs) {
+            py::gil_scoped_release rel;
+            return b.waitSend(ms(timeoutMs));
+          },
+          py::arg("timeout_ms") = -1)
+      .def("abort_wait_recv", &transport::UnboundBuffer::abortWaitRecv)
+      .def("abort_wait_send", &transport::UnboundBuffer::abortWaitSend);
+
+  py::class_<transport::Buffer>(m, "Buffer")
+      .def(
+          "send",
+          [](transport::Buffer& b, size_t off, long length, size_t roff) {
+            py::gil_scoped_release rel;
+            b.send(off, length < 0 ? b.size() - off : size_t(length), roff);
+          },
+          py::arg("offset") = 0,
+          py::arg("length") = -1,
+          py::arg("roffset") = 0)
+      .def("wait_recv",
+           [](transport::Buffer& b) {
+             py::gil_scoped_release rel;
+             b.waitRecv();
+           })
+      .def("wait_send", [](transport::Buffer& b) {
+        py::gil_scoped_release rel;
+        b.waitSend();
+      });
+
+  py::class_<transport::Pair>(m, "Pair")
+      .def(
+          "create_send_buffer",
+          [](transport::Pair& p, uint64_t slot, uintptr_t ptr, size_t size) {
+            return p.createSendBuffer(slot, reinterpret_cast<void*>(ptr),
+                                      size);
+          },
+          py::keep_alive<0, 1>())
+      .def(
+          "create_recv_buffer",
+          [](transport::Pair& p, uint64_t slot, uintptr_t ptr, size_t size) {
+            return p.createRecvBuffer(slot, reinterpret_cast<void*>(ptr),
+                                      size);
+          },
+          py::keep_alive<0, 1>())
+      .def("is_connected", &transport::Pair::isConnected)
+      .def("__str__", &transport::Pair::str);
+
+  // --- context --------------------------------------------------------------
+  py::class_<Context, std::shared_ptr<Context>>(m, "Context")
+      .def(py::init<int, int, int>(), py::arg("rank"), py::arg("size"),
+           py::arg("base") = 2)
+      .def_readonly("rank", &Context::rank)
+      .def_readonly("size", &Context::size)
+      .def_readwrite("base", &Context::base)
+      .def(
+          "connect_full_mesh",
+          [](Context& ctx, std::shared_ptr<IStore> store,
+             std::shared_ptr<transport::Device> dev) {
+            py::gil_scoped_release rel;
+            ctx.connectFullMesh(*store, dev);
+          },
+          py::arg("store"),
+          py::arg("device"))
+      .def("set_timeout",
+           [](Context& ctx, long timeoutMs) { ctx.setTimeout(ms(timeoutMs)); })
+      .def("close", &Context::closeConnections,
+           py::call_guard<py::gil_scoped_release>())
+      .def("next_slot", &Context::nextSlot, py::arg("num_to_skip") = 1)
+      .def(
+          "get_pair",
+          [](Context& ctx, int rank) { return ctx.getPair(rank); },
+          py::return_value_policy::reference_internal)
+      .def(
+          "create_unbound_buffer",
+          [](Context& ctx, uintptr_t ptr, size_t size) {
+            return ctx.createUnboundBuffer(reinterpret_cast<void*>(ptr),
+                                           size);
+          },
+          py::keep_alive<0, 1>());
+
+  // --- collectives ----------------------------------------------------------
+  m.def(
+      "allreduce",
+      [](std::shared_ptr<Context> ctx, std::vector<uintptr_t> outputs,
+         size_t elements, DType dtype, ReduceOp op,
+         std::vector<uintptr_t> inputs, uint32_t tag,
+         const std::string& algorithm, size_t maxSegmentSize,
+         long timeoutMs) {
+        AllreduceOptions opts(ctx);
+        for (auto p : outputs) {
+          opts.outputs.push_back(reinterpret_cast<void*>(p));
+        }
+        for (auto p : inputs) {
+          opts.inputs.push_back(reinterpret_cast<void*>(p));
+        }
+        opts.elements = elements;
+        opts.elementSize = dtypeSize(dtype);
+        opts.reduce = cpuReduceFn(dtype, op);
+        opts.tag = tag;
+        opts.algorithm = algoFromString(algorithm);
+        opts.maxSegmentSize = maxSegmentSize;
+        opts.timeout = ms(timeoutMs);
+        py::gil_scoped_release rel;
+        allreduce(opts);
+      },
+      py::arg("context"),
+      py::arg("outputs"),
+      py::arg("elements"),
+      py::arg("dtype") = DType::F32,
+      py::arg("op") = ReduceOp::SUM,
+      py::arg("inputs") = std::vector<uintptr_t>(),
+      py::arg("tag") = 0,
+      py::arg("algorithm") = "ring",
+      py::arg("max_segment_size") = kDefaultMaxSegmentSize,
+      py::arg("timeout_ms") = 0);
+
+  m.def(
+      "allgather",
+      [](std::shared_ptr<Context> ctx, uintptr_t output, uintptr_t input,
+         size_t inElements, DType dtype, uint32_t tag, long timeoutMs) {
+        AllgatherOptions opts(ctx);
+        opts.output = reinterpret_cast<void*>(output);
+        opts.input = reinterpret_cast<void*>(input);
+        opts.inElements = inElements;
+        opts.elementSize = dtypeSize(dtype);
+        opts.tag = tag;
+        opts.timeout = ms(timeoutMs);
+        py::gil_scoped_release rel;
+        allgather(opts);
+      },
+      py::arg("context"),
+      py::arg("output"),
+      py::arg("input"),
+      py::arg("in_elements"),
+      py::arg("dtype") = DType::F32,
+      py::arg("tag") = 0,
+      py::arg("timeout_ms") = 0);
+
+  m.def(
+      "allgatherv",
+      [](std::shared_ptr<Context> ctx, uintptr_t output, uintptr_t input,
+         std::vector<size_t> counts, DType dtype, uint32_t tag,
+         long timeoutMs) {
+        AllgathervOptions opts(ctx);
+        opts.output = reinterpret_cast<void*>(output);
+        opts.input = reinterpret_cast<void*>(input);
+        opts.counts = std::move(counts);
+        opts.elementSize = dtypeSize(dtype);
+        opts.tag = tag;
+        opts.timeout = ms(timeoutMs);
+        py::gil_scoped_release rel;
+        allgatherv(opts);
+      },
+      py::arg("context"),
+      py::arg("output"),
+      py::arg("input"),
+      py::arg("counts"),
+      py::arg("dtype") = DType::F32,
+      py::arg("tag") = 0,
+      py::arg("timeout_ms") = 0);
+
+  m.def(
+      "alltoall",
+      [](std::shared_ptr<Context> ctx, uintptr_t output, uintptr_t input,
+         size_t perRankElements, DType dtype, uint32_t tag, long timeoutMs) {
+        AlltoallOptions opts(ctx);
+        opts.output = reinterpret_cast<void*>(output);
+        opts.input = reinterpret_cast<void*>(input);
+        opts.perRankElements = perRankElements;
+        opts.elementSize = dtypeSize(dtype);
+        opts.tag = tag;
+        opts.timeout = ms(timeoutMs);
+        py::gil_scoped_release rel;
+        alltoall(opts);
+      },
+      py::arg("context"),
+      py::arg("output"),
+      py::arg("input"),
+      py::arg("per_rank_elements"),
+      py::arg("dtype") = DType::F32,
+      py::arg("tag") = 0,
+      py::arg("timeout_ms") = 0);
+
+  m.def(
+      "alltoallv",
+      [](std::shared_ptr<Context> ctx, uintptr_t output, uintptr_t input,
+         std::vector<size_t> inCounts, std::vector<size_t> outCounts,
+         DType dtype, uint32_t tag, long timeoutMs) {
+        AlltoallvOptions opts(ctx);
+        opts.output = reinterpret_cast<void*>(output);
+        opts.input = reinterpret_cast<void*>(input);
+        opts.inCounts = std::move(inCounts);
+        opts.outCounts = std::move(outCounts);
+        opts.elementSize = dtypeSize(dtype);
+        opts.tag = tag;
+        opts.timeout = ms(timeoutMs);
+        py::gil_scoped_release rel;
+        alltoallv(opts);
+      },
+      py::arg("context"),
+      py::arg("output"),
+      py::arg("input"),
+      py::arg("in_counts"),
+      py::arg("out_counts"),
+      py::arg("dtype") = DType::F32,
+      py::arg("tag") = 0,
+      py::arg("timeout_ms") = 0);
+
+  m.def(
+      "barrier",
+      [](std::shared_ptr<Context> ctx, uint32_t tag, long timeoutMs) {
+        BarrierOptions opts(ctx);
+        opts.tag = tag;
+        opts.timeout = ms(timeoutMs);
+        py::gil_scoped_release rel;
+        barrier(opts);
+      },
+      py::arg("context"),
+      py::arg("tag") = 0,
+      py::arg("timeout_ms") = 0);
+
+  m.def(
+      "broadcast",
+      [](std::shared_ptr<Context> ctx, uintptr_t output, uintptr_t input,
+         size_t elements, DType dtype, int root, uint32_t tag,
+         long timeoutMs) {
+        BroadcastOptions opts(ctx);
+        opts.output = reinterpret_cast<void*>(output);
+        opts.input = reinterpret_cast<void*>(input);
+        opts.elements = elements;
+        opts.elementSize = dtypeSize(dtype);
+        opts.root = root;
+        opts.tag = tag;
+        opts.timeout = ms(timeoutMs);
+        py::gil_scoped_release rel;
+        broadcast(opts);
+      },
+      py::arg("context"),
+      py::arg("output"),
+      py::arg("input"),
+      py::arg("elements"),
+      py::arg("dtype") = DType::F32,
+      py::arg("root") = 0,
+      py::arg("tag") = 0,
+      py::arg("timeout_ms") = 0);
+
+  m.def(
+      "gather",
+      [](std::shared_ptr<Context> ctx, uintptr_t output, uintptr_t input,
+         size_t inElements, DType dtype, int root, uint32_t tag,
+         long timeoutMs) {
+        GatherOptions opts(ctx);
+        opts.output = reinterpret_cast<void*>(output);
+        opts.input = reinterpret_cast<void*>(input);
+        opts.inElements = inElements;
+        opts.elementSize = dtypeSize(dtype);
+        opts.root = root;
+        opts.tag = tag;
+        opts.timeout = ms(timeoutMs);
+        py::gil_scoped_release rel;
+        gather(opts);
+      },
+      py::arg("context"),
+      py::arg("output"),
+      py::arg("input"),
+      py::arg("in_elements"),
+      py::arg("dtype") = DType::F32,
+      py::arg("root") = 0,
+      py::arg("tag") = 0,
+      py::arg("timeout_ms") = 0);
+
+  m.def(
+      "gatherv",
+      [](std::shared_ptr<Context> ctx, uintptr_t output, uintptr_t input,
+         std::vector<size_t> counts, DType dtype, int root, uint32_t tag,
+         long timeoutMs) {
+        GathervOptions opts(ctx);
+        opts.output = reinterpret_cast<void*>(output);
+        opts.input = reinterpret_cast<void*>(input);
+        opts.counts = std::move(counts);
+        opts.elementSize = dtypeSize(dtype);
+        opts.root = root;
+        opts.tag = tag;
+        opts.timeout = ms(timeoutMs);
+        py::gil_scoped_release rel;
+        gatherv(opts);
+      },
+      py::arg("context"),
+      py::arg("output"),
+      py::arg("input"),
+      py::arg("counts"),
+      py::arg("dtype") = DType::F32,
+      py::arg("root") = 0,
+      py::arg("tag") = 0,
+      py::arg("timeout_ms") = 0);
+
+  m.def(
+      "reduce",
+      [](std::shared_ptr<Context> ctx, uintptr_t output, uintptr_t input,
+         size_t elements, DType dtype, ReduceOp op, int root, uint32_t tag,
+         long timeoutMs) {
+        ReduceOptions opts(ctx);
+        opts.output = reinterpret_cast<void*>(output);
+        opts.input = reinterpret_cast<void*>(input);
+        opts.elements = elements;
+        opts.elementSize = dtypeSize(dtype);
+        opts.reduce = cpuReduceFn(dtype, op);
+        opts.root = root;
+        opts.tag = tag;
+        opts.timeout = ms(timeoutMs);
+        py::gil_scoped_release rel;
+        reduce(opts);
+      },
+      py::arg("context"),
+      py::arg("output"),
+      py::arg("input"),
+      py::arg("elements"),
+      py::arg("dtype") = DType::F32,
+      py::arg("op") = ReduceOp::SUM,
+      py::arg("root") = 0,
+      py::arg("tag") = 0,
+      py::arg("timeout_ms") = 0);
+
+  m.def(
+      "scatter",
+      [](std::shared_ptr<Context> ctx, uintptr_t output, uintptr_t input,
+         size_t outElements, DType dtype, int root, uint32_t tag,
+         long timeoutMs) {
+        ScatterOptions opts(ctx);
+        opts.output = reinterpret_cast<void*>(output);
+        opts.input = reinterpret_cast<void*>(input);
+        opts.outElements = outElements;
+        opts.elementSize = dtypeSize(dtype);
+        opts.root = root;
+        opts.tag = tag;
+        opts.timeout = ms(timeoutMs);
+        py::gil_scoped_release rel;
+        scatter(opts);
+      },
+      py::arg("context"),
+      py::arg("output"),
+      py::arg("input"),
+      py::arg("out_elements"),
+      py::arg("dtype") = DType::F32,
+      py::arg("root") = 0,
+      py::arg("tag") = 0,
+      py::arg("timeout_ms") = 0);
+
+  m.def(
+      "reduce_scatter",
+      [](std::shared_ptr<Context> ctx, uintptr_t output, uintptr_t input,
+         size_t recvElements, DType dtype, ReduceOp op, uint32_t tag,
+         long timeoutMs) {
+        ReduceScatterOptions opts(ctx);
+        opts.output = reinterpret_cast<void*>(output);
+        opts.input = reinterpret_cast<void*>(input);
+        opts.recvElements = recvElements;
+        opts.elementSize = dtypeSize(dtype);
+        opts.reduce = cpuReduceFn(dtype, op);
+        opts.tag = tag;
+        opts.timeout = ms(timeoutMs);
+        py::gil_scoped_release rel;
+        reduce_scatter(opts);
+      },
+      py::arg("context"),
+      py::arg("output"),
+      py::arg("input"),
+      py::arg("recv_elements"),
+      py::arg("dtype") = DType::F32,
+      py::arg("op") = ReduceOp::SUM,
+      py::arg("tag") = 0,
+      py::arg("timeout_ms") = 0);
+}

@@ -1,0 +1,129 @@
+#include "transport/tcp/loop.h"
+
+#include <sys/epoll.h>
+#include <sys/eventfd.h>
+#include <unistd.h>
+
+#include <array>
+#include <cstring>
+
+#include "common/logging.h"
+
+namespace glooamd {
+namespace tcp {
+
+Loop::Loop() {
+  epfd_ = epoll_create1(EPOLL_CLOEXEC);
+  GA_ENFORCE_GE(epfd_, 0, "epoll_create1: ", strerror(errno));
+  evfd_ = eventfd(0, EFD_NONBLOCK | EFD_CLOEXEC);
+  GA_ENFORCE_GE(evfd_, 0, "eventfd: ", strerror(errno));
+
+  struct epoll_event ev;
+  std::memset(&ev, 0, sizeof(ev));
+  ev.events = EPOLLIN;
+  ev.data.ptr = nullptr; // nullptr marks the wakeup fd
+  GA_ENFORCE_EQ(epoll_ctl(epfd_, EPOLL_CTL_ADD, evfd_, &ev), 0);
+
+  thread_ = std::thread([this] { run(); });
+  threadId_ = thread_.get_id();
+}
+
+Loop::~Loop() {
+  done_ = true;
+  uint64_t one = 1;
+  (void)!write(evfd_, &one, sizeof(one));
+  if (thread_.joinable()) {
+    thread_.join();
+  }
+  close(evfd_);
+  close(epfd_);
+}
+
+void Loop::registerDescriptor(int fd, uint32_t events, Handler* h) {
+  struct epoll_event ev;
+  std::memset(&ev, 0, sizeof(ev));
+  ev.events = events;
+  ev.data.ptr = h;
+  int rv = epoll_ctl(epfd_, EPOLL_CTL_ADD, fd, &ev);
+  GA_ENFORCE_EQ(rv, 0, "epoll_ctl ADD: ", strerror(errno));
+}
+
+void Loop::modifyDescriptor(int fd, uint32_t events, Handler* h) {
+  struct epoll_event ev;
+  std::memset(&ev, 0, sizeof(ev));
+  ev.events = events;
+  ev.data.ptr = h;
+  int rv = epoll_ctl(epfd_, EPOLL_CTL_MOD, fd, &ev);
+  GA_ENFORCE_EQ(rv, 0, "epoll_ctl MOD: ", strerror(errno));
+}
+
+void Loop::unregisterNoWait(int fd) {
+  int rv = epoll_ctl(epfd_, EPOLL_CTL_DEL, fd, nullptr);
+  GA_ENFORCE_EQ(rv, 0, "epoll_ctl DEL: ", strerror(errno));
+}
+
+void Loop::unregisterDescriptor(int fd) {
+  int rv = epoll_ctl(epfd_, EPOLL_CTL_DEL, fd, nullptr);
+  GA_ENFORCE_EQ(rv, 0, "epoll_ctl DEL: ", strerror(errno));
+  // If called off the loop thread, wait until the loop has moved past the
+  // current dispatch batch so no handler for this fd can still be running.
+  if (!inLoopThread()) {
+    waitForTick();
+  }
+}
+
+void Loop::waitForTick() {
+  std::unique_lock<std::mutex> lock(mu_);
+  uint64_t current = tick_;
+  uint64_t one = 1;
+  (void)!write(evfd_, &one, sizeof(one));
+  cv_.wait(lock, [&] { return tick_ > current || done_.load(); });
+}
+
+void Loop::defer(std::function<void()> fn) {
+  {
+    std::lock_guard<std::mutex> lock(mu_);
+    deferred_.push_back(std::move(fn));
+  }
+  uint64_t one = 1;
+  (void)!write(evfd_, &one, sizeof(one));
+}
+
+void Loop::run() {
+  std::array<struct epoll_event, 64> events;
+  while (!done_.load()) {
+    int n = epoll_wait(epfd_, events.data(), events.size(), 100);
+    if (n < 0 && errno != EINTR) {
+      GA_ERROR << "epoll_wait: " << strerror(errno);
+      break;
+    }
+    for (int i = 0; i < n; i++) {
+      auto* h = static_cast<Handler*>(events[i].data.ptr);
+      if (h == nullptr) {
+        uint64_t val;
+        (void)!read(evfd_, &val, sizeof(val));
+        continue;
+      }
+      h->handleEvents(events[i].events);
+    }
+    std::vector<std::function<void()>> fns;
+    {
+      std::lock_guard<std::mutex> lock(mu_);
+      tick_++;
+      fns.swap(deferred_);
+    }
+    cv_.notify_all();
+    for (auto& fn : fns) {
+      fn();
+    }
+  }
+  // final tick so waiters don't hang at shutdown
+  {
+    std::lock_guard<std::mutex> lock(mu_);
+    tick_++;
+  }
+  cv_.notify_all();
+}
+
+} // namespace tcp
+} // namespace glooamd

@@ -1,0 +1,50 @@
+import os
+import sys
+
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def pytest_configure(config):
+    config.addinivalue_line("markers", "gpu: requires an MI355X GPU")
+
+
+@pytest.fixture
+def spawn_threads():
+    """gloo-style multi-rank harness: N python threads, each with its own
+    tcp Device + Context, meeting through a shared in-process HashStore
+    (reference strategy: gloo/test/base_test.h:89-192)."""
+    import threading
+
+    import gloo_amd as ga
+
+    def run(size, fn, base=2, timeout=60.0):
+        store = ga.HashStore()
+        results = [None] * size
+        errors = []
+
+        def worker(rank):
+            try:
+                dev = ga.create_tcp_device()
+                ctx = ga.Context(rank, size, base)
+                ctx.connect_full_mesh(store, dev)
+                results[rank] = fn(ctx, rank, size)
+            except Exception as e:  # noqa: BLE001
+                import traceback
+
+                errors.append((rank, e, traceback.format_exc()))
+
+        threads = [
+            threading.Thread(target=worker, args=(r,), daemon=True)
+            for r in range(size)
+        ]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout)
+            assert not t.is_alive(), f"worker thread hung (>{timeout}s)"
+        assert not errors, f"rank errors: {errors[0][2]}"
+        return results
+
+    return run

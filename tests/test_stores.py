@@ -1,0 +1,72 @@
+"""Rendezvous store tests (reference parity: gloo/rendezvous/*_store)."""
+import threading
+
+import pytest
+
+import gloo_amd as ga
+
+
+def _exercise(store, store2=None):
+    if store2 is None:
+        store2 = store
+    store.set("key1", b"hello")
+    assert store2.get("key1") == b"hello"
+    store.set("key2", b"")
+    store2.wait(["key1", "key2"], timeout_ms=2000)
+
+    def setter():
+        import time
+
+        time.sleep(0.2)
+        store.set("late", b"v" * 10000)
+
+    t = threading.Thread(target=setter)
+    t.start()
+    store2.wait(["late"], timeout_ms=5000)
+    assert store2.get("late") == b"v" * 10000
+    t.join()
+
+
+def test_hash_store():
+    _exercise(ga.HashStore())
+
+
+def test_hash_store_wait_timeout():
+    store = ga.HashStore()
+    with pytest.raises(ga.TimeoutError):
+        store.wait(["missing"], timeout_ms=100)
+
+
+def test_file_store(tmp_path):
+    _exercise(ga.FileStore(str(tmp_path)), ga.FileStore(str(tmp_path)))
+
+
+def test_file_store_weird_keys(tmp_path):
+    store = ga.FileStore(str(tmp_path))
+    store.set("a/b/c:d", b"x")
+    assert store.get("a/b/c:d") == b"x"
+
+
+def test_prefix_store():
+    inner = ga.HashStore()
+    a = ga.PrefixStore("ns1", inner)
+    b = ga.PrefixStore("ns2", inner)
+    a.set("k", b"va")
+    b.set("k", b"vb")
+    assert a.get("k") == b"va"
+    assert b.get("k") == b"vb"
+
+
+def test_prefix_store_exercise():
+    inner = ga.HashStore()
+    _exercise(ga.PrefixStore("p", inner), ga.PrefixStore("p", inner))
+
+
+def test_tcp_store():
+    import random
+
+    port = random.randint(20000, 40000)
+    server = ga.TcpStore("127.0.0.1", port, is_server=True)
+    client = ga.TcpStore("127.0.0.1", port, is_server=False)
+    _exercise(server, client)
+    _exercise(client, server)

@@ -1,0 +1,278 @@
+"""TCP transport tests: tagged send/recv, recv-from-any, bound buffers,
+timeouts, aborts (reference parity: gloo/test/send_recv_test.cc,
+transport_test.cc, buffer_test.cc)."""
+import numpy as np
+import pytest
+
+import gloo_amd as ga
+
+
+def test_send_recv_basic(spawn_threads):
+    def fn(ctx, rank, size):
+        n = 1000
+        buf = np.full(n, rank, dtype=np.float32)
+        ub = ctx.create_unbound_buffer(buf.ctypes.data, buf.nbytes)
+        peer = (rank + 1) % size
+        src = (rank - 1) % size
+        out = np.zeros(n, dtype=np.float32)
+        ub_out = ctx.create_unbound_buffer(out.ctypes.data, out.nbytes)
+        ub_out.recv(src, slot=7)
+        ub.send(peer, slot=7)
+        ok, got_src = ub_out.wait_recv()
+        assert ok and got_src == src
+        ub.wait_send()
+        assert np.all(out == src)
+        return True
+
+    assert all(spawn_threads(4, fn))
+
+
+def test_send_recv_offsets(spawn_threads):
+    def fn(ctx, rank, size):
+        buf = np.arange(100, dtype=np.int64) * (rank + 1)
+        ub = ctx.create_unbound_buffer(buf.ctypes.data, buf.nbytes)
+        if rank == 0:
+            ub.send(1, slot=3, offset=80, nbytes=160)  # elements 10..30
+        elif rank == 1:
+            out = np.zeros(100, dtype=np.int64)
+            ub2 = ctx.create_unbound_buffer(out.ctypes.data, out.nbytes)
+            ub2.recv(0, slot=3, offset=0, nbytes=160)
+            ub2.wait_recv()
+            assert np.all(out[:20] == np.arange(10, 30, dtype=np.int64))
+        if rank == 0:
+            ub.wait_send()
+        return True
+
+    spawn_threads(2, fn)
+
+
+def test_send_to_self(spawn_threads):
+    def fn(ctx, rank, size):
+        a = np.arange(50, dtype=np.float64)
+        b = np.zeros(50, dtype=np.float64)
+        ua = ctx.create_unbound_buffer(a.ctypes.data, a.nbytes)
+        ub = ctx.create_unbound_buffer(b.ctypes.data, b.nbytes)
+        ua.send(rank, slot=11)
+        ub.recv(rank, slot=11)
+        ub.wait_recv()
+        ua.wait_send()
+        assert np.all(b == a)
+        return True
+
+    spawn_threads(2, fn)
+
+
+def test_recv_from_any(spawn_threads):
+    """Rank 0 posts recv-from-any; every other rank sends once."""
+
+    def fn(ctx, rank, size):
+        n = 16
+        if rank == 0:
+            out = np.zeros(n, dtype=np.float32)
+            seen = []
+            for _ in range(size - 1):
+                ub = ctx.create_unbound_buffer(out.ctypes.data, out.nbytes)
+                ub.recv_any(list(range(1, size)), slot=21)
+                ok, src = ub.wait_recv()
+                assert ok
+                assert np.all(out == src)
+                seen.append(src)
+            assert sorted(seen) == list(range(1, size))
+        else:
+            buf = np.full(n, rank, dtype=np.float32)
+            ub = ctx.create_unbound_buffer(buf.ctypes.data, buf.nbytes)
+            ub.send(0, slot=21)
+            ub.wait_send()
+        return True
+
+    spawn_threads(4, fn)
+
+
+def test_recv_from_any_subset_ordering(spawn_threads):
+    """Sends race ahead of the posted any-recv (SEND_READY tally path)."""
+
+    def fn(ctx, rank, size):
+        import time
+
+        n = 8
+        if rank == 0:
+            time.sleep(0.3)  # let SEND_READYs arrive first
+            out = np.zeros(n, dtype=np.float32)
+            got = []
+            for _ in range(2):
+                ub = ctx.create_unbound_buffer(out.ctypes.data, out.nbytes)
+                ub.recv_any([1, 2], slot=33)
+                ok, src = ub.wait_recv()
+                assert ok
+                got.append(src)
+            assert sorted(got) == [1, 2]
+        elif rank in (1, 2):
+            buf = np.full(n, rank, dtype=np.float32)
+            ub = ctx.create_unbound_buffer(buf.ctypes.data, buf.nbytes)
+            ub.send(0, slot=33)
+            ub.wait_send()
+        return True
+
+    spawn_threads(3, fn)
+
+
+def test_mixed_targeted_and_any_recv(spawn_threads):
+    """A targeted recv must not steal/strand tally entries for any-recv."""
+
+    def fn(ctx, rank, size):
+        import time
+
+        n = 4
+        if rank == 0:
+            time.sleep(0.2)
+            a = np.zeros(n, dtype=np.float32)
+            b = np.zeros(n, dtype=np.float32)
+            ua = ctx.create_unbound_buffer(a.ctypes.data, a.nbytes)
+            ua.recv(1, slot=5)  # targeted at rank 1's first send
+            ua.wait_recv()
+            assert np.all(a == 1)
+            ub = ctx.create_unbound_buffer(b.ctypes.data, b.nbytes)
+            ub.recv_any([1, 2], slot=5)
+            ok, src = ub.wait_recv()
+            assert ok and src == 2 and np.all(b == 2)
+        elif rank == 1:
+            buf = np.full(n, 1, dtype=np.float32)
+            ub = ctx.create_unbound_buffer(buf.ctypes.data, buf.nbytes)
+            ub.send(0, slot=5)
+            ub.wait_send()
+        elif rank == 2:
+            buf = np.full(n, 2, dtype=np.float32)
+            ub = ctx.create_unbound_buffer(buf.ctypes.data, buf.nbytes)
+            ub.send(0, slot=5)
+            ub.wait_send()
+        return True
+
+    spawn_threads(3, fn)
+
+
+def test_many_concurrent_slots(spawn_threads):
+    def fn(ctx, rank, size):
+        n = 64
+        bufs = []
+        peer = (rank + 1) % size
+        src = (rank - 1) % size
+        for s in range(32):
+            out = np.zeros(n, dtype=np.int32)
+            ub = ctx.create_unbound_buffer(out.ctypes.data, out.nbytes)
+            ub.recv(src, slot=100 + s)
+            bufs.append((ub, out))
+        sends = []
+        for s in range(32):
+            data = np.full(n, s, dtype=np.int32)
+            ub = ctx.create_unbound_buffer(data.ctypes.data, data.nbytes)
+            ub.send(peer, slot=100 + s)
+            sends.append((ub, data))
+        for s, (ub, out) in enumerate(bufs):
+            ub.wait_recv()
+            assert np.all(out == s)
+        for ub, _ in sends:
+            ub.wait_send()
+        return True
+
+    spawn_threads(3, fn)
+
+
+def test_bound_buffers(spawn_threads):
+    """Legacy one-sided-write style buffers with remote offset."""
+
+    def fn(ctx, rank, size):
+        n = 100
+        src = np.arange(n, dtype=np.float32) + rank * 1000
+        dst = np.zeros(n, dtype=np.float32)
+        peer = (rank + 1) % size
+        left = (rank - 1) % size
+        sb = ctx.get_pair(peer).create_send_buffer(42, src.ctypes.data, src.nbytes)
+        rb = ctx.get_pair(left).create_recv_buffer(42, dst.ctypes.data, dst.nbytes)
+        # send elements [10, 35) to remote offset 40 bytes (element 10)
+        sb.send(offset=40, length=100, roffset=40)
+        rb.wait_recv()
+        sb.wait_send()
+        assert np.all(dst[10:35] == np.arange(10, 35, dtype=np.float32) + left * 1000)
+        assert np.all(dst[:10] == 0) and np.all(dst[35:] == 0)
+        return True
+
+    spawn_threads(3, fn)
+
+
+def test_bound_buffer_early_data(spawn_threads):
+    """One-sided write that lands before createRecvBuffer is called."""
+
+    def fn(ctx, rank, size):
+        import time
+
+        n = 16
+        if rank == 0:
+            src = np.full(n, 7.0, dtype=np.float32)
+            sb = ctx.get_pair(1).create_send_buffer(9, src.ctypes.data, src.nbytes)
+            sb.send()
+            sb.wait_send()
+        else:
+            time.sleep(0.3)  # data arrives before registration
+            dst = np.zeros(n, dtype=np.float32)
+            rb = ctx.get_pair(0).create_recv_buffer(9, dst.ctypes.data, dst.nbytes)
+            rb.wait_recv()
+            assert np.all(dst == 7.0)
+        return True
+
+    spawn_threads(2, fn)
+
+
+def test_recv_timeout(spawn_threads):
+    def fn(ctx, rank, size):
+        import time
+
+        buf = np.zeros(4, dtype=np.float32)
+        ub = ctx.create_unbound_buffer(buf.ctypes.data, buf.nbytes)
+        if rank == 0:
+            ub.recv(1, slot=77)
+            with pytest.raises(ga.TimeoutError):
+                ub.wait_recv(timeout_ms=200)
+        else:
+            time.sleep(1.0)  # stay alive so EOF doesn't beat the timeout
+        return True
+
+    spawn_threads(2, fn)
+
+
+def test_abort_wait(spawn_threads):
+    def fn(ctx, rank, size):
+        import threading as th
+
+        buf = np.zeros(4, dtype=np.float32)
+        ub = ctx.create_unbound_buffer(buf.ctypes.data, buf.nbytes)
+        if rank == 0:
+            ub.recv(1, slot=78)
+            t = th.Timer(0.2, ub.abort_wait_recv)
+            t.start()
+            ok, _ = ub.wait_recv(timeout_ms=10000)
+            assert not ok  # aborted
+            t.join()
+        return True
+
+    spawn_threads(2, fn)
+
+
+def test_big_transfer(spawn_threads):
+    """Multi-MB payload exercises partial read/write state machines."""
+
+    def fn(ctx, rank, size):
+        n = 3_000_000
+        if rank == 0:
+            data = np.arange(n, dtype=np.float32)
+            ub = ctx.create_unbound_buffer(data.ctypes.data, data.nbytes)
+            ub.send(1, slot=88)
+            ub.wait_send()
+        else:
+            out = np.zeros(n, dtype=np.float32)
+            ub = ctx.create_unbound_buffer(out.ctypes.data, out.nbytes)
+            ub.recv(0, slot=88)
+            ub.wait_recv()
+            assert np.array_equal(out, np.arange(n, dtype=np.float32))
+        return True
+
+    spawn_threads(2, fn)
