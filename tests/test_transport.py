@@ -242,6 +242,7 @@ def test_recv_timeout(spawn_threads):
 def test_abort_wait(spawn_threads):
     def fn(ctx, rank, size):
         import threading as th
+        import time
 
         buf = np.zeros(4, dtype=np.float32)
         ub = ctx.create_unbound_buffer(buf.ctypes.data, buf.nbytes)
@@ -252,6 +253,8 @@ def test_abort_wait(spawn_threads):
             ok, _ = ub.wait_recv(timeout_ms=10000)
             assert not ok  # aborted
             t.join()
+        else:
+            time.sleep(1.0)  # keep the pair open while rank 0 waits
         return True
 
     spawn_threads(2, fn)
