@@ -14,6 +14,8 @@
 #include "collectives/reduce_fns.h"
 #include "common/store.h"
 #include "context.h"
+#include "hip/algorithms.h"
+#include "hip/kernels.h"
 #include "rendezvous/stores.h"
 #include "transport/tcp/context.h"
 #include "transport/tcp/device.h"
@@ -568,4 +570,120 @@ PYBIND11_MODULE(_C, m) {
       py::arg("op") = ReduceOp::SUM,
       py::arg("tag") = 0,
       py::arg("timeout_ms") = 0);
+
+  // --- HIP / xGMI device collectives ---------------------------------------
+  m.def("hip_available", &hip::available);
+  m.def("hip_device_count", [] {
+    return hip::available() ? hip::deviceCount() : 0;
+  });
+
+  // Raw kernel entry points for numerics tests (device pointers; blocks
+  // until complete on the default-constructed stream semantics: caller
+  // synchronizes via torch).
+  m.def(
+      "hip_reduce2",
+      [](uintptr_t dst, uintptr_t a, uintptr_t b, size_t n, DType dt,
+         ReduceOp op, uintptr_t stream) {
+        py::gil_scoped_release rel;
+        hip::launchReduce2(
+            reinterpret_cast<void*>(dst),
+            reinterpret_cast<const void*>(a),
+            reinterpret_cast<const void*>(b),
+            n,
+            dt,
+            op,
+            reinterpret_cast<hipStream_t>(stream));
+      },
+      py::arg("dst"),
+      py::arg("a"),
+      py::arg("b"),
+      py::arg("n"),
+      py::arg("dtype"),
+      py::arg("op"),
+      py::arg("stream") = 0);
+
+  m.def(
+      "hip_allreduce_local",
+      [](std::vector<uintptr_t> ptrs, size_t n, DType dt, ReduceOp op,
+         int device) {
+        std::vector<void*> p;
+        for (auto v : ptrs) {
+          p.push_back(reinterpret_cast<void*>(v));
+        }
+        py::gil_scoped_release rel;
+        hip::hipAllreduceLocal(p, n, dt, op, device);
+      },
+      py::arg("ptrs"),
+      py::arg("n"),
+      py::arg("dtype"),
+      py::arg("op"),
+      py::arg("device") = 0);
+
+  py::class_<hip::HipAllreduceRing>(m, "HipAllreduceRing")
+      .def(
+          py::init([](std::shared_ptr<Context> ctx, int device, bool chunked,
+                      size_t inboxCap) {
+            py::gil_scoped_release rel;
+            return std::make_unique<hip::HipAllreduceRing>(
+                ctx, device, chunked, inboxCap);
+          }),
+          py::arg("context"),
+          py::arg("device"),
+          py::arg("chunked") = true,
+          py::arg("inbox_cap") = 0)
+      .def(
+          "run",
+          [](hip::HipAllreduceRing& a, uintptr_t ptr, size_t n, DType dt,
+             ReduceOp op) {
+            py::gil_scoped_release rel;
+            a.run(reinterpret_cast<void*>(ptr), n, dt, op);
+          },
+          py::arg("ptr"),
+          py::arg("elements"),
+          py::arg("dtype") = DType::F32,
+          py::arg("op") = ReduceOp::SUM);
+
+  py::class_<hip::HipAllreduceHalvingDoubling>(m, "HipAllreduceHalvingDoubling")
+      .def(
+          py::init([](std::shared_ptr<Context> ctx, int device,
+                      size_t inboxCap) {
+            py::gil_scoped_release rel;
+            return std::make_unique<hip::HipAllreduceHalvingDoubling>(
+                ctx, device, inboxCap);
+          }),
+          py::arg("context"),
+          py::arg("device"),
+          py::arg("inbox_cap") = 0)
+      .def(
+          "run",
+          [](hip::HipAllreduceHalvingDoubling& a, uintptr_t ptr, size_t n,
+             DType dt, ReduceOp op) {
+            py::gil_scoped_release rel;
+            a.run(reinterpret_cast<void*>(ptr), n, dt, op);
+          },
+          py::arg("ptr"),
+          py::arg("elements"),
+          py::arg("dtype") = DType::F32,
+          py::arg("op") = ReduceOp::SUM);
+
+  py::class_<hip::HipBroadcastOneToAll>(m, "HipBroadcastOneToAll")
+      .def(
+          py::init([](std::shared_ptr<Context> ctx, int device, int root,
+                      int numStreams) {
+            py::gil_scoped_release rel;
+            return std::make_unique<hip::HipBroadcastOneToAll>(
+                ctx, device, root, numStreams);
+          }),
+          py::arg("context"),
+          py::arg("device"),
+          py::arg("root") = 0,
+          py::arg("num_streams") = 4)
+      .def(
+          "run",
+          [](hip::HipBroadcastOneToAll& a, uintptr_t ptr, size_t bytes) {
+            py::gil_scoped_release rel;
+            a.run(reinterpret_cast<void*>(ptr), bytes);
+          },
+          py::arg("ptr"),
+          py::arg("bytes"));
 }

@@ -7,6 +7,7 @@
 #include <cstring>
 
 #include "collectives/collectives.h"
+#include "collectives/schedule.h"
 #include "common/logging.h"
 #include "common/utils.h"
 #include "types.h"
@@ -15,37 +16,10 @@ namespace glooamd {
 
 namespace {
 
-struct Seg {
-  size_t off; // element offset
-  size_t len; // element count
-};
-
-// Block b of an N-element buffer split across P ranks (tail-clamped).
-Seg blockOf(size_t N, int P, int b) {
-  size_t perRank = (N + P - 1) / P;
-  size_t start = std::min<size_t>(static_cast<size_t>(b) * perRank, N);
-  size_t end = std::min<size_t>(start + perRank, N);
-  return {start, end - start};
-}
-
-// Segment s (of S) within block b.
-Seg segmentOf(size_t N, int P, int b, int s, int S) {
-  Seg blk = blockOf(N, P, b);
-  size_t perSeg = (blk.len + S - 1) / S;
-  size_t start = std::min(blk.off + static_cast<size_t>(s) * perSeg,
-                          blk.off + blk.len);
-  size_t end = std::min(start + perSeg, blk.off + blk.len);
-  return {start, end - start};
-}
-
-// Sub-span j of base for span [off, len) — consistent across ranks.
-Seg subspanOf(Seg span, int j, int base) {
-  size_t q = span.len / base;
-  size_t r = span.len % base;
-  size_t start = span.off + static_cast<size_t>(j) * q + std::min<size_t>(j, r);
-  size_t len = q + (static_cast<size_t>(j) < r ? 1 : 0);
-  return {start, len};
-}
+using sched::Seg;
+using sched::blockOf;
+using sched::segmentOf;
+using sched::subspanOf;
 
 void localReduceInputs(const AllreduceOptions& opts, char* out) {
   const size_t nbytes = opts.elements * opts.elementSize;

@@ -1,0 +1,55 @@
+// Shared block/segment math for ring and bcube schedules (used by the
+// CPU v2 collectives and the hip_* device algorithms so both sides of a
+// wire agree on every chunk boundary).
+#pragma once
+
+#include <algorithm>
+#include <cstddef>
+
+namespace glooamd {
+namespace sched {
+
+struct Seg {
+  size_t off; // element offset
+  size_t len; // element count
+};
+
+// Block b of an N-element buffer split across P ranks (tail-clamped).
+inline Seg blockOf(size_t N, int P, int b) {
+  size_t perRank = (N + P - 1) / P;
+  size_t start = std::min<size_t>(static_cast<size_t>(b) * perRank, N);
+  size_t end = std::min<size_t>(start + perRank, N);
+  return {start, end - start};
+}
+
+// Segment s (of S) within block b.
+inline Seg segmentOf(size_t N, int P, int b, int s, int S) {
+  Seg blk = blockOf(N, P, b);
+  size_t perSeg = (blk.len + S - 1) / S;
+  size_t start =
+      std::min(blk.off + static_cast<size_t>(s) * perSeg, blk.off + blk.len);
+  size_t end = std::min(start + perSeg, blk.off + blk.len);
+  return {start, end - start};
+}
+
+// Sub-span j of `base` equal parts of span (remainder spread left).
+inline Seg subspanOf(Seg span, int j, int base) {
+  size_t q = span.len / base;
+  size_t r = span.len % base;
+  size_t start =
+      span.off + static_cast<size_t>(j) * q + std::min<size_t>(j, r);
+  size_t len = q + (static_cast<size_t>(j) < r ? 1 : 0);
+  return {start, len};
+}
+
+// Chunk c (of nc) within a span.
+inline Seg chunkOf(Seg span, int c, int nc) {
+  size_t per = (span.len + nc - 1) / nc;
+  size_t start = std::min(span.off + static_cast<size_t>(c) * per,
+                          span.off + span.len);
+  size_t end = std::min(start + per, span.off + span.len);
+  return {start, end - start};
+}
+
+} // namespace sched
+} // namespace glooamd

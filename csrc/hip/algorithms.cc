@@ -1,0 +1,533 @@
+#include "hip/algorithms.h"
+
+#include <chrono>
+#include <thread>
+
+#include "collectives/collectives.h"
+#include "collectives/schedule.h"
+#include "common/utils.h"
+#include "hip/kernels.h"
+
+namespace glooamd {
+namespace hip {
+
+using sched::Seg;
+using sched::chunkOf;
+using sched::segmentOf;
+using sched::subspanOf;
+
+size_t onDeviceThreshold() {
+  static size_t v = static_cast<size_t>(
+      getEnvInt("GLOO_AMD_DEVICE_THRESHOLD", 256 * 1024));
+  return v;
+}
+
+namespace {
+constexpr size_t kDefaultInboxCap = 4 << 20; // 4 MiB per inbox
+
+// Wait for `ev` with a deadline; on expiry poison the mesh flags so the
+// GPU-side waits drain, then throw. Fail-fast model like the reference's
+// pair exception fan-out.
+void watchdogWait(
+    HipEvent& ev,
+    XgmiMesh& mesh,
+    std::chrono::milliseconds timeout,
+    const char* what) {
+  auto deadline = std::chrono::steady_clock::now() + timeout;
+  while (!ev.query()) {
+    if (timeout.count() > 0 && std::chrono::steady_clock::now() > deadline) {
+      mesh.poisonFlags();
+      (void)hipDeviceSynchronize();
+      GA_THROW_IO("hip collective timed out in ", what);
+    }
+    std::this_thread::yield();
+  }
+}
+} // namespace
+
+// ===========================================================================
+// HipAllreduceRing
+// ===========================================================================
+
+HipAllreduceRing::HipAllreduceRing(
+    std::shared_ptr<Context> ctx,
+    int device,
+    bool chunked,
+    size_t inboxCap)
+    : ctx_(std::move(ctx)),
+      device_(device),
+      chunked_(chunked),
+      inboxCap_(inboxCap == 0 ? kDefaultInboxCap : inboxCap) {
+  GA_HIP_CHECK(hipSetDevice(device_));
+  mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, inboxCap_);
+  cs_ = std::make_unique<HipStream>(device_, /*highPriority=*/true);
+  ks_ = std::make_unique<HipStream>(device_);
+  initEvent_ = std::make_unique<HipEvent>(device_);
+  doneEvent_ = std::make_unique<HipEvent>(device_);
+  fDATA_ = mesh_->allocFlags(2);
+  fACK_ = mesh_->allocFlags(2);
+}
+
+void HipAllreduceRing::run(
+    void* devPtr,
+    size_t elements,
+    DType dtype,
+    ReduceOp op) {
+  GA_HIP_CHECK(hipSetDevice(device_));
+  const size_t es = dtypeSize(dtype);
+  const size_t bytes = elements * es;
+  if (ctx_->size == 1 || elements == 0) {
+    return;
+  }
+  if (bytes < onDeviceThreshold()) {
+    runHostStaged(static_cast<char*>(devPtr), bytes, elements, dtype, op);
+  } else {
+    runDevice(static_cast<char*>(devPtr), bytes, elements, dtype, op);
+  }
+}
+
+void HipAllreduceRing::runHostStaged(
+    char* buf,
+    size_t bytes,
+    size_t n,
+    DType dt,
+    ReduceOp op) {
+  if (hostStageCap_ < bytes) {
+    if (hostStage_ != nullptr) {
+      (void)hipHostFree(hostStage_);
+    }
+    GA_HIP_CHECK(hipHostMalloc(&hostStage_, bytes));
+    hostStageCap_ = bytes;
+  }
+  GA_HIP_CHECK(hipMemcpyAsync(
+      hostStage_, buf, bytes, hipMemcpyDeviceToHost, cs_->stream()));
+  cs_->synchronize();
+  AllreduceOptions opts(ctx_);
+  opts.outputs = {hostStage_};
+  opts.elements = n;
+  opts.elementSize = dtypeSize(dt);
+  opts.reduce = cpuReduceFn(dt, op);
+  opts.tag = 0x7fff0000u; // reserved tag band for hip host staging
+  allreduce(opts);
+  GA_HIP_CHECK(hipMemcpyAsync(
+      buf, hostStage_, bytes, hipMemcpyHostToDevice, cs_->stream()));
+  cs_->synchronize();
+}
+
+void HipAllreduceRing::runDevice(
+    char* buf,
+    size_t bytes,
+    size_t n,
+    DType dt,
+    ReduceOp op) {
+  const int P = ctx_->size;
+  const int r = ctx_->rank;
+  const size_t es = bytes / n;
+  const int right = (r + 1) % P;
+  const int left = (r - 1 + P) % P;
+
+  const size_t perRank = (n + P - 1) / P;
+  const int S = chunked_
+      ? std::max<int>(
+            2,
+            static_cast<int>(
+                (perRank * es + inboxCap_ - 1) / inboxCap_))
+      : 1;
+  const size_t segCapBytes = ((perRank + S - 1) / S) * es;
+  mesh_->ensureCapacity(bytes, segCapBytes);
+
+  struct Step {
+    Seg send;
+    Seg recv;
+    bool reduceStep;
+  };
+  const int K1 = (P - 1) * S;
+  std::vector<Step> steps;
+  steps.reserve(2 * K1);
+  for (int k = 0; k < K1; k++) {
+    int i = k / S, s = k % S;
+    steps.push_back({segmentOf(n, P, (r - i + P) % P, s, S),
+                     segmentOf(n, P, (r - i - 1 + 2 * P) % P, s, S),
+                     true});
+  }
+  for (int k = 0; k < K1; k++) {
+    int i = k / S, s = k % S;
+    steps.push_back({segmentOf(n, P, (r + 1 - i + P) % P, s, S),
+                     segmentOf(n, P, (r - i + P) % P, s, S),
+                     false});
+  }
+  const int K = static_cast<int>(steps.size());
+
+  const int pool = S + 2;
+  while (static_cast<int>(events_.size()) < pool) {
+    events_.push_back(std::make_unique<HipEvent>(device_));
+  }
+
+  char* work = mesh_->work();
+  auto csm = cs_->stream();
+  auto ksm = ks_->stream();
+
+  // Stage in.
+  GA_HIP_CHECK(
+      hipMemcpyAsync(work, buf, bytes, hipMemcpyDeviceToDevice, csm));
+  initEvent_->record(csm);
+
+  auto seqOf = [&](int k) { return seqBase_ + k + 1; };
+
+  for (int k = 0; k < K; k++) {
+    const Step& st = steps[k];
+    const int par = k & 1;
+
+    // --- sender side (ks): work[send] -> right's inbox[par] ---
+    // Inbox-reuse gate: right must have consumed what we last put in
+    // inbox[par] (covers both the k-2 step of this run and the tail of
+    // the previous run).
+    const uint64_t prevAck = (k >= 2) ? seqOf(k - 2)
+                                      : (seqBase_ >= 2 ? seqBase_ - 2 + par + 1
+                                                       : 0);
+    if (prevAck > 0) {
+      launchWaitFlagGte(mesh_->flag(fACK_ + par), prevAck, ksm);
+    }
+    if (k >= S) {
+      events_[(k - S) % pool]->streamWait(ksm);
+    } else {
+      initEvent_->streamWait(ksm);
+    }
+    if (st.send.len > 0) {
+      GA_HIP_CHECK(hipMemcpyAsync(
+          mesh_->peerInbox(right, par),
+          work + st.send.off * es,
+          st.send.len * es,
+          hipMemcpyDeviceToDevice,
+          ksm));
+    }
+    launchWriteFlag(mesh_->peerFlag(right, fDATA_ + par), seqOf(k), ksm);
+
+    // --- receiver side (cs): inbox[par] -> work[recv] ---
+    launchWaitFlagGte(mesh_->flag(fDATA_ + par), seqOf(k), csm);
+    if (st.recv.len > 0) {
+      if (st.reduceStep) {
+        launchReduce2(
+            work + st.recv.off * es,
+            work + st.recv.off * es,
+            mesh_->inbox(par),
+            st.recv.len,
+            dt,
+            op,
+            csm);
+      } else {
+        GA_HIP_CHECK(hipMemcpyAsync(
+            work + st.recv.off * es,
+            mesh_->inbox(par),
+            st.recv.len * es,
+            hipMemcpyDeviceToDevice,
+            csm));
+      }
+    }
+    launchWriteFlag(mesh_->peerFlag(left, fACK_ + par), seqOf(k), csm);
+    events_[k % pool]->record(csm);
+  }
+
+  // Stage out.
+  GA_HIP_CHECK(
+      hipMemcpyAsync(buf, work, bytes, hipMemcpyDeviceToDevice, csm));
+  doneEvent_->record(csm);
+
+  auto timeout = ctx_->getTimeout();
+  watchdogWait(*doneEvent_, *mesh_, timeout, "hip_allreduce_ring (cs)");
+  // ks tail: the final data-flag writes; drains once peers ack.
+  initEvent_->record(ksm);
+  watchdogWait(*initEvent_, *mesh_, timeout, "hip_allreduce_ring (ks)");
+  cs_->synchronize();
+  ks_->synchronize();
+  seqBase_ += K;
+}
+
+// ===========================================================================
+// HipAllreduceHalvingDoubling
+// ===========================================================================
+
+HipAllreduceHalvingDoubling::HipAllreduceHalvingDoubling(
+    std::shared_ptr<Context> ctx,
+    int device,
+    size_t inboxCap)
+    : ctx_(std::move(ctx)),
+      device_(device),
+      inboxCap_(inboxCap == 0 ? kDefaultInboxCap : inboxCap) {
+  const int P = ctx_->size;
+  GA_ENFORCE(
+      P > 0 && (P & (P - 1)) == 0,
+      "hip_allreduce_halving_doubling requires power-of-2 size, got ",
+      P,
+      " (use hip_allreduce_ring)");
+  log2P_ = 0;
+  while ((1 << log2P_) < P) {
+    log2P_++;
+  }
+  GA_HIP_CHECK(hipSetDevice(device_));
+  mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, inboxCap_);
+  cs_ = std::make_unique<HipStream>(device_, true);
+  ks_ = std::make_unique<HipStream>(device_);
+  stepEvent_ = std::make_unique<HipEvent>(device_);
+  initEvent_ = std::make_unique<HipEvent>(device_);
+  doneEvent_ = std::make_unique<HipEvent>(device_);
+  const int T = std::max(1, log2P_);
+  fDATA_ = mesh_->allocFlags(T * 2);
+  fACK_ = mesh_->allocFlags(T * 2);
+  fAGD_ = mesh_->allocFlags(T);
+}
+
+void HipAllreduceHalvingDoubling::run(
+    void* devPtr,
+    size_t elements,
+    DType dtype,
+    ReduceOp op) {
+  GA_HIP_CHECK(hipSetDevice(device_));
+  const int P = ctx_->size;
+  const int r = ctx_->rank;
+  const size_t es = dtypeSize(dtype);
+  const size_t bytes = elements * es;
+  if (P == 1 || elements == 0) {
+    return;
+  }
+  char* buf = static_cast<char*>(devPtr);
+  const int T = log2P_;
+  mesh_->ensureCapacity(bytes, inboxCap_);
+  // Sub-inbox layout: the 2*inboxCap inbox region split into T steps x 2
+  // parities.
+  const size_t subBytes = (2 * mesh_->inboxCap()) / (2 * T);
+  auto subInbox = [&](int t, int par) {
+    return mesh_->inbox(0) + (static_cast<size_t>(t) * 2 + par) * subBytes;
+  };
+  auto peerSubInbox = [&](int rank, int t, int par) {
+    return mesh_->peerInbox(rank, 0) +
+        (static_cast<size_t>(t) * 2 + par) * subBytes;
+  };
+
+  char* work = mesh_->work();
+  auto csm = cs_->stream();
+  auto ksm = ks_->stream();
+
+  GA_HIP_CHECK(
+      hipMemcpyAsync(work, buf, bytes, hipMemcpyDeviceToDevice, csm));
+  initEvent_->record(csm);
+
+  // Host-tracked per-(step,parity) last-issued ack/data seqs; identical
+  // on every rank (symmetric schedule), persistent across runs.
+  if (lastAck_.empty()) {
+    lastAck_.assign(T * 2, 0);
+    lastAgd_.assign(T, 0);
+  }
+
+  Seg span{0, elements};
+  std::vector<Seg> spanAt(T);
+  std::vector<std::unique_ptr<HipEvent>>& evs = stepEvents_;
+  while (static_cast<int>(evs.size()) < T) {
+    evs.push_back(std::make_unique<HipEvent>(device_));
+  }
+
+  // ---- reduce-scatter (vector halving, distance doubling) ----
+  for (int t = 0; t < T; t++) {
+    spanAt[t] = span;
+    const int peer = r ^ (1 << t);
+    const int bit = (r >> t) & 1;
+    Seg kp = subspanOf(span, bit, 2);
+    Seg gv = subspanOf(span, 1 - bit, 2);
+    const size_t maxHalf = std::max(kp.len, gv.len) * es;
+    const int nc =
+        std::max<int>(1, static_cast<int>((maxHalf + subBytes - 1) / subBytes));
+
+    // gv must be fully produced (previous step's reduces on cs).
+    if (t == 0) {
+      initEvent_->streamWait(ksm);
+    } else {
+      evs[t - 1]->streamWait(ksm);
+    }
+    for (int c = 0; c < nc; c++) {
+      const int par = c & 1;
+      seq_++;
+      const uint64_t dseq = seq_;
+      if (lastAck_[t * 2 + par] > 0) {
+        launchWaitFlagGte(
+            mesh_->flag(fACK_ + t * 2 + par), lastAck_[t * 2 + par], ksm);
+      }
+      Seg gch = chunkOf(gv, c, nc);
+      if (gch.len > 0) {
+        GA_HIP_CHECK(hipMemcpyAsync(
+            peerSubInbox(peer, t, par),
+            work + gch.off * es,
+            gch.len * es,
+            hipMemcpyDeviceToDevice,
+            ksm));
+      }
+      launchWriteFlag(mesh_->peerFlag(peer, fDATA_ + t * 2 + par), dseq, ksm);
+
+      launchWaitFlagGte(mesh_->flag(fDATA_ + t * 2 + par), dseq, csm);
+      Seg kch = chunkOf(kp, c, nc);
+      if (kch.len > 0) {
+        launchReduce2(
+            work + kch.off * es,
+            work + kch.off * es,
+            subInbox(t, par),
+            kch.len,
+            dtype,
+            op,
+            csm);
+      }
+      launchWriteFlag(mesh_->peerFlag(peer, fACK_ + t * 2 + par), dseq, csm);
+      lastAck_[t * 2 + par] = dseq;
+    }
+    evs[t]->record(csm);
+    span = kp;
+  }
+
+  // ---- allgather (mirror; direct writes into peer's work) ----
+  Seg cur = span;
+  for (int t = T - 1; t >= 0; t--) {
+    const int peer = r ^ (1 << t);
+    if (t == T - 1) {
+      evs[T - 1]->streamWait(ksm);
+    } else {
+      // Halves received in step t+1 are part of cur now.
+      if (lastAgd_[t + 1] > 0) {
+        launchWaitFlagGte(mesh_->flag(fAGD_ + t + 1), lastAgd_[t + 1], ksm);
+      }
+    }
+    seq_++;
+    const uint64_t aseq = seq_;
+    if (cur.len > 0) {
+      GA_HIP_CHECK(hipMemcpyAsync(
+          mesh_->peerWork(peer) + cur.off * es,
+          work + cur.off * es,
+          cur.len * es,
+          hipMemcpyDeviceToDevice,
+          ksm));
+    }
+    launchWriteFlag(mesh_->peerFlag(peer, fAGD_ + t), aseq, ksm);
+    lastAgd_[t] = aseq;
+    cur = spanAt[t];
+  }
+
+  // Final: wait all incoming allgather halves, stage out.
+  for (int t = 0; t < T; t++) {
+    launchWaitFlagGte(mesh_->flag(fAGD_ + t), lastAgd_[t], csm);
+  }
+  GA_HIP_CHECK(
+      hipMemcpyAsync(buf, work, bytes, hipMemcpyDeviceToDevice, csm));
+  doneEvent_->record(csm);
+
+  auto timeout = ctx_->getTimeout();
+  watchdogWait(*doneEvent_, *mesh_, timeout, "hip_allreduce_hd (cs)");
+  initEvent_->record(ksm);
+  watchdogWait(*initEvent_, *mesh_, timeout, "hip_allreduce_hd (ks)");
+  cs_->synchronize();
+  ks_->synchronize();
+}
+
+// ===========================================================================
+// HipBroadcastOneToAll
+// ===========================================================================
+
+HipBroadcastOneToAll::HipBroadcastOneToAll(
+    std::shared_ptr<Context> ctx,
+    int device,
+    int root,
+    int numStreams)
+    : ctx_(std::move(ctx)), device_(device), root_(root) {
+  GA_HIP_CHECK(hipSetDevice(device_));
+  mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, 4096);
+  cs_ = std::make_unique<HipStream>(device_);
+  for (int i = 0; i < numStreams; i++) {
+    fanout_.push_back(std::make_unique<HipStream>(device_));
+  }
+  fBDATA_ = mesh_->allocFlags(1);
+  fBACK_ = mesh_->allocFlags(ctx_->size);
+}
+
+void HipBroadcastOneToAll::run(void* devPtr, size_t bytes) {
+  GA_HIP_CHECK(hipSetDevice(device_));
+  const int P = ctx_->size;
+  const int r = ctx_->rank;
+  if (P == 1 || bytes == 0) {
+    return;
+  }
+  mesh_->ensureCapacity(bytes, 4096);
+  char* work = mesh_->work();
+  char* buf = static_cast<char*>(devPtr);
+  seq_++;
+
+  if (r == root_) {
+    GA_HIP_CHECK(hipMemcpyAsync(
+        work, buf, bytes, hipMemcpyDeviceToDevice, cs_->stream()));
+    HipEvent staged(device_);
+    staged.record(cs_->stream());
+    int si = 0;
+    for (int i = 0; i < P; i++) {
+      if (i == root_) {
+        continue;
+      }
+      auto& st = *fanout_[si++ % fanout_.size()];
+      staged.streamWait(st.stream());
+      if (seq_ > 1) {
+        launchWaitFlagGte(mesh_->flag(fBACK_ + i), seq_ - 1, st.stream());
+      }
+      GA_HIP_CHECK(hipMemcpyAsync(
+          mesh_->peerWork(i),
+          work,
+          bytes,
+          hipMemcpyDeviceToDevice,
+          st.stream()));
+      launchWriteFlag(mesh_->peerFlag(i, fBDATA_), seq_, st.stream());
+    }
+    cs_->synchronize();
+    for (auto& st : fanout_) {
+      st->synchronize();
+    }
+  } else {
+    launchWaitFlagGte(mesh_->flag(fBDATA_), seq_, cs_->stream());
+    GA_HIP_CHECK(hipMemcpyAsync(
+        buf, work, bytes, hipMemcpyDeviceToDevice, cs_->stream()));
+    launchWriteFlag(mesh_->peerFlag(root_, fBACK_ + r), seq_, cs_->stream());
+    cs_->synchronize();
+  }
+}
+
+// ===========================================================================
+// hipAllreduceLocal
+// ===========================================================================
+
+void hipAllreduceLocal(
+    const std::vector<void*>& ptrs,
+    size_t elements,
+    DType dtype,
+    ReduceOp op,
+    int device) {
+  GA_ENFORCE(!ptrs.empty());
+  if (ptrs.size() == 1 || elements == 0) {
+    return;
+  }
+  GA_HIP_CHECK(hipSetDevice(device));
+  HipStream s(device);
+  const size_t es = dtypeSize(dtype);
+  launchReduceN(
+      ptrs[0],
+      const_cast<const void* const*>(ptrs.data()),
+      static_cast<int>(std::min<size_t>(ptrs.size(), 8)),
+      elements,
+      dtype,
+      op,
+      s.stream());
+  // >8 pointers: fold the rest pairwise.
+  for (size_t i = 8; i < ptrs.size(); i++) {
+    launchReduce2(ptrs[0], ptrs[0], ptrs[i], elements, dtype, op, s.stream());
+  }
+  for (size_t i = 1; i < ptrs.size(); i++) {
+    GA_HIP_CHECK(hipMemcpyAsync(
+        ptrs[i], ptrs[0], elements * es, hipMemcpyDeviceToDevice,
+        s.stream()));
+  }
+  s.synchronize();
+}
+
+} // namespace hip
+} // namespace glooamd

@@ -1,0 +1,141 @@
+// Device-side collective algorithms over the xGMI/IPC mesh.
+//
+// Reference parity set (gloo/cuda_allreduce_ring.{h,cc},
+// cuda_allreduce_ring_chunked.{h,cc}, cuda_allreduce_halving_doubling,
+// cuda_allreduce_local, cuda_broadcast_one_to_all), re-derived for
+// MI355X (see csrc/hip/mesh.h header for the transport re-design):
+//
+//  * HipAllreduceRing (chunked=false -> whole-block ring like
+//    cuda_allreduce_ring; chunked=true -> >=2 pipelined segments per
+//    block, the reference's ring_chunked schedule). The entire
+//    schedule — peer SDMA copies, doorbells, reduction kernels — is
+//    enqueued on two HIP streams with GPU-side flag waits; the host
+//    leaves the critical path (the reference pays a host round trip
+//    per chunk, cuda_allreduce_ring_chunked.cc:168-213).
+//  * HipAllreduceHalvingDoubling: recursive vector halving / distance
+//    doubling for power-of-2 sizes; each step's half moves in
+//    inbox-capacity sub-chunks (double-buffered); the allgather mirror
+//    writes straight into the peer's work region (single writer per
+//    region, so no inbox hop is needed).
+//  * HipBroadcastOneToAll: root fans out over several streams so
+//    multiple xGMI links run concurrently.
+//  * hipAllreduceLocal: single-process multi-pointer fused reduce +
+//    broadcast (reference cuda_allreduce_local.cc).
+//
+// Buffers below kOnDeviceThreshold take the host-staged path (pinned
+// D2H -> CPU tcp/shm collective -> H2D), mirroring the reference's
+// kOnDeviceThreshold host/device workspace split (gloo/algorithm.cc:16)
+// with MI355X-tuned default (256 KiB, env GLOO_AMD_DEVICE_THRESHOLD).
+#pragma once
+
+#include <memory>
+#include <vector>
+
+#include "collectives/reduce_fns.h"
+#include "hip/core.h"
+#include "hip/mesh.h"
+
+namespace glooamd {
+namespace hip {
+
+size_t onDeviceThreshold(); // bytes; below this, host-staged path
+
+class HipAllreduceRing {
+ public:
+  HipAllreduceRing(
+      std::shared_ptr<Context> ctx,
+      int device,
+      bool chunked = true,
+      size_t inboxCap = 0 /*0 -> default 4 MiB*/);
+
+  // In-place allreduce of a device buffer. Blocking (streams synced).
+  void run(void* devPtr, size_t elements, DType dtype, ReduceOp op);
+
+  XgmiMesh* mesh() {
+    return mesh_.get();
+  }
+
+ private:
+  void runDevice(char* buf, size_t bytes, size_t n, DType dt, ReduceOp op);
+  void runHostStaged(char* buf, size_t bytes, size_t n, DType dt, ReduceOp op);
+
+  std::shared_ptr<Context> ctx_;
+  int device_;
+  bool chunked_;
+  size_t inboxCap_;
+  std::unique_ptr<XgmiMesh> mesh_;
+  std::unique_ptr<HipStream> cs_; // compute: reduces, local copies, acks
+  std::unique_ptr<HipStream> ks_; // copies out to peers + data doorbells
+  std::vector<std::unique_ptr<HipEvent>> events_;
+  std::unique_ptr<HipEvent> initEvent_;
+  std::unique_ptr<HipEvent> doneEvent_;
+  int fDATA_, fACK_; // flag indices (x2 parity each)
+  uint64_t seqBase_{0};
+  // pinned staging for the host path
+  void* hostStage_{nullptr};
+  size_t hostStageCap_{0};
+};
+
+class HipAllreduceHalvingDoubling {
+ public:
+  HipAllreduceHalvingDoubling(
+      std::shared_ptr<Context> ctx,
+      int device,
+      size_t inboxCap = 0);
+
+  void run(void* devPtr, size_t elements, DType dtype, ReduceOp op);
+
+ private:
+  std::shared_ptr<Context> ctx_;
+  int device_;
+  size_t inboxCap_;
+  int log2P_;
+  std::unique_ptr<XgmiMesh> mesh_;
+  std::unique_ptr<HipStream> cs_;
+  std::unique_ptr<HipStream> ks_;
+  std::unique_ptr<HipEvent> stepEvent_;
+  std::unique_ptr<HipEvent> initEvent_;
+  std::unique_ptr<HipEvent> doneEvent_;
+  std::vector<std::unique_ptr<HipEvent>> stepEvents_;
+  int fDATA_, fACK_, fAGD_; // per step t: [t*2 + parity]
+  uint64_t seq_{0};
+  // Last-issued doorbell seq per (step, parity) / per step; identical on
+  // all ranks (symmetric schedule) and persistent across runs so inbox
+  // reuse is safe across run boundaries.
+  std::vector<uint64_t> lastAck_;
+  std::vector<uint64_t> lastAgd_;
+};
+
+class HipBroadcastOneToAll {
+ public:
+  HipBroadcastOneToAll(
+      std::shared_ptr<Context> ctx,
+      int device,
+      int root = 0,
+      int numStreams = 4);
+
+  void run(void* devPtr, size_t bytes);
+
+ private:
+  std::shared_ptr<Context> ctx_;
+  int device_;
+  int root_;
+  std::unique_ptr<XgmiMesh> mesh_;
+  std::unique_ptr<HipStream> cs_;
+  std::vector<std::unique_ptr<HipStream>> fanout_;
+  int fBDATA_;
+  int fBACK_;
+  uint64_t seq_{0};
+};
+
+// Single-process multi-pointer allreduce: fused k-way reduction into
+// ptrs[0] then broadcast copies (all on one device/stream; blocking).
+void hipAllreduceLocal(
+    const std::vector<void*>& ptrs,
+    size_t elements,
+    DType dtype,
+    ReduceOp op,
+    int device);
+
+} // namespace hip
+} // namespace glooamd

@@ -1,0 +1,469 @@
+// CDNA4 (gfx950) elementwise reduction kernels.
+//
+// Re-designed replacements for the reference's CUDA kernels
+// (gloo/cuda.cu:283-409). MI355X-first choices:
+//  * memory-bound roofline: every kernel moves 16 B per lane per
+//    iteration (alignas(16) packs -> global_load/store_dwordx4), the
+//    coalescing sweet spot on CDNA4 (guide §2/G13: scalar 16-bit loads
+//    are 2-2.5x slower).
+//  * 256-thread blocks (4 waves of 64), grid capped at 2048 workgroups
+//    with a grid-stride loop (guide G11: >=8 blocks/CU over 256 CUs).
+//  * f16/bf16 reduce through fp32 VALU (v_cvt packed) — numerically
+//    safer than native packed-math min/max NaN semantics and still
+//    >40x below the VALU roofline at HBM speed.
+//  * no warp-32 idioms anywhere; wave64 is irrelevant here because the
+//    kernels are lane-independent.
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+#include <hip/hip_runtime.h>
+
+#include "hip/kernels.h"
+
+namespace glooamd {
+namespace hip {
+
+namespace {
+
+// --- op functors (computed in Acc, stored as T) ----------------------------
+
+template <typename T>
+struct AccOf {
+  using type = T;
+};
+template <>
+struct AccOf<__half> {
+  using type = float;
+};
+template <>
+struct AccOf<__hip_bfloat16> {
+  using type = float;
+};
+
+template <typename T>
+__device__ inline typename AccOf<T>::type toAcc(T v) {
+  return static_cast<typename AccOf<T>::type>(v);
+}
+template <>
+__device__ inline float toAcc<__half>(__half v) {
+  return __half2float(v);
+}
+template <>
+__device__ inline float toAcc<__hip_bfloat16>(__hip_bfloat16 v) {
+  return __bfloat162float(v);
+}
+
+template <typename T>
+__device__ inline T fromAcc(typename AccOf<T>::type v) {
+  return static_cast<T>(v);
+}
+template <>
+__device__ inline __half fromAcc<__half>(float v) {
+  return __float2half(v);
+}
+template <>
+__device__ inline __hip_bfloat16 fromAcc<__hip_bfloat16>(float v) {
+  return __float2bfloat16(v);
+}
+
+struct OpSum {
+  template <typename A>
+  __device__ static A apply(A a, A b) {
+    return a + b;
+  }
+};
+struct OpProd {
+  template <typename A>
+  __device__ static A apply(A a, A b) {
+    return a * b;
+  }
+};
+struct OpMin {
+  template <typename A>
+  __device__ static A apply(A a, A b) {
+    return b < a ? b : a;
+  }
+};
+struct OpMax {
+  template <typename A>
+  __device__ static A apply(A a, A b) {
+    return a < b ? b : a;
+  }
+};
+
+// --- 16-byte pack ----------------------------------------------------------
+
+template <typename T>
+struct VecOf {
+  static constexpr int value = 16 / sizeof(T);
+};
+
+template <typename T, int V = VecOf<T>::value>
+struct alignas(16) Pack {
+  T v[V];
+};
+
+// --- kernels ---------------------------------------------------------------
+
+constexpr int kBlock = 256;
+constexpr int kMaxGrid = 2048; // 256 CU x 8 workgroups
+
+template <typename T, typename OP>
+__global__ __launch_bounds__(kBlock) void reduce2Kernel(
+    T* __restrict__ dst,
+    const T* __restrict__ a,
+    const T* __restrict__ b,
+    size_t npacks,
+    size_t ntail) {
+  constexpr int V = VecOf<T>::value;
+  using P = Pack<T>;
+  const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = gridDim.x * blockDim.x;
+  auto* dp = reinterpret_cast<P*>(dst);
+  auto* ap = reinterpret_cast<const P*>(a);
+  auto* bp = reinterpret_cast<const P*>(b);
+  for (size_t i = tid; i < npacks; i += stride) {
+    P pa = ap[i];
+    P pb = bp[i];
+    P pd;
+#pragma unroll
+    for (int j = 0; j < V; j++) {
+      pd.v[j] = fromAcc<T>(OP::apply(toAcc(pa.v[j]), toAcc(pb.v[j])));
+    }
+    dp[i] = pd;
+  }
+  // tail elements past the packed region
+  const size_t base = npacks * V;
+  for (size_t i = tid; i < ntail; i += stride) {
+    dst[base + i] =
+        fromAcc<T>(OP::apply(toAcc(a[base + i]), toAcc(b[base + i])));
+  }
+}
+
+// Unaligned fallback: scalar grid-stride.
+template <typename T, typename OP>
+__global__ __launch_bounds__(kBlock) void reduce2ScalarKernel(
+    T* __restrict__ dst,
+    const T* __restrict__ a,
+    const T* __restrict__ b,
+    size_t n) {
+  const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = gridDim.x * blockDim.x;
+  for (size_t i = tid; i < n; i += stride) {
+    dst[i] = fromAcc<T>(OP::apply(toAcc(a[i]), toAcc(b[i])));
+  }
+}
+
+template <typename T, typename OP, int K>
+__global__ __launch_bounds__(kBlock) void reduceNKernel(
+    T* __restrict__ dst,
+    const T* __restrict__ s0,
+    const T* __restrict__ s1,
+    const T* __restrict__ s2,
+    const T* __restrict__ s3,
+    const T* __restrict__ s4,
+    const T* __restrict__ s5,
+    const T* __restrict__ s6,
+    const T* __restrict__ s7,
+    size_t npacks,
+    size_t ntail) {
+  constexpr int V = VecOf<T>::value;
+  using P = Pack<T>;
+  using A = typename AccOf<T>::type;
+  const T* srcs[8] = {s0, s1, s2, s3, s4, s5, s6, s7};
+  const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = gridDim.x * blockDim.x;
+  auto* dp = reinterpret_cast<P*>(dst);
+  for (size_t i = tid; i < npacks; i += stride) {
+    P acc = reinterpret_cast<const P*>(srcs[0])[i];
+    A accv[V];
+#pragma unroll
+    for (int j = 0; j < V; j++) {
+      accv[j] = toAcc(acc.v[j]);
+    }
+#pragma unroll
+    for (int k = 1; k < K; k++) {
+      P pk = reinterpret_cast<const P*>(srcs[k])[i];
+#pragma unroll
+      for (int j = 0; j < V; j++) {
+        accv[j] = OP::apply(accv[j], toAcc(pk.v[j]));
+      }
+    }
+    P out;
+#pragma unroll
+    for (int j = 0; j < V; j++) {
+      out.v[j] = fromAcc<T>(accv[j]);
+    }
+    dp[i] = out;
+  }
+  const size_t base = npacks * V;
+  for (size_t i = tid; i < ntail; i += stride) {
+    A acc = toAcc(srcs[0][base + i]);
+#pragma unroll
+    for (int k = 1; k < K; k++) {
+      acc = OP::apply(acc, toAcc(srcs[k][base + i]));
+    }
+    dst[base + i] = fromAcc<T>(acc);
+  }
+}
+
+__global__ __launch_bounds__(64) void writeFlagKernel(
+    uint64_t* addr,
+    uint64_t val) {
+  if (threadIdx.x == 0) {
+    __hip_atomic_store(addr, val, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_SYSTEM);
+  }
+}
+
+__global__ __launch_bounds__(64) void waitFlagGteKernel(
+    const uint64_t* addr,
+    uint64_t val) {
+  if (threadIdx.x == 0) {
+    while (__hip_atomic_load(addr, __ATOMIC_ACQUIRE,
+                             __HIP_MEMORY_SCOPE_SYSTEM) < val) {
+      __builtin_amdgcn_s_sleep(16);
+    }
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(kBlock) void fillPatternKernel(
+    T* __restrict__ ptr,
+    size_t n,
+    double val,
+    double stride_) {
+  const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = gridDim.x * blockDim.x;
+  for (size_t i = tid; i < n; i += stride) {
+    ptr[i] = fromAcc<T>(static_cast<typename AccOf<T>::type>(
+        double(i % 47) * stride_ + val));
+  }
+}
+
+inline int gridFor(size_t work) {
+  if (work == 0) {
+    return 1;
+  }
+  size_t blocks = (work + kBlock - 1) / kBlock;
+  return static_cast<int>(blocks < kMaxGrid ? blocks : kMaxGrid);
+}
+
+template <typename T, typename OP>
+void launchReduce2T(
+    void* dst,
+    const void* a,
+    const void* b,
+    size_t n,
+    hipStream_t stream) {
+  constexpr int V = VecOf<T>::value;
+  const auto aligned = [](const void* p) {
+    return (reinterpret_cast<uintptr_t>(p) & 15) == 0;
+  };
+  if (aligned(dst) && aligned(a) && aligned(b)) {
+    size_t npacks = n / V;
+    size_t ntail = n % V;
+    hipLaunchKernelGGL(
+        (reduce2Kernel<T, OP>),
+        dim3(gridFor(npacks + ntail)),
+        dim3(kBlock),
+        0,
+        stream,
+        static_cast<T*>(dst),
+        static_cast<const T*>(a),
+        static_cast<const T*>(b),
+        npacks,
+        ntail);
+  } else {
+    hipLaunchKernelGGL(
+        (reduce2ScalarKernel<T, OP>),
+        dim3(gridFor(n)),
+        dim3(kBlock),
+        0,
+        stream,
+        static_cast<T*>(dst),
+        static_cast<const T*>(a),
+        static_cast<const T*>(b),
+        n);
+  }
+}
+
+template <typename T>
+void launchReduce2Op(
+    void* dst,
+    const void* a,
+    const void* b,
+    size_t n,
+    ReduceOp op,
+    hipStream_t stream) {
+  switch (op) {
+    case ReduceOp::SUM:
+      return launchReduce2T<T, OpSum>(dst, a, b, n, stream);
+    case ReduceOp::PRODUCT:
+      return launchReduce2T<T, OpProd>(dst, a, b, n, stream);
+    case ReduceOp::MIN:
+      return launchReduce2T<T, OpMin>(dst, a, b, n, stream);
+    case ReduceOp::MAX:
+      return launchReduce2T<T, OpMax>(dst, a, b, n, stream);
+  }
+}
+
+template <typename T, typename OP>
+void launchReduceNT(
+    void* dst,
+    const void* const* srcs,
+    int k,
+    size_t n,
+    hipStream_t stream) {
+  constexpr int V = VecOf<T>::value;
+  const T* s[8] = {nullptr};
+  bool aligned = (reinterpret_cast<uintptr_t>(dst) & 15) == 0;
+  for (int i = 0; i < k; i++) {
+    s[i] = static_cast<const T*>(srcs[i]);
+    aligned = aligned && (reinterpret_cast<uintptr_t>(s[i]) & 15) == 0;
+  }
+  for (int i = k; i < 8; i++) {
+    s[i] = s[0];
+  }
+  size_t npacks = aligned ? n / V : 0;
+  size_t ntail = n - npacks * V;
+  auto launch = [&](auto kval) {
+    constexpr int K = decltype(kval)::value;
+    hipLaunchKernelGGL(
+        (reduceNKernel<T, OP, K>),
+        dim3(gridFor(npacks + ntail)),
+        dim3(kBlock),
+        0,
+        stream,
+        static_cast<T*>(dst),
+        s[0], s[1], s[2], s[3], s[4], s[5], s[6], s[7],
+        npacks,
+        ntail);
+  };
+  switch (k) {
+    case 2:
+      return launch(std::integral_constant<int, 2>{});
+    case 3:
+      return launch(std::integral_constant<int, 3>{});
+    case 4:
+      return launch(std::integral_constant<int, 4>{});
+    case 5:
+      return launch(std::integral_constant<int, 5>{});
+    case 6:
+      return launch(std::integral_constant<int, 6>{});
+    case 7:
+      return launch(std::integral_constant<int, 7>{});
+    case 8:
+      return launch(std::integral_constant<int, 8>{});
+    default:
+      return; // k==1 handled by caller (plain copy)
+  }
+}
+
+template <typename T>
+void launchReduceNOp(
+    void* dst,
+    const void* const* srcs,
+    int k,
+    size_t n,
+    ReduceOp op,
+    hipStream_t stream) {
+  switch (op) {
+    case ReduceOp::SUM:
+      return launchReduceNT<T, OpSum>(dst, srcs, k, n, stream);
+    case ReduceOp::PRODUCT:
+      return launchReduceNT<T, OpProd>(dst, srcs, k, n, stream);
+    case ReduceOp::MIN:
+      return launchReduceNT<T, OpMin>(dst, srcs, k, n, stream);
+    case ReduceOp::MAX:
+      return launchReduceNT<T, OpMax>(dst, srcs, k, n, stream);
+  }
+}
+
+template <typename F>
+auto dispatchDType(DType d, F&& f) {
+  switch (d) {
+    case DType::F32:
+      return f(float{});
+    case DType::F64:
+      return f(double{});
+    case DType::F16:
+      return f(__half{});
+    case DType::BF16:
+      return f(__hip_bfloat16{});
+    case DType::I8:
+      return f(int8_t{});
+    case DType::U8:
+      return f(uint8_t{});
+    case DType::I32:
+      return f(int32_t{});
+    case DType::I64:
+      return f(int64_t{});
+    case DType::U64:
+      return f(uint64_t{});
+  }
+  __builtin_unreachable();
+}
+
+} // namespace
+
+void launchReduce2(
+    void* dst,
+    const void* a,
+    const void* b,
+    size_t n,
+    DType dtype,
+    ReduceOp op,
+    hipStream_t stream) {
+  dispatchDType(dtype, [&](auto t) {
+    launchReduce2Op<decltype(t)>(dst, a, b, n, op, stream);
+  });
+}
+
+void launchReduceN(
+    void* dst,
+    const void* const* srcs,
+    int k,
+    size_t n,
+    DType dtype,
+    ReduceOp op,
+    hipStream_t stream) {
+  dispatchDType(dtype, [&](auto t) {
+    launchReduceNOp<decltype(t)>(dst, srcs, k, n, op, stream);
+  });
+}
+
+void launchWriteFlag(uint64_t* addr, uint64_t val, hipStream_t stream) {
+  hipLaunchKernelGGL(writeFlagKernel, dim3(1), dim3(64), 0, stream, addr, val);
+}
+
+void launchWaitFlagGte(
+    const uint64_t* addr,
+    uint64_t val,
+    hipStream_t stream) {
+  hipLaunchKernelGGL(
+      waitFlagGteKernel, dim3(1), dim3(64), 0, stream, addr, val);
+}
+
+void launchFillPattern(
+    void* ptr,
+    size_t n,
+    DType dtype,
+    double val,
+    double stride,
+    hipStream_t stream) {
+  dispatchDType(dtype, [&](auto t) {
+    using T = decltype(t);
+    hipLaunchKernelGGL(
+        (fillPatternKernel<T>),
+        dim3(gridFor(n)),
+        dim3(kBlock),
+        0,
+        stream,
+        static_cast<T*>(ptr),
+        n,
+        val,
+        stride);
+  });
+}
+
+} // namespace hip
+} // namespace glooamd

@@ -1,0 +1,181 @@
+#include "hip/mesh.h"
+
+#include <unistd.h>
+
+#include <cstring>
+
+#include "collectives/collectives.h"
+#include "common/logging.h"
+
+namespace glooamd {
+namespace hip {
+
+namespace {
+// hipExtMallocWithFlags with fine-grained coherence for the flag page:
+// cross-device system-scope atomics over xGMI require fine-grained
+// allocations on CDNA.
+void* allocFineGrained(size_t bytes) {
+  void* p = nullptr;
+  hipError_t err =
+      hipExtMallocWithFlags(&p, bytes, hipDeviceMallocFinegrained);
+  if (err != hipSuccess || p == nullptr) {
+    // Fallback: pinned host memory is always fine-grained & peer-visible
+    // (slower polls, still correct).
+    GA_WARN << "fine-grained device alloc failed ("
+            << hipGetErrorString(err) << "); flags fall back to hipMalloc";
+    GA_HIP_CHECK(hipMalloc(&p, bytes));
+  }
+  return p;
+}
+
+struct HandleBlob {
+  int32_t pid;
+  int32_t device;
+  uint64_t dataPtr;
+  uint64_t flagsPtr;
+  hipIpcMemHandle_t dataHandle;
+  hipIpcMemHandle_t flagsHandle;
+};
+} // namespace
+
+XgmiMesh::XgmiMesh(
+    std::shared_ptr<Context> ctx,
+    int device,
+    size_t workCap,
+    size_t inboxCap)
+    : ctx_(std::move(ctx)),
+      device_(device),
+      workCap_(workCap),
+      inboxCap_(inboxCap) {
+  GA_HIP_CHECK(hipSetDevice(device_));
+  const size_t total = workCap_ + 2 * inboxCap_;
+  {
+    std::lock_guard<std::mutex> lock(allocMutex());
+    GA_HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&data_), total));
+  }
+  flags_ = static_cast<uint64_t*>(allocFineGrained(kNumFlags * 8));
+  GA_HIP_CHECK(hipMemset(flags_, 0, kNumFlags * 8));
+  GA_HIP_CHECK(hipDeviceSynchronize());
+  exchange();
+}
+
+void XgmiMesh::exchange() {
+  const int P = ctx_->size;
+  peerData_.assign(P, nullptr);
+  peerFlags_.assign(P, nullptr);
+  peerSameProcess_.assign(P, false);
+
+  HandleBlob mine;
+  std::memset(&mine, 0, sizeof(mine));
+  mine.pid = getpid();
+  mine.device = device_;
+  mine.dataPtr = reinterpret_cast<uint64_t>(data_);
+  mine.flagsPtr = reinterpret_cast<uint64_t>(flags_);
+  if (P > 1) {
+    // Exporting handles only matters with real peers; it also fails for
+    // some allocation types, so only do it when needed.
+    GA_HIP_CHECK(hipIpcGetMemHandle(&mine.dataHandle, data_));
+    GA_HIP_CHECK(hipIpcGetMemHandle(&mine.flagsHandle, flags_));
+  }
+
+  std::vector<HandleBlob> all(P);
+  AllgatherOptions opts(ctx_);
+  opts.input = &mine;
+  opts.output = all.data();
+  opts.inElements = sizeof(HandleBlob);
+  opts.elementSize = 1;
+  opts.tag = ctx_->nextSlot();
+  allgather(opts);
+
+  for (int r = 0; r < P; r++) {
+    if (r == ctx_->rank) {
+      peerData_[r] = data_;
+      peerFlags_[r] = flags_;
+      peerSameProcess_[r] = true;
+      continue;
+    }
+    if (all[r].pid == mine.pid) {
+      // Thread-spawned rank in this process: IPC open of a same-process
+      // handle is invalid; the raw pointers are directly usable.
+      peerData_[r] = reinterpret_cast<char*>(all[r].dataPtr);
+      peerFlags_[r] = reinterpret_cast<uint64_t*>(all[r].flagsPtr);
+      peerSameProcess_[r] = true;
+    } else {
+      void* p = nullptr;
+      GA_HIP_CHECK(hipIpcOpenMemHandle(
+          &p, all[r].dataHandle, hipIpcMemLazyEnablePeerAccess));
+      peerData_[r] = static_cast<char*>(p);
+      void* f = nullptr;
+      GA_HIP_CHECK(hipIpcOpenMemHandle(
+          &f, all[r].flagsHandle, hipIpcMemLazyEnablePeerAccess));
+      peerFlags_[r] = static_cast<uint64_t*>(f);
+    }
+  }
+
+  // Everyone has mapped everyone before first use.
+  BarrierOptions bar(ctx_);
+  bar.tag = ctx_->nextSlot();
+  barrier(bar);
+}
+
+void XgmiMesh::releasePeers() {
+  for (int r = 0; r < static_cast<int>(peerData_.size()); r++) {
+    if (r == ctx_->rank || peerSameProcess_[r]) {
+      continue;
+    }
+    if (peerData_[r] != nullptr) {
+      (void)hipIpcCloseMemHandle(peerData_[r]);
+    }
+    if (peerFlags_[r] != nullptr) {
+      (void)hipIpcCloseMemHandle(peerFlags_[r]);
+    }
+  }
+  peerData_.clear();
+  peerFlags_.clear();
+}
+
+void XgmiMesh::ensureCapacity(size_t workCap, size_t inboxCap) {
+  if (workCap <= workCap_ && inboxCap <= inboxCap_) {
+    // Still a collective decision: all ranks compute the same sizes from
+    // the same collective arguments, so either all grow or none do.
+    return;
+  }
+  workCap_ = std::max(workCap_, workCap);
+  inboxCap_ = std::max(inboxCap_, inboxCap);
+  releasePeers();
+  {
+    std::lock_guard<std::mutex> lock(allocMutex());
+    (void)hipFree(data_);
+    GA_HIP_CHECK(hipMalloc(
+        reinterpret_cast<void**>(&data_), workCap_ + 2 * inboxCap_));
+  }
+  exchange();
+}
+
+int XgmiMesh::allocFlags(int count) {
+  GA_ENFORCE_LE(nextFlag_ + count, kNumFlags, "out of doorbell flags");
+  int base = nextFlag_;
+  nextFlag_ += count;
+  return base;
+}
+
+void XgmiMesh::poisonFlags() {
+  std::vector<uint64_t> poison(kNumFlags, ~uint64_t(0) >> 1);
+  (void)hipMemcpy(flags_, poison.data(), kNumFlags * 8,
+                  hipMemcpyHostToDevice);
+}
+
+XgmiMesh::~XgmiMesh() {
+  (void)hipDeviceSynchronize();
+  releasePeers();
+  if (data_ != nullptr) {
+    std::lock_guard<std::mutex> lock(allocMutex());
+    (void)hipFree(data_);
+  }
+  if (flags_ != nullptr) {
+    (void)hipFree(flags_);
+  }
+}
+
+} // namespace hip
+} // namespace glooamd
