@@ -17,6 +17,7 @@
 #include <hip/hip_fp16.h>
 #include <hip/hip_runtime.h>
 
+#include "hip/core.h"
 #include "hip/kernels.h"
 
 namespace glooamd {
@@ -272,6 +273,7 @@ void launchReduce2T(
         static_cast<const T*>(b),
         npacks,
         ntail);
+  GA_HIP_CHECK(hipGetLastError());
   } else {
     hipLaunchKernelGGL(
         (reduce2ScalarKernel<T, OP>),
@@ -283,6 +285,7 @@ void launchReduce2T(
         static_cast<const T*>(a),
         static_cast<const T*>(b),
         n);
+  GA_HIP_CHECK(hipGetLastError());
   }
 }
 
@@ -337,6 +340,7 @@ void launchReduceNT(
         s[0], s[1], s[2], s[3], s[4], s[5], s[6], s[7],
         npacks,
         ntail);
+  GA_HIP_CHECK(hipGetLastError());
   };
   switch (k) {
     case 2:
@@ -433,6 +437,7 @@ void launchReduceN(
 
 void launchWriteFlag(uint64_t* addr, uint64_t val, hipStream_t stream) {
   hipLaunchKernelGGL(writeFlagKernel, dim3(1), dim3(64), 0, stream, addr, val);
+  GA_HIP_CHECK(hipGetLastError());
 }
 
 void launchWaitFlagGte(
@@ -441,6 +446,7 @@ void launchWaitFlagGte(
     hipStream_t stream) {
   hipLaunchKernelGGL(
       waitFlagGteKernel, dim3(1), dim3(64), 0, stream, addr, val);
+  GA_HIP_CHECK(hipGetLastError());
 }
 
 void launchFillPattern(
@@ -462,6 +468,7 @@ void launchFillPattern(
         n,
         val,
         stride);
+  GA_HIP_CHECK(hipGetLastError());
   });
 }
 
