@@ -6,6 +6,16 @@ CPU collectives over an epoll TCP transport, and HIP/CDNA4 device
 collectives over xGMI (hipIpcMemHandle peer transport) for GPU tensors.
 """
 
+# Load torch (and its bundled HIP runtime) BEFORE our extension: _C's
+# libamdhip64.so.7 dependency then resolves by SONAME onto torch's
+# already-loaded runtime, so one process has exactly one HIP runtime.
+# (The system ROCm userspace at /opt/rocm can fail to enumerate the GPU
+# on some hosts while torch's bundled runtime works.)
+try:
+    import torch  # noqa: F401
+except ImportError:
+    pass
+
 from gloo_amd._C import (  # noqa: F401
     Buffer,
     Context,
