@@ -71,6 +71,8 @@ def main():
     esize = ga.dtype_size(gdt)
 
     if have_gpu:
+        # one GPU per rank; wrap on boxes with fewer GPUs than ranks
+        local_rank = local_rank % torch.cuda.device_count()
         torch.cuda.set_device(local_rank)
         tdt = getattr(torch, tmap[args.dtype])
         buf = torch.rand(args.elements, dtype=torch.float32,

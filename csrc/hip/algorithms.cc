@@ -507,7 +507,13 @@ void hipAllreduceLocal(
     return;
   }
   GA_HIP_CHECK(hipSetDevice(device));
-  HipStream s(device);
+  // Cached per-thread stream: creation costs ~300us, far more than the
+  // kernels themselves for MB-scale buffers.
+  static thread_local std::unique_ptr<HipStream> cached;
+  if (!cached || cached->device() != device) {
+    cached = std::make_unique<HipStream>(device);
+  }
+  HipStream& s = *cached;
   const size_t es = dtypeSize(dtype);
   launchReduceN(
       ptrs[0],
