@@ -86,7 +86,36 @@ PYBIND11_MODULE(_C, m) {
   m.def("dtype_size", &dtypeSize);
 
   // --- stores ---------------------------------------------------------------
-  py::class_<IStore, std::shared_ptr<IStore>>(m, "Store")
+  // Trampoline so Python classes (e.g. a torch.distributed Store adapter)
+  // can implement the rendezvous interface consumed by connectFullMesh.
+  class PyStore : public IStore {
+   public:
+    using IStore::IStore;
+    void set(const std::string& key, const std::vector<char>& data) override {
+      py::gil_scoped_acquire gil;
+      py::function f = py::get_override(this, "set");
+      GA_ENFORCE(f, "Store.set not implemented");
+      f(key, py::bytes(data.data(), data.size()));
+    }
+    std::vector<char> get(const std::string& key) override {
+      py::gil_scoped_acquire gil;
+      py::function f = py::get_override(this, "get");
+      GA_ENFORCE(f, "Store.get not implemented");
+      std::string s = f(key).cast<std::string>();
+      return std::vector<char>(s.begin(), s.end());
+    }
+    void wait(
+        const std::vector<std::string>& keys,
+        const std::chrono::milliseconds& timeout) override {
+      py::gil_scoped_acquire gil;
+      py::function f = py::get_override(this, "wait");
+      GA_ENFORCE(f, "Store.wait not implemented");
+      f(keys, timeout.count());
+    }
+  };
+
+  py::class_<IStore, PyStore, std::shared_ptr<IStore>>(m, "Store")
+      .def(py::init<>())
       .def(
           "set",
           [](IStore& s, const std::string& key, py::bytes data) {

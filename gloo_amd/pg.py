@@ -1,0 +1,367 @@
+"""torch.distributed ProcessGroup backend backed by gloo_amd.
+
+Register/usage:
+    import gloo_amd.pg  # registers backend "glooamd" (cpu + cuda)
+    torch.distributed.init_process_group("glooamd", ...)
+
+CPU tensors run the v2 TCP collectives; CUDA tensors route allreduce /
+broadcast through the xGMI/IPC device algorithms (csrc/hip/) and the
+remaining collectives through pinned-host staging (device-native
+versions are per-op upgrades, not API changes).
+
+This is the MI355X counterpart of the reference's role as the `gloo`
+backend of torch.distributed.
+"""
+import threading
+from datetime import timedelta
+
+import torch
+import torch.distributed as dist
+from torch._C._distributed_c10d import _create_work_from_future
+from torch.futures import Future
+
+import gloo_amd as ga
+
+def _map_op(op):
+    """c10d passes ReduceOp instances whose hash differs from the enum
+    members; compare by equality."""
+    if op == dist.ReduceOp.SUM or op == dist.ReduceOp.AVG:
+        return ga.ReduceOp.sum  # AVG divides after
+    if op == dist.ReduceOp.PRODUCT:
+        return ga.ReduceOp.product
+    if op == dist.ReduceOp.MIN:
+        return ga.ReduceOp.min
+    if op == dist.ReduceOp.MAX:
+        return ga.ReduceOp.max
+    raise ValueError(f"unsupported reduce op {op}")
+
+
+class _OPSView:
+    def __getitem__(self, op):
+        return _map_op(op)
+
+    def get(self, op):
+        try:
+            return _map_op(op)
+        except ValueError:
+            return None
+
+
+_OPS = _OPSView()
+
+
+def _ret_work(result=None):
+    fut = Future()
+    fut.set_result(result)
+    return _create_work_from_future(fut)
+
+
+class _TorchStoreAdapter(ga.Store):
+    """Bridges a c10d Store into the gloo_amd rendezvous interface."""
+
+    def __init__(self, tstore):
+        super().__init__()
+        self._s = tstore
+
+    def set(self, key, data):
+        self._s.set(key, data)
+
+    def get(self, key):
+        return bytes(self._s.get(key))
+
+    def wait(self, keys, timeout_ms):
+        self._s.wait(list(keys))
+
+
+def _gdtype(t):
+    if t.dtype == torch.bool:
+        return ga.DType.u8
+    return ga.dtype_from_torch(t.dtype)
+
+
+class ProcessGroupGlooAmd(dist.ProcessGroup):
+    def __init__(self, store, rank, size, timeout=timedelta(seconds=300)):
+        super().__init__(rank, size)
+        if isinstance(store, ga.Store):
+            self._store = store
+        else:
+            self._store = _TorchStoreAdapter(store)
+        self._ctx = ga.Context(rank, size)
+        self._ctx.set_timeout(int(timeout.total_seconds() * 1000))
+        self._ctx.connect_full_mesh(self._store, ga.create_tcp_device())
+        self._lock = threading.Lock()
+        self._hip_ring = {}  # device -> HipAllreduceRing
+        self._hip_bcast = {}  # (device, root) -> HipBroadcastOneToAll
+        self._pin = {}  # device -> pinned staging tensor
+
+    # -- helpers -------------------------------------------------------------
+
+    def _tag(self):
+        return self._ctx.next_slot()
+
+    def _ring(self, device):
+        if device not in self._hip_ring:
+            self._hip_ring[device] = ga._C.HipAllreduceRing(self._ctx, device)
+        return self._hip_ring[device]
+
+    def _bcaster(self, device, root):
+        key = (device, root)
+        if key not in self._hip_bcast:
+            self._hip_bcast[key] = ga._C.HipBroadcastOneToAll(
+                self._ctx, device, root)
+        return self._hip_bcast[key]
+
+    def _staged(self, t, fn):
+        """Run a CPU collective on a host copy of a CUDA tensor."""
+        host = t.detach().cpu()
+        fn(host)
+        t.detach().copy_(host)
+
+    # -- collectives ---------------------------------------------------------
+
+    def allreduce(self, tensors, opts=None):
+        op = opts.reduceOp if opts is not None else dist.ReduceOp.SUM
+        gop = _OPS.get(op)
+        if gop is None:
+            raise ValueError(f"unsupported reduce op {op}")
+        with self._lock:
+            for t in tensors:
+                t_ = t.detach()
+                assert t_.is_contiguous(), "gloo_amd needs contiguous tensors"
+                if t_.is_cuda:
+                    self._ring(t_.get_device()).run(
+                        t_.data_ptr(), t_.numel(), _gdtype(t_), gop)
+                else:
+                    ga.allreduce(
+                        self._ctx, [t_.data_ptr()], t_.numel(), _gdtype(t_),
+                        gop, tag=self._tag())
+                if op == dist.ReduceOp.AVG:
+                    t_.div_(self.size())
+        return _ret_work(tensors)
+
+    def broadcast(self, tensors, opts=None):
+        root = opts.rootRank if opts is not None else 0
+        with self._lock:
+            for t in tensors:
+                t_ = t.detach()
+                assert t_.is_contiguous()
+                if t_.is_cuda:
+                    self._bcaster(t_.get_device(), root).run(
+                        t_.data_ptr(), t_.numel() * t_.element_size())
+                else:
+                    ga.broadcast(
+                        self._ctx, t_.data_ptr(), 0, t_.numel(), _gdtype(t_),
+                        root=root, tag=self._tag())
+        return _ret_work(tensors)
+
+    def _allgather_base(self, output, input, opts=None):
+        out = output.detach()
+        inp = input.detach()
+        assert out.is_contiguous() and inp.is_contiguous()
+        with self._lock:
+            tag = self._tag()
+            if out.is_cuda:
+                h_out = out.cpu()
+                h_in = inp.cpu()
+                ga.allgather(self._ctx, h_out.data_ptr(), h_in.data_ptr(),
+                             h_in.numel(), _gdtype(inp), tag=tag)
+                out.copy_(h_out)
+            else:
+                ga.allgather(self._ctx, out.data_ptr(), inp.data_ptr(),
+                             inp.numel(), _gdtype(inp), tag=tag)
+        return _ret_work(output)
+
+    def allgather(self, output_tensors, input_tensors, opts=None):
+        for out_list, inp in zip(output_tensors, input_tensors):
+            flat = torch.empty(
+                inp.numel() * self.size(), dtype=inp.dtype,
+                device="cpu")
+            inp_c = inp.detach().contiguous()
+            with self._lock:
+                h_in = inp_c.cpu() if inp_c.is_cuda else inp_c
+                ga.allgather(self._ctx, flat.data_ptr(), h_in.data_ptr(),
+                             h_in.numel(), _gdtype(inp), tag=self._tag())
+            for r, out in enumerate(out_list):
+                out.detach().copy_(
+                    flat[r * inp.numel():(r + 1) * inp.numel()].view_as(out))
+        return _ret_work(output_tensors)
+
+    def reduce(self, tensors, opts=None):
+        root = opts.rootRank if opts is not None else 0
+        op = _OPS[opts.reduceOp] if opts is not None else ga.ReduceOp.sum
+        for t in tensors:
+            t_ = t.detach()
+
+            def run(host):
+                ga.reduce(self._ctx, host.data_ptr(), host.data_ptr(),
+                          host.numel(), _gdtype(t_), op, root=root,
+                          tag=self._tag())
+
+            with self._lock:
+                if t_.is_cuda:
+                    self._staged(t_, run)
+                else:
+                    run(t_)
+            if opts is not None and opts.reduceOp == dist.ReduceOp.AVG \
+                    and self.rank() == root:
+                t_.div_(self.size())
+        return _ret_work(tensors)
+
+    def _reduce_scatter_base(self, output, input, opts=None):
+        op = _OPS[opts.reduceOp] if opts is not None else ga.ReduceOp.sum
+        out = output.detach()
+        inp = input.detach().contiguous()
+        with self._lock:
+            tag = self._tag()
+            if out.is_cuda:
+                h_out = out.cpu()
+                h_in = inp.cpu()
+                ga.reduce_scatter(self._ctx, h_out.data_ptr(), h_in.data_ptr(),
+                                  h_out.numel(), _gdtype(out), op, tag=tag)
+                out.copy_(h_out)
+            else:
+                ga.reduce_scatter(self._ctx, out.data_ptr(), inp.data_ptr(),
+                                  out.numel(), _gdtype(out), op, tag=tag)
+        if opts is not None and opts.reduceOp == dist.ReduceOp.AVG:
+            out.div_(self.size())
+        return _ret_work(output)
+
+    def reduce_scatter(self, output_tensors, input_tensor_lists, opts=None):
+        for out, in_list in zip(output_tensors, input_tensor_lists):
+            flat = torch.cat([t.detach().reshape(-1).cpu() for t in in_list])
+            h_out = torch.empty(out.numel(), dtype=out.dtype)
+            op = _OPS[opts.reduceOp] if opts is not None else ga.ReduceOp.sum
+            with self._lock:
+                ga.reduce_scatter(self._ctx, h_out.data_ptr(), flat.data_ptr(),
+                                  h_out.numel(), _gdtype(out), op,
+                                  tag=self._tag())
+            out.detach().copy_(h_out.view_as(out))
+            if opts is not None and opts.reduceOp == dist.ReduceOp.AVG:
+                out.detach().div_(self.size())
+        return _ret_work(output_tensors)
+
+    def alltoall_base(self, output, input, output_split_sizes,
+                      input_split_sizes, opts=None):
+        out = output.detach()
+        inp = input.detach().contiguous()
+        h_out = out.cpu() if out.is_cuda else out
+        h_in = inp.cpu() if inp.is_cuda else inp
+        with self._lock:
+            tag = self._tag()
+            if not output_split_sizes and not input_split_sizes:
+                per = inp.numel() // self.size()
+                row = inp[0].numel() if inp.dim() > 0 else 1
+                ga.alltoall(self._ctx, h_out.data_ptr(), h_in.data_ptr(),
+                            per, _gdtype(inp), tag=tag)
+            else:
+                row = inp.numel() // inp.shape[0] if inp.dim() > 0 else 1
+                in_counts = [int(s) * row for s in input_split_sizes]
+                out_counts = [int(s) * row for s in output_split_sizes]
+                ga.alltoallv(self._ctx, h_out.data_ptr(), h_in.data_ptr(),
+                             in_counts, out_counts, _gdtype(inp), tag=tag)
+        if out.is_cuda:
+            out.copy_(h_out)
+        return _ret_work(output)
+
+    def alltoall(self, output_tensors, input_tensors, opts=None):
+        flat_in = torch.cat(
+            [t.detach().reshape(-1).cpu() for t in input_tensors])
+        counts_in = [t.numel() for t in input_tensors]
+        counts_out = [t.numel() for t in output_tensors]
+        flat_out = torch.empty(sum(counts_out), dtype=output_tensors[0].dtype)
+        with self._lock:
+            ga.alltoallv(self._ctx, flat_out.data_ptr(), flat_in.data_ptr(),
+                         counts_in, counts_out,
+                         _gdtype(input_tensors[0]), tag=self._tag())
+        off = 0
+        for t in output_tensors:
+            t.detach().copy_(flat_out[off:off + t.numel()].view_as(t))
+            off += t.numel()
+        return _ret_work(output_tensors)
+
+    def gather(self, output_tensors, input_tensors, opts=None):
+        root = opts.rootRank if opts is not None else 0
+        inp = input_tensors[0].detach().contiguous()
+        h_in = inp.cpu() if inp.is_cuda else inp
+        if self.rank() == root:
+            outs = output_tensors[0]
+            flat = torch.empty(inp.numel() * self.size(), dtype=inp.dtype)
+            with self._lock:
+                ga.gather(self._ctx, flat.data_ptr(), h_in.data_ptr(),
+                          h_in.numel(), _gdtype(inp), root=root,
+                          tag=self._tag())
+            for r, out in enumerate(outs):
+                out.detach().copy_(
+                    flat[r * inp.numel():(r + 1) * inp.numel()].view_as(out))
+        else:
+            with self._lock:
+                ga.gather(self._ctx, 0, h_in.data_ptr(), h_in.numel(),
+                          _gdtype(inp), root=root, tag=self._tag())
+        return _ret_work(output_tensors)
+
+    def scatter(self, output_tensors, input_tensors, opts=None):
+        root = opts.rootRank if opts is not None else 0
+        out = output_tensors[0].detach()
+        h_out = torch.empty(out.numel(), dtype=out.dtype)
+        if self.rank() == root:
+            flat = torch.cat(
+                [t.detach().reshape(-1).cpu() for t in input_tensors[0]])
+            with self._lock:
+                ga.scatter(self._ctx, h_out.data_ptr(), flat.data_ptr(),
+                           out.numel(), _gdtype(out), root=root,
+                           tag=self._tag())
+        else:
+            with self._lock:
+                ga.scatter(self._ctx, h_out.data_ptr(), 0, out.numel(),
+                           _gdtype(out), root=root, tag=self._tag())
+        out.copy_(h_out.view_as(out))
+        return _ret_work(output_tensors)
+
+    def barrier(self, opts=None):
+        with self._lock:
+            ga.barrier(self._ctx, tag=self._tag())
+        return _ret_work(True)
+
+    def send(self, tensors, dst_rank, tag=0):
+        for t in tensors:
+            t_ = t.detach().contiguous()
+            h = t_.cpu() if t_.is_cuda else t_
+            ub = self._ctx.create_unbound_buffer(
+                h.data_ptr(), h.numel() * h.element_size())
+            ub.send(dst_rank, self._p2p_slot(tag))
+            ub.wait_send()
+        return _ret_work(tensors)
+
+    def recv(self, tensors, src_rank, tag=0):
+        for t in tensors:
+            t_ = t.detach()
+            h = torch.empty(
+                t_.numel(), dtype=t_.dtype) if t_.is_cuda else t_
+            ub = self._ctx.create_unbound_buffer(
+                h.data_ptr(), h.numel() * h.element_size())
+            ub.recv(src_rank, self._p2p_slot(tag))
+            ub.wait_recv()
+            if t_.is_cuda:
+                t_.copy_(h.view_as(t_))
+        return _ret_work(tensors)
+
+    @staticmethod
+    def _p2p_slot(tag):
+        # user tag band, below the context's next_slot() counter space
+        return 0x0900_0000_0000_0000 | (tag & 0xFFFFF)
+
+    def getBackendName(self):
+        return "glooamd"
+
+    @property
+    def options(self):  # some callers poke at this
+        return None
+
+
+def _create_glooamd_pg(prefix_store, rank, world_size, timeout):
+    return ProcessGroupGlooAmd(prefix_store, rank, world_size, timeout)
+
+
+if "GLOOAMD" not in getattr(dist.Backend, "_plugins", {}):
+    dist.Backend.register_backend(
+        "glooamd", _create_glooamd_pg, devices=["cpu", "cuda"])

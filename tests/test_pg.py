@@ -1,0 +1,199 @@
+"""torch.distributed ProcessGroup backend tests (CPU tensors).
+
+Direct-construction tests run world=2/4 in threads; the subprocess test
+goes through init_process_group("glooamd") + DDP (see pg_worker.py).
+"""
+import os
+import subprocess
+import sys
+import threading
+
+import pytest
+import torch
+import torch.distributed as dist
+
+import gloo_amd as ga
+from gloo_amd.pg import ProcessGroupGlooAmd
+
+
+def spawn_pg(size, fn):
+    store = ga.HashStore()
+    errors = []
+    results = [None] * size
+
+    def worker(rank):
+        try:
+            pg = ProcessGroupGlooAmd(store, rank, size)
+            results[rank] = fn(pg, rank, size)
+        except Exception:  # noqa: BLE001
+            import traceback
+
+            errors.append(traceback.format_exc())
+
+    ths = [threading.Thread(target=worker, args=(r,)) for r in range(size)]
+    [t.start() for t in ths]
+    [t.join(60) for t in ths]
+    assert not errors, errors[0]
+    return results
+
+
+def test_pg_allreduce():
+    def fn(pg, rank, size):
+        t = torch.arange(1000, dtype=torch.float32) + rank
+        pg.allreduce([t]).wait()
+        expected = sum(torch.arange(1000, dtype=torch.float32) + r
+                       for r in range(size))
+        assert torch.allclose(t, expected)
+        return True
+
+    spawn_pg(4, fn)
+
+
+def test_pg_allreduce_ops():
+    def fn(pg, rank, size):
+        opts = dist.AllreduceOptions()
+        opts.reduceOp = dist.ReduceOp.MAX
+        t = torch.tensor([float(rank), -float(rank)])
+        pg.allreduce([t], opts).wait()
+        assert torch.allclose(t, torch.tensor([float(size - 1), 0.0]))
+
+        opts.reduceOp = dist.ReduceOp.AVG
+        t = torch.full((10,), float(rank))
+        pg.allreduce([t], opts).wait()
+        assert torch.allclose(t, torch.full((10,), (size - 1) / 2))
+        return True
+
+    spawn_pg(2, fn)
+
+
+def test_pg_broadcast():
+    def fn(pg, rank, size):
+        opts = dist.BroadcastOptions()
+        opts.rootRank = 1
+        t = (torch.arange(64, dtype=torch.float64) if rank == 1
+             else torch.zeros(64, dtype=torch.float64))
+        pg.broadcast([t], opts).wait()
+        assert torch.allclose(t, torch.arange(64, dtype=torch.float64))
+        return True
+
+    spawn_pg(3, fn)
+
+
+def test_pg_allgather():
+    def fn(pg, rank, size):
+        inp = torch.full((8,), float(rank))
+        outs = [[torch.zeros(8) for _ in range(size)]]
+        pg.allgather(outs, [inp]).wait()
+        for r in range(size):
+            assert torch.allclose(outs[0][r], torch.full((8,), float(r)))
+        # _allgather_base
+        flat = torch.zeros(8 * size)
+        pg._allgather_base(flat, inp).wait()
+        for r in range(size):
+            assert torch.allclose(flat[r * 8:(r + 1) * 8],
+                                  torch.full((8,), float(r)))
+        return True
+
+    spawn_pg(3, fn)
+
+
+def test_pg_reduce_scatter_base():
+    def fn(pg, rank, size):
+        inp = torch.arange(size * 4, dtype=torch.float32) + rank
+        out = torch.zeros(4)
+        pg._reduce_scatter_base(out, inp).wait()
+        total = sum(torch.arange(size * 4, dtype=torch.float32) + r
+                    for r in range(size))
+        assert torch.allclose(out, total[rank * 4:(rank + 1) * 4])
+        return True
+
+    spawn_pg(4, fn)
+
+
+def test_pg_alltoall_base():
+    def fn(pg, rank, size):
+        inp = torch.arange(size * 3, dtype=torch.float32) + 100 * rank
+        out = torch.zeros(size * 3)
+        pg.alltoall_base(out, inp, [], []).wait()
+        for s in range(size):
+            expect = torch.arange(rank * 3, rank * 3 + 3,
+                                  dtype=torch.float32) + 100 * s
+            assert torch.allclose(out[s * 3:(s + 1) * 3], expect)
+        return True
+
+    spawn_pg(3, fn)
+
+
+def test_pg_gather_scatter():
+    def fn(pg, rank, size):
+        opts = dist.GatherOptions()
+        opts.rootRank = 0
+        inp = torch.full((5,), float(rank))
+        outs = [[torch.zeros(5) for _ in range(size)]] if rank == 0 else []
+        pg.gather(outs, [inp], opts).wait()
+        if rank == 0:
+            for r in range(size):
+                assert torch.allclose(outs[0][r], torch.full((5,), float(r)))
+
+        sopts = dist.ScatterOptions()
+        sopts.rootRank = 0
+        sout = torch.zeros(5)
+        sins = ([[torch.full((5,), float(r) + 10) for r in range(size)]]
+                if rank == 0 else [])
+        pg.scatter([sout], sins, sopts).wait()
+        assert torch.allclose(sout, torch.full((5,), float(rank) + 10))
+        return True
+
+    spawn_pg(3, fn)
+
+
+def test_pg_send_recv_barrier():
+    def fn(pg, rank, size):
+        if rank == 0:
+            t = torch.arange(32, dtype=torch.int64)
+            pg.send([t], 1, tag=5).wait()
+        elif rank == 1:
+            t = torch.zeros(32, dtype=torch.int64)
+            pg.recv([t], 0, tag=5).wait()
+            assert torch.equal(t, torch.arange(32, dtype=torch.int64))
+        pg.barrier().wait()
+        return True
+
+    spawn_pg(2, fn)
+
+
+def test_pg_reduce():
+    def fn(pg, rank, size):
+        opts = dist.ReduceOptions()
+        opts.rootRank = 1
+        opts.reduceOp = dist.ReduceOp.SUM
+        t = torch.full((100,), float(rank + 1))
+        pg.reduce([t], opts).wait()
+        if rank == 1:
+            assert torch.allclose(t, torch.full((100,), float(
+                sum(range(1, size + 1)))))
+        return True
+
+    spawn_pg(3, fn)
+
+
+def test_init_process_group_ddp():
+    """Full stack: init_process_group('glooamd') + DDP training step in
+    subprocesses."""
+    worker = os.path.join(os.path.dirname(__file__), "pg_worker.py")
+    import random
+
+    port = random.randint(20000, 40000)
+    procs = [
+        subprocess.Popen(
+            [sys.executable, worker, str(r), "2", str(port)],
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        )
+        for r in range(2)
+    ]
+    outs = []
+    for p in procs:
+        out, _ = p.communicate(timeout=180)
+        outs.append(out.decode())
+    assert all(p.returncode == 0 for p in procs), "\n".join(outs)
+    assert all("DDP-OK" in o for o in outs), "\n".join(outs)
