@@ -10,6 +10,7 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include "algorithms/factory.h"
 #include "collectives/collectives.h"
 #include "collectives/reduce_fns.h"
 #include "common/store.h"
@@ -599,6 +600,35 @@ PYBIND11_MODULE(_C, m) {
       py::arg("op") = ReduceOp::SUM,
       py::arg("tag") = 0,
       py::arg("timeout_ms") = 0);
+
+  // --- legacy Algorithm classes --------------------------------------------
+  py::class_<Algorithm>(m, "Algorithm")
+      .def("run", &Algorithm::run, py::call_guard<py::gil_scoped_release>());
+
+  m.def(
+      "create_algorithm",
+      [](const std::string& name, std::shared_ptr<Context> ctx,
+         std::vector<uintptr_t> ptrs, size_t count, DType dtype, ReduceOp op,
+         int root, std::vector<int> recvElems, size_t bytes, int steps) {
+        std::vector<void*> p;
+        for (auto v : ptrs) {
+          p.push_back(reinterpret_cast<void*>(v));
+        }
+        py::gil_scoped_release rel;
+        return createAlgorithm(
+            name, ctx, p, count, dtype, op, root, recvElems, bytes, steps);
+      },
+      py::arg("name"),
+      py::arg("context"),
+      py::arg("ptrs") = std::vector<uintptr_t>(),
+      py::arg("count") = 0,
+      py::arg("dtype") = DType::F32,
+      py::arg("op") = ReduceOp::SUM,
+      py::arg("root") = 0,
+      py::arg("recv_elems") = std::vector<int>(),
+      py::arg("bytes") = 0,
+      py::arg("steps") = 1,
+      py::keep_alive<0, 2>());
 
   // --- HIP / xGMI device collectives ---------------------------------------
   m.def("hip_available", &hip::available);

@@ -567,8 +567,22 @@ void TcpPair::finishRx() {
     rxBbuf_->recvCount_++;
     rxBbuf_ = nullptr;
   } else if (rxIsSpill_) {
-    earlyBoundData_[rxPre_.slot].emplace_back(
-        rxPre_.roffset, std::move(rxSpill_));
+    // The slot may have been registered while we were reading the
+    // payload; deliver directly in that case (otherwise the stash would
+    // never be replayed).
+    auto it = recvBufs_.find(rxPre_.slot);
+    if (it != recvBufs_.end()) {
+      TcpBuffer* buf = it->second;
+      GA_ENFORCE_LE(rxPre_.roffset + rxSpill_.size(), buf->size());
+      std::memcpy(
+          static_cast<char*>(buf->ptr()) + rxPre_.roffset,
+          rxSpill_.data(),
+          rxSpill_.size());
+      buf->recvCount_++;
+    } else {
+      earlyBoundData_[rxPre_.slot].emplace_back(
+          rxPre_.roffset, std::move(rxSpill_));
+    }
     rxSpill_ = std::string();
     rxIsSpill_ = false;
   }

@@ -30,6 +30,10 @@ def spawn_threads():
                 ctx = ga.Context(rank, size, base)
                 ctx.connect_full_mesh(store, dev)
                 results[rank] = fn(ctx, rank, size)
+                # Sync before teardown so no rank closes its pairs while a
+                # peer is still mid-collective (reference strategy:
+                # gloo/test/base_test.h:161-173).
+                ga.barrier(ctx, tag=0xFFFF0)
             except Exception as e:  # noqa: BLE001
                 import traceback
 
