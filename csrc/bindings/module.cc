@@ -46,10 +46,12 @@ PYBIND11_MODULE(_C, m) {
   m.doc() = "gloo_amd: MI355X-native collective communications";
 
   // --- exceptions -----------------------------------------------------------
-  static py::exception<EnforceNotMet> excEnforce(m, "EnforceNotMet");
-  static py::exception<TimeoutException> excTimeout(m, "TimeoutError");
-  static py::exception<IoException> excIo(m, "IoError");
   static py::exception<Exception> excBase(m, "GlooAmdError");
+  static py::exception<EnforceNotMet> excEnforce(
+      m, "EnforceNotMet", excBase.ptr());
+  static py::exception<IoException> excIo(m, "IoError", excBase.ptr());
+  static py::exception<TimeoutException> excTimeout(
+      m, "TimeoutError", excIo.ptr());
   py::register_exception_translator([](std::exception_ptr p) {
     try {
       if (p) {

@@ -32,8 +32,12 @@ def spawn_threads():
                 results[rank] = fn(ctx, rank, size)
                 # Sync before teardown so no rank closes its pairs while a
                 # peer is still mid-collective (reference strategy:
-                # gloo/test/base_test.h:161-173).
-                ga.barrier(ctx, tag=0xFFFF0)
+                # gloo/test/base_test.h:161-173). Best-effort: tests that
+                # deliberately poison the context (timeouts) skip it.
+                try:
+                    ga.barrier(ctx, tag=0xFFFF0)
+                except ga.GlooAmdError:
+                    pass
             except Exception as e:  # noqa: BLE001
                 import traceback
 
