@@ -45,3 +45,15 @@ clean:
 	rm -rf build $(TARGET) $(BENCH)
 
 .PHONY: all clean
+
+bench: $(BENCH)
+
+BENCH_OBJS := build/bench/main.o
+
+$(BENCH): $(CC_OBJS) $(HIP_OBJS) $(BENCH_OBJS)
+	@mkdir -p bin
+	$(CXX) -o $@ $^ $(LDFLAGS)
+
+build/bench/%.o: csrc/bench/%.cc
+	@mkdir -p $(dir $@)
+	$(CXX) $(CXXFLAGS) -c $< -o $@
