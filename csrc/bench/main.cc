@@ -297,7 +297,8 @@ Bench makeCpuBench(
       name != "allreduce_local";
   // Like the reference, the timed loop re-reduces the running values
   // (no per-iteration reset; verification uses fresh fixtures).
-  b.run = [algo] { algo->run(); };
+  // data/out are captured to keep the buffers alive for algo's lifetime.
+  b.run = [algo, data, out] { algo->run(); };
   int sz = ctx->size;
   if (isAllreduce && o.verify) {
     b.verify = [algo, data, reset, sz, numInputs] {
@@ -480,6 +481,12 @@ int main(int argc, char** argv) {
     for (long n : sweep) {
       runOne(o, ctx, n);
     }
+  }
+  // Final sync so no rank tears down while a peer is mid-collective.
+  {
+    BarrierOptions barOpts(ctx);
+    barOpts.tag = ctx->nextSlot();
+    barrier(barOpts);
   }
   return 0;
 }

@@ -385,6 +385,13 @@ void TcpPair::armEpollOutLocked() {
 // --- loop thread: events ----------------------------------------------------
 
 void TcpPair::handleEvents(uint32_t events) {
+  if (events & EPOLLIN) {
+    // Drain readable data FIRST: when the peer closes right after
+    // sending, EPOLLHUP arrives together with the final payload bytes;
+    // failing before reading would drop them. read()==0 inside the loop
+    // handles the eventual EOF.
+    readLoop();
+  }
   if (events & (EPOLLERR | EPOLLHUP)) {
     std::lock_guard<std::mutex> lock(ctx_->mu_);
     failLocked(std::make_exception_ptr(
@@ -398,9 +405,6 @@ void TcpPair::handleEvents(uint32_t events) {
       dev_->loop().modifyDescriptor(fd_, EPOLLIN, this);
       epollOutArmed_ = false;
     }
-  }
-  if (events & EPOLLIN) {
-    readLoop();
   }
 }
 
