@@ -65,7 +65,8 @@ struct Options {
           "  pairwise_exchange, reduce_scatter_halving_doubling,\n"
           "  new_allreduce_ring, new_allreduce_bcube, sendrecv_roundtrip,\n"
           "  hip_allreduce_ring, hip_allreduce_ring_chunked,\n"
-          "  hip_allreduce_halving_doubling, hip_broadcast_one_to_all\n");
+          "  hip_allreduce_halving_doubling, hip_broadcast_one_to_all,\n"
+          "  hip_allgather_ring, hip_reduce_scatter, hip_alltoall\n");
   exit(1);
 }
 
@@ -347,6 +348,60 @@ Bench makeHipBench(
     b.run = [algo, devPtr, elements, es] {
       algo->run(devPtr, elements * es);
     };
+  } else if (name == "hip_allgather_ring") {
+    void* outPtr = nullptr;
+    GA_HIP_CHECK(hipMalloc(
+        &outPtr, std::max<size_t>(elements * es * ctx->size, 16)));
+    auto algo = std::make_shared<hip::HipAllgatherRing>(ctx, device);
+    b.run = [algo, devPtr, outPtr, elements, es] {
+      algo->run(devPtr, outPtr, elements, es);
+    };
+    b.teardown = [outPtr, devPtr] {
+      (void)hipFree(outPtr);
+      (void)hipFree(devPtr);
+    };
+    return b;
+  } else if (name == "hip_reduce_scatter") {
+    // devPtr holds size*elements; out holds elements
+    void* outPtr = nullptr;
+    GA_HIP_CHECK(hipFree(devPtr));
+    GA_HIP_CHECK(hipMalloc(
+        &devPtr,
+        std::max<size_t>(elements * es * ctx->size, 16)));
+    hip::launchFillPattern(devPtr, elements * ctx->size, dtype, ctx->rank,
+                           1.0, nullptr);
+    GA_HIP_CHECK(hipMalloc(&outPtr, std::max<size_t>(elements * es, 16)));
+    GA_HIP_CHECK(hipDeviceSynchronize());
+    auto algo = std::make_shared<hip::HipReduceScatterRing>(ctx, device);
+    DType dt = dtype;
+    b.run = [algo, devPtr, outPtr, elements, dt] {
+      algo->run(devPtr, outPtr, elements, dt, ReduceOp::SUM);
+    };
+    b.teardown = [outPtr, devPtr] {
+      (void)hipFree(outPtr);
+      (void)hipFree(devPtr);
+    };
+    return b;
+  } else if (name == "hip_alltoall") {
+    void* outPtr = nullptr;
+    GA_HIP_CHECK(hipFree(devPtr));
+    GA_HIP_CHECK(hipMalloc(
+        &devPtr,
+        std::max<size_t>(elements * es * ctx->size, 16)));
+    hip::launchFillPattern(devPtr, elements * ctx->size, dtype, ctx->rank,
+                           1.0, nullptr);
+    GA_HIP_CHECK(
+        hipMalloc(&outPtr, std::max<size_t>(elements * es * ctx->size, 16)));
+    GA_HIP_CHECK(hipDeviceSynchronize());
+    auto algo = std::make_shared<hip::HipAlltoall>(ctx, device);
+    b.run = [algo, devPtr, outPtr, elements, es] {
+      algo->run(devPtr, outPtr, elements, es);
+    };
+    b.teardown = [outPtr, devPtr] {
+      (void)hipFree(outPtr);
+      (void)hipFree(devPtr);
+    };
+    return b;
   } else {
     GA_THROW("unknown hip benchmark: ", name);
   }

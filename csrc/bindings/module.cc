@@ -727,6 +727,79 @@ PYBIND11_MODULE(_C, m) {
           py::arg("dtype") = DType::F32,
           py::arg("op") = ReduceOp::SUM);
 
+  py::class_<hip::HipAllgatherRing>(m, "HipAllgatherRing")
+      .def(
+          py::init([](std::shared_ptr<Context> ctx, int device,
+                      size_t inboxCap) {
+            py::gil_scoped_release rel;
+            return std::make_unique<hip::HipAllgatherRing>(
+                ctx, device, inboxCap);
+          }),
+          py::arg("context"),
+          py::arg("device"),
+          py::arg("inbox_cap") = 0)
+      .def(
+          "run",
+          [](hip::HipAllgatherRing& a, uintptr_t in, uintptr_t out,
+             size_t inElements, size_t es) {
+            py::gil_scoped_release rel;
+            a.run(reinterpret_cast<const void*>(in),
+                  reinterpret_cast<void*>(out), inElements, es);
+          },
+          py::arg("in_ptr"),
+          py::arg("out_ptr"),
+          py::arg("in_elements"),
+          py::arg("element_size") = 4);
+
+  py::class_<hip::HipReduceScatterRing>(m, "HipReduceScatterRing")
+      .def(
+          py::init([](std::shared_ptr<Context> ctx, int device,
+                      size_t inboxCap) {
+            py::gil_scoped_release rel;
+            return std::make_unique<hip::HipReduceScatterRing>(
+                ctx, device, inboxCap);
+          }),
+          py::arg("context"),
+          py::arg("device"),
+          py::arg("inbox_cap") = 0)
+      .def(
+          "run",
+          [](hip::HipReduceScatterRing& a, uintptr_t in, uintptr_t out,
+             size_t recvElements, DType dt, ReduceOp op) {
+            py::gil_scoped_release rel;
+            a.run(reinterpret_cast<const void*>(in),
+                  reinterpret_cast<void*>(out), recvElements, dt, op);
+          },
+          py::arg("in_ptr"),
+          py::arg("out_ptr"),
+          py::arg("recv_elements"),
+          py::arg("dtype") = DType::F32,
+          py::arg("op") = ReduceOp::SUM);
+
+  py::class_<hip::HipAlltoall>(m, "HipAlltoall")
+      .def(
+          py::init([](std::shared_ptr<Context> ctx, int device,
+                      int numStreams) {
+            py::gil_scoped_release rel;
+            return std::make_unique<hip::HipAlltoall>(ctx, device,
+                                                      numStreams);
+          }),
+          py::arg("context"),
+          py::arg("device"),
+          py::arg("num_streams") = 4)
+      .def(
+          "run",
+          [](hip::HipAlltoall& a, uintptr_t in, uintptr_t out,
+             size_t perRankElements, size_t es) {
+            py::gil_scoped_release rel;
+            a.run(reinterpret_cast<const void*>(in),
+                  reinterpret_cast<void*>(out), perRankElements, es);
+          },
+          py::arg("in_ptr"),
+          py::arg("out_ptr"),
+          py::arg("per_rank_elements"),
+          py::arg("element_size") = 4);
+
   py::class_<hip::HipBroadcastOneToAll>(m, "HipBroadcastOneToAll")
       .def(
           py::init([](std::shared_ptr<Context> ctx, int device, int root,

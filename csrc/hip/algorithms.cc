@@ -493,6 +493,335 @@ void HipBroadcastOneToAll::run(void* devPtr, size_t bytes) {
 }
 
 // ===========================================================================
+// HipAllgatherRing
+// ===========================================================================
+
+HipAllgatherRing::HipAllgatherRing(
+    std::shared_ptr<Context> ctx,
+    int device,
+    size_t inboxCap)
+    : ctx_(std::move(ctx)),
+      device_(device),
+      inboxCap_(inboxCap == 0 ? kDefaultInboxCap : inboxCap) {
+  GA_HIP_CHECK(hipSetDevice(device_));
+  mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, inboxCap_);
+  cs_ = std::make_unique<HipStream>(device_, true);
+  ks_ = std::make_unique<HipStream>(device_);
+  initEvent_ = std::make_unique<HipEvent>(device_);
+  doneEvent_ = std::make_unique<HipEvent>(device_);
+  fDATA_ = mesh_->allocFlags(2);
+  fACK_ = mesh_->allocFlags(2);
+}
+
+void HipAllgatherRing::run(
+    const void* devIn,
+    void* devOut,
+    size_t inElements,
+    size_t es) {
+  GA_HIP_CHECK(hipSetDevice(device_));
+  const int P = ctx_->size;
+  const int r = ctx_->rank;
+  const size_t blockBytes = inElements * es;
+  if (P == 1) {
+    if (devOut != devIn) {
+      GA_HIP_CHECK(hipMemcpy(devOut, devIn, blockBytes,
+                             hipMemcpyDeviceToDevice));
+    }
+    return;
+  }
+  const size_t totalBytes = blockBytes * P;
+  const int S = std::max<int>(
+      1, static_cast<int>((blockBytes + inboxCap_ - 1) / inboxCap_));
+  const size_t segCapBytes = ((inElements + S - 1) / S) * es;
+  mesh_->ensureCapacity(totalBytes, segCapBytes);
+
+  const int right = (r + 1) % P;
+  const int left = (r - 1 + P) % P;
+  const int pool = S + 2;
+  while (static_cast<int>(events_.size()) < pool) {
+    events_.push_back(std::make_unique<HipEvent>(device_));
+  }
+  char* work = mesh_->work();
+  auto csm = cs_->stream();
+  auto ksm = ks_->stream();
+
+  GA_HIP_CHECK(hipMemcpyAsync(
+      work + r * blockBytes, devIn, blockBytes, hipMemcpyDeviceToDevice,
+      csm));
+  initEvent_->record(csm);
+
+  auto seqOf = [&](int k) { return seqBase_ + k + 1; };
+  const int K = (P - 1) * S;
+  for (int k = 0; k < K; k++) {
+    const int i = k / S, s = k % S;
+    const int par = k & 1;
+    const sched::Seg send =
+        sched::segmentOf(inElements, 1, 0, s, S); // within-block segment
+    const size_t sendBlock = (r - i + 2 * P) % P;
+    const size_t recvBlock = (r - i - 1 + 2 * P) % P;
+
+    const uint64_t prevAck = (k >= 2) ? seqOf(k - 2) : lastAck_[par];
+    if (prevAck > 0) {
+      launchWaitFlagGte(mesh_->flag(fACK_ + par), prevAck, ksm);
+    }
+    if (k >= S) {
+      events_[(k - S) % pool]->streamWait(ksm);
+    } else {
+      initEvent_->streamWait(ksm);
+    }
+    if (send.len > 0) {
+      GA_HIP_CHECK(hipMemcpyAsync(
+          mesh_->peerInbox(right, par),
+          work + sendBlock * blockBytes + send.off * es,
+          send.len * es,
+          hipMemcpyDeviceToDevice,
+          ksm));
+    }
+    launchWriteFlag(mesh_->peerFlag(right, fDATA_ + par), seqOf(k), ksm);
+
+    launchWaitFlagGte(mesh_->flag(fDATA_ + par), seqOf(k), csm);
+    if (send.len > 0) {
+      GA_HIP_CHECK(hipMemcpyAsync(
+          work + recvBlock * blockBytes + send.off * es,
+          mesh_->inbox(par),
+          send.len * es,
+          hipMemcpyDeviceToDevice,
+          csm));
+    }
+    launchWriteFlag(mesh_->peerFlag(left, fACK_ + par), seqOf(k), csm);
+    events_[k % pool]->record(csm);
+  }
+
+  GA_HIP_CHECK(hipMemcpyAsync(
+      devOut, work, totalBytes, hipMemcpyDeviceToDevice, csm));
+  doneEvent_->record(csm);
+  auto timeout = ctx_->getTimeout();
+  watchdogWait(*doneEvent_, *mesh_, timeout, "hip_allgather_ring (cs)");
+  initEvent_->record(ksm);
+  watchdogWait(*initEvent_, *mesh_, timeout, "hip_allgather_ring (ks)");
+  cs_->synchronize();
+  ks_->synchronize();
+  for (int k = std::max(0, K - 2); k < K; k++) {
+    lastAck_[k & 1] = seqOf(k);
+  }
+  seqBase_ += K;
+}
+
+// ===========================================================================
+// HipReduceScatterRing
+// ===========================================================================
+
+HipReduceScatterRing::HipReduceScatterRing(
+    std::shared_ptr<Context> ctx,
+    int device,
+    size_t inboxCap)
+    : ctx_(std::move(ctx)),
+      device_(device),
+      inboxCap_(inboxCap == 0 ? kDefaultInboxCap : inboxCap) {
+  GA_HIP_CHECK(hipSetDevice(device_));
+  mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, inboxCap_);
+  cs_ = std::make_unique<HipStream>(device_, true);
+  ks_ = std::make_unique<HipStream>(device_);
+  initEvent_ = std::make_unique<HipEvent>(device_);
+  doneEvent_ = std::make_unique<HipEvent>(device_);
+  fDATA_ = mesh_->allocFlags(2);
+  fACK_ = mesh_->allocFlags(2);
+}
+
+void HipReduceScatterRing::run(
+    const void* devIn,
+    void* devOut,
+    size_t recvElements,
+    DType dtype,
+    ReduceOp op) {
+  GA_HIP_CHECK(hipSetDevice(device_));
+  const int P = ctx_->size;
+  const int r = ctx_->rank;
+  const size_t es = dtypeSize(dtype);
+  const size_t blockBytes = recvElements * es;
+  if (P == 1) {
+    if (devOut != devIn) {
+      GA_HIP_CHECK(hipMemcpy(devOut, devIn, blockBytes,
+                             hipMemcpyDeviceToDevice));
+    }
+    return;
+  }
+  const size_t totalBytes = blockBytes * P;
+  const int S = std::max<int>(
+      1, static_cast<int>((blockBytes + inboxCap_ - 1) / inboxCap_));
+  const size_t segCapBytes = ((recvElements + S - 1) / S) * es;
+  mesh_->ensureCapacity(totalBytes, segCapBytes);
+
+  const int right = (r + 1) % P;
+  const int left = (r - 1 + P) % P;
+  const int pool = S + 2;
+  while (static_cast<int>(events_.size()) < pool) {
+    events_.push_back(std::make_unique<HipEvent>(device_));
+  }
+  char* work = mesh_->work();
+  auto csm = cs_->stream();
+  auto ksm = ks_->stream();
+
+  GA_HIP_CHECK(hipMemcpyAsync(
+      work, devIn, totalBytes, hipMemcpyDeviceToDevice, csm));
+  initEvent_->record(csm);
+
+  // Ring reduce-scatter ending with rank r owning block r:
+  // iteration i: send block (r-i-1), reduce block (r-i-2).
+  auto seqOf = [&](int k) { return seqBase_ + k + 1; };
+  const int K = (P - 1) * S;
+  for (int k = 0; k < K; k++) {
+    const int i = k / S, s = k % S;
+    const int par = k & 1;
+    const sched::Seg seg = sched::segmentOf(recvElements, 1, 0, s, S);
+    const size_t sendBlock = (r - i - 1 + 2 * P) % P;
+    const size_t recvBlock = (r - i - 2 + 2 * P) % P;
+
+    const uint64_t prevAck = (k >= 2) ? seqOf(k - 2) : lastAck_[par];
+    if (prevAck > 0) {
+      launchWaitFlagGte(mesh_->flag(fACK_ + par), prevAck, ksm);
+    }
+    if (k >= S) {
+      events_[(k - S) % pool]->streamWait(ksm);
+    } else {
+      initEvent_->streamWait(ksm);
+    }
+    if (seg.len > 0) {
+      GA_HIP_CHECK(hipMemcpyAsync(
+          mesh_->peerInbox(right, par),
+          work + sendBlock * blockBytes + seg.off * es,
+          seg.len * es,
+          hipMemcpyDeviceToDevice,
+          ksm));
+    }
+    launchWriteFlag(mesh_->peerFlag(right, fDATA_ + par), seqOf(k), ksm);
+
+    launchWaitFlagGte(mesh_->flag(fDATA_ + par), seqOf(k), csm);
+    if (seg.len > 0) {
+      launchReduce2(
+          work + recvBlock * blockBytes + seg.off * es,
+          work + recvBlock * blockBytes + seg.off * es,
+          mesh_->inbox(par),
+          seg.len,
+          dtype,
+          op,
+          csm);
+    }
+    launchWriteFlag(mesh_->peerFlag(left, fACK_ + par), seqOf(k), csm);
+    events_[k % pool]->record(csm);
+  }
+
+  GA_HIP_CHECK(hipMemcpyAsync(
+      devOut, work + r * blockBytes, blockBytes, hipMemcpyDeviceToDevice,
+      csm));
+  doneEvent_->record(csm);
+  auto timeout = ctx_->getTimeout();
+  watchdogWait(*doneEvent_, *mesh_, timeout, "hip_reduce_scatter (cs)");
+  initEvent_->record(ksm);
+  watchdogWait(*initEvent_, *mesh_, timeout, "hip_reduce_scatter (ks)");
+  cs_->synchronize();
+  ks_->synchronize();
+  for (int k = std::max(0, K - 2); k < K; k++) {
+    lastAck_[k & 1] = seqOf(k);
+  }
+  seqBase_ += K;
+}
+
+// ===========================================================================
+// HipAlltoall
+// ===========================================================================
+
+HipAlltoall::HipAlltoall(
+    std::shared_ptr<Context> ctx,
+    int device,
+    int numStreams)
+    : ctx_(std::move(ctx)), device_(device) {
+  GA_HIP_CHECK(hipSetDevice(device_));
+  mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, 4096);
+  cs_ = std::make_unique<HipStream>(device_, true);
+  for (int i = 0; i < numStreams; i++) {
+    fanout_.push_back(std::make_unique<HipStream>(device_));
+  }
+  fDATA_ = mesh_->allocFlags(ctx_->size);
+  fACK_ = mesh_->allocFlags(ctx_->size);
+}
+
+void HipAlltoall::run(
+    const void* devIn,
+    void* devOut,
+    size_t perRankElements,
+    size_t es) {
+  GA_HIP_CHECK(hipSetDevice(device_));
+  const int P = ctx_->size;
+  const int r = ctx_->rank;
+  const size_t blockBytes = perRankElements * es;
+  const char* in = static_cast<const char*>(devIn);
+  char* out = static_cast<char*>(devOut);
+  if (P == 1) {
+    if (devOut != devIn) {
+      GA_HIP_CHECK(hipMemcpy(out, in, blockBytes, hipMemcpyDeviceToDevice));
+    }
+    return;
+  }
+  mesh_->ensureCapacity(blockBytes * P, 4096);
+  char* work = mesh_->work();
+  const uint64_t seq = ++seq_;
+
+  // Push my block for each destination straight into its work region.
+  int si = 0;
+  for (int d = 0; d < P; d++) {
+    if (d == r) {
+      continue;
+    }
+    auto& st = *fanout_[si++ % fanout_.size()];
+    if (seq > 1) {
+      // d must have copied out the previous run's block from its work.
+      launchWaitFlagGte(mesh_->flag(fACK_ + d), seq - 1, st.stream());
+    }
+    if (blockBytes > 0) {
+      GA_HIP_CHECK(hipMemcpyAsync(
+          mesh_->peerWork(d) + static_cast<size_t>(r) * blockBytes,
+          in + static_cast<size_t>(d) * blockBytes,
+          blockBytes,
+          hipMemcpyDeviceToDevice,
+          st.stream()));
+    }
+    launchWriteFlag(mesh_->peerFlag(d, fDATA_ + r), seq, st.stream());
+  }
+  // Local block.
+  GA_HIP_CHECK(hipMemcpyAsync(
+      out + static_cast<size_t>(r) * blockBytes,
+      in + static_cast<size_t>(r) * blockBytes,
+      blockBytes,
+      hipMemcpyDeviceToDevice,
+      cs_->stream()));
+  // Collect incoming blocks as they land.
+  for (int s = 0; s < P; s++) {
+    if (s == r) {
+      continue;
+    }
+    launchWaitFlagGte(mesh_->flag(fDATA_ + s), seq, cs_->stream());
+    GA_HIP_CHECK(hipMemcpyAsync(
+        out + static_cast<size_t>(s) * blockBytes,
+        work + static_cast<size_t>(s) * blockBytes,
+        blockBytes,
+        hipMemcpyDeviceToDevice,
+        cs_->stream()));
+    launchWriteFlag(mesh_->peerFlag(s, fACK_ + r), seq, cs_->stream());
+  }
+  {
+    auto timeout = ctx_->getTimeout();
+    HipEvent done(device_);
+    done.record(cs_->stream());
+    watchdogWait(done, *mesh_, timeout, "hip_alltoall (cs)");
+  }
+  cs_->synchronize();
+  for (auto& st : fanout_) {
+    st->synchronize();
+  }
+}
+
+// ===========================================================================
 // hipAllreduceLocal
 // ===========================================================================
 
@@ -515,22 +844,25 @@ void hipAllreduceLocal(
   }
   HipStream& s = *cached;
   const size_t es = dtypeSize(dtype);
-  launchReduceN(
-      ptrs[0],
-      const_cast<const void* const*>(ptrs.data()),
-      static_cast<int>(std::min<size_t>(ptrs.size(), 8)),
-      elements,
-      dtype,
-      op,
-      s.stream());
-  // >8 pointers: fold the rest pairwise.
-  for (size_t i = 8; i < ptrs.size(); i++) {
-    launchReduce2(ptrs[0], ptrs[0], ptrs[i], elements, dtype, op, s.stream());
-  }
-  for (size_t i = 1; i < ptrs.size(); i++) {
-    GA_HIP_CHECK(hipMemcpyAsync(
-        ptrs[i], ptrs[0], elements * es, hipMemcpyDeviceToDevice,
-        s.stream()));
+  if (ptrs.size() <= 8) {
+    // Fused: one pass reduces and broadcasts (every pointer gets the
+    // result) — 2k*N traffic instead of (2k+1)*N plus copies.
+    launchReduceNAll(
+        ptrs.data(), static_cast<int>(ptrs.size()), elements, dtype, op,
+        s.stream());
+  } else {
+    launchReduceN(
+        ptrs[0], const_cast<const void* const*>(ptrs.data()), 8, elements,
+        dtype, op, s.stream());
+    for (size_t i = 8; i < ptrs.size(); i++) {
+      launchReduce2(ptrs[0], ptrs[0], ptrs[i], elements, dtype, op,
+                    s.stream());
+    }
+    for (size_t i = 1; i < ptrs.size(); i++) {
+      GA_HIP_CHECK(hipMemcpyAsync(
+          ptrs[i], ptrs[0], elements * es, hipMemcpyDeviceToDevice,
+          s.stream()));
+    }
   }
   s.synchronize();
 }

@@ -128,6 +128,88 @@ class HipBroadcastOneToAll {
   uint64_t seq_{0};
 };
 
+// Device-native allgather ring: out (size*inElements) assembled via the
+// same segmented inbox pipeline as the allreduce ring's allgather phase.
+class HipAllgatherRing {
+ public:
+  HipAllgatherRing(
+      std::shared_ptr<Context> ctx,
+      int device,
+      size_t inboxCap = 0);
+  // out must hold size * inElements elements.
+  void run(const void* devIn, void* devOut, size_t inElements, size_t es);
+
+ private:
+  std::shared_ptr<Context> ctx_;
+  int device_;
+  size_t inboxCap_;
+  std::unique_ptr<XgmiMesh> mesh_;
+  std::unique_ptr<HipStream> cs_;
+  std::unique_ptr<HipStream> ks_;
+  std::vector<std::unique_ptr<HipEvent>> events_;
+  std::unique_ptr<HipEvent> initEvent_;
+  std::unique_ptr<HipEvent> doneEvent_;
+  int fDATA_, fACK_;
+  uint64_t seqBase_{0};
+  uint64_t lastAck_[2] = {0, 0};
+};
+
+// Device-native ring reduce-scatter: rank r ends with the reduced
+// block r (recvElements) of the size*recvElements input.
+class HipReduceScatterRing {
+ public:
+  HipReduceScatterRing(
+      std::shared_ptr<Context> ctx,
+      int device,
+      size_t inboxCap = 0);
+  void run(
+      const void* devIn,
+      void* devOut,
+      size_t recvElements,
+      DType dtype,
+      ReduceOp op);
+
+ private:
+  std::shared_ptr<Context> ctx_;
+  int device_;
+  size_t inboxCap_;
+  std::unique_ptr<XgmiMesh> mesh_;
+  std::unique_ptr<HipStream> cs_;
+  std::unique_ptr<HipStream> ks_;
+  std::vector<std::unique_ptr<HipEvent>> events_;
+  std::unique_ptr<HipEvent> initEvent_;
+  std::unique_ptr<HipEvent> doneEvent_;
+  int fDATA_, fACK_;
+  uint64_t seqBase_{0};
+  uint64_t lastAck_[2] = {0, 0};
+};
+
+// Device-native all-to-all: every rank writes its block for rank d
+// directly into d's work region over xGMI (single writer per region;
+// fan-out over several streams so multiple links run concurrently).
+class HipAlltoall {
+ public:
+  HipAlltoall(
+      std::shared_ptr<Context> ctx,
+      int device,
+      int numStreams = 4);
+  void run(
+      const void* devIn,
+      void* devOut,
+      size_t perRankElements,
+      size_t es);
+
+ private:
+  std::shared_ptr<Context> ctx_;
+  int device_;
+  std::unique_ptr<XgmiMesh> mesh_;
+  std::unique_ptr<HipStream> cs_;
+  std::vector<std::unique_ptr<HipStream>> fanout_;
+  int fDATA_; // [src]
+  int fACK_; // [src]
+  uint64_t seq_{0};
+};
+
 // Single-process multi-pointer allreduce: fused k-way reduction into
 // ptrs[0] then broadcast copies (all on one device/stream; blocking).
 void hipAllreduceLocal(

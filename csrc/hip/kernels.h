@@ -40,6 +40,16 @@ void launchReduceN(
     ReduceOp op,
     hipStream_t stream);
 
+// In-place k-way allreduce: every pointer receives the reduced result
+// in one fused pass (k <= 8).
+void launchReduceNAll(
+    void* const* ptrs,
+    int k,
+    size_t n,
+    DType dtype,
+    ReduceOp op,
+    hipStream_t stream);
+
 // Stream-ordered doorbell ops over system-scope atomics (fine-grained
 // flag memory, peer-writable over xGMI). One 64-thread workgroup.
 void launchWriteFlag(uint64_t* addr, uint64_t val, hipStream_t stream);

@@ -154,6 +154,67 @@ __global__ __launch_bounds__(kBlock) void reduce2ScalarKernel(
   }
 }
 
+// In-place k-way allreduce: every source pointer receives the reduced
+// result (saves the separate broadcast pass: 2k*N traffic instead of
+// (2k+1)*N + k copies).
+template <typename T, typename OP, int K>
+__global__ __launch_bounds__(kBlock) void reduceNAllKernel(
+    T* __restrict__ s0,
+    T* __restrict__ s1,
+    T* __restrict__ s2,
+    T* __restrict__ s3,
+    T* __restrict__ s4,
+    T* __restrict__ s5,
+    T* __restrict__ s6,
+    T* __restrict__ s7,
+    size_t npacks,
+    size_t ntail) {
+  constexpr int V = VecOf<T>::value;
+  using P = Pack<T>;
+  using A = typename AccOf<T>::type;
+  T* srcs[8] = {s0, s1, s2, s3, s4, s5, s6, s7};
+  const size_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = gridDim.x * blockDim.x;
+  for (size_t i = tid; i < npacks; i += stride) {
+    P acc = reinterpret_cast<const P*>(srcs[0])[i];
+    A accv[V];
+#pragma unroll
+    for (int j = 0; j < V; j++) {
+      accv[j] = toAcc(acc.v[j]);
+    }
+#pragma unroll
+    for (int k = 1; k < K; k++) {
+      P pk = reinterpret_cast<const P*>(srcs[k])[i];
+#pragma unroll
+      for (int j = 0; j < V; j++) {
+        accv[j] = OP::apply(accv[j], toAcc(pk.v[j]));
+      }
+    }
+    P out;
+#pragma unroll
+    for (int j = 0; j < V; j++) {
+      out.v[j] = fromAcc<T>(accv[j]);
+    }
+#pragma unroll
+    for (int k = 0; k < K; k++) {
+      reinterpret_cast<P*>(srcs[k])[i] = out;
+    }
+  }
+  const size_t base = npacks * V;
+  for (size_t i = tid; i < ntail; i += stride) {
+    A acc = toAcc(srcs[0][base + i]);
+#pragma unroll
+    for (int k = 1; k < K; k++) {
+      acc = OP::apply(acc, toAcc(srcs[k][base + i]));
+    }
+    T r = fromAcc<T>(acc);
+#pragma unroll
+    for (int k = 0; k < K; k++) {
+      srcs[k][base + i] = r;
+    }
+  }
+}
+
 template <typename T, typename OP, int K>
 __global__ __launch_bounds__(kBlock) void reduceNKernel(
     T* __restrict__ dst,
@@ -362,6 +423,76 @@ void launchReduceNT(
   }
 }
 
+template <typename T, typename OP>
+void launchReduceNAllT(
+    void* const* ptrs,
+    int k,
+    size_t n,
+    hipStream_t stream) {
+  constexpr int V = VecOf<T>::value;
+  T* s[8] = {nullptr};
+  bool aligned = true;
+  for (int i = 0; i < k; i++) {
+    s[i] = static_cast<T*>(ptrs[i]);
+    aligned = aligned && (reinterpret_cast<uintptr_t>(s[i]) & 15) == 0;
+  }
+  for (int i = k; i < 8; i++) {
+    s[i] = s[0];
+  }
+  size_t npacks = aligned ? n / V : 0;
+  size_t ntail = n - npacks * V;
+  auto launch = [&](auto kval) {
+    constexpr int K = decltype(kval)::value;
+    hipLaunchKernelGGL(
+        (reduceNAllKernel<T, OP, K>),
+        dim3(gridFor(npacks + ntail)),
+        dim3(kBlock),
+        0,
+        stream,
+        s[0], s[1], s[2], s[3], s[4], s[5], s[6], s[7],
+        npacks,
+        ntail);
+    GA_HIP_CHECK(hipGetLastError());
+  };
+  switch (k) {
+    case 2:
+      return launch(std::integral_constant<int, 2>{});
+    case 3:
+      return launch(std::integral_constant<int, 3>{});
+    case 4:
+      return launch(std::integral_constant<int, 4>{});
+    case 5:
+      return launch(std::integral_constant<int, 5>{});
+    case 6:
+      return launch(std::integral_constant<int, 6>{});
+    case 7:
+      return launch(std::integral_constant<int, 7>{});
+    case 8:
+      return launch(std::integral_constant<int, 8>{});
+    default:
+      return;
+  }
+}
+
+template <typename T>
+void launchReduceNAllOp(
+    void* const* ptrs,
+    int k,
+    size_t n,
+    ReduceOp op,
+    hipStream_t stream) {
+  switch (op) {
+    case ReduceOp::SUM:
+      return launchReduceNAllT<T, OpSum>(ptrs, k, n, stream);
+    case ReduceOp::PRODUCT:
+      return launchReduceNAllT<T, OpProd>(ptrs, k, n, stream);
+    case ReduceOp::MIN:
+      return launchReduceNAllT<T, OpMin>(ptrs, k, n, stream);
+    case ReduceOp::MAX:
+      return launchReduceNAllT<T, OpMax>(ptrs, k, n, stream);
+  }
+}
+
 template <typename T>
 void launchReduceNOp(
     void* dst,
@@ -432,6 +563,18 @@ void launchReduceN(
     hipStream_t stream) {
   dispatchDType(dtype, [&](auto t) {
     launchReduceNOp<decltype(t)>(dst, srcs, k, n, op, stream);
+  });
+}
+
+void launchReduceNAll(
+    void* const* ptrs,
+    int k,
+    size_t n,
+    DType dtype,
+    ReduceOp op,
+    hipStream_t stream) {
+  dispatchDType(dtype, [&](auto t) {
+    launchReduceNAllOp<decltype(t)>(ptrs, k, n, op, stream);
   });
 }
 

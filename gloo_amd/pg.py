@@ -92,7 +92,9 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
         self._lock = threading.Lock()
         self._hip_ring = {}  # device -> HipAllreduceRing
         self._hip_bcast = {}  # (device, root) -> HipBroadcastOneToAll
-        self._pin = {}  # device -> pinned staging tensor
+        self._hip_ag = {}  # device -> HipAllgatherRing
+        self._hip_rs = {}  # device -> HipReduceScatterRing
+        self._hip_a2a = {}  # device -> HipAlltoall
 
     # -- helpers -------------------------------------------------------------
 
@@ -110,6 +112,22 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
             self._hip_bcast[key] = ga._C.HipBroadcastOneToAll(
                 self._ctx, device, root)
         return self._hip_bcast[key]
+
+    def _ag(self, device):
+        if device not in self._hip_ag:
+            self._hip_ag[device] = ga._C.HipAllgatherRing(self._ctx, device)
+        return self._hip_ag[device]
+
+    def _rs(self, device):
+        if device not in self._hip_rs:
+            self._hip_rs[device] = ga._C.HipReduceScatterRing(
+                self._ctx, device)
+        return self._hip_rs[device]
+
+    def _a2a(self, device):
+        if device not in self._hip_a2a:
+            self._hip_a2a[device] = ga._C.HipAlltoall(self._ctx, device)
+        return self._hip_a2a[device]
 
     def _staged(self, t, fn):
         """Run a CPU collective on a host copy of a CUDA tensor."""
@@ -161,11 +179,9 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
         with self._lock:
             tag = self._tag()
             if out.is_cuda:
-                h_out = out.cpu()
-                h_in = inp.cpu()
-                ga.allgather(self._ctx, h_out.data_ptr(), h_in.data_ptr(),
-                             h_in.numel(), _gdtype(inp), tag=tag)
-                out.copy_(h_out)
+                self._ag(out.get_device()).run(
+                    inp.data_ptr(), out.data_ptr(), inp.numel(),
+                    inp.element_size())
             else:
                 ga.allgather(self._ctx, out.data_ptr(), inp.data_ptr(),
                              inp.numel(), _gdtype(inp), tag=tag)
@@ -214,11 +230,9 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
         with self._lock:
             tag = self._tag()
             if out.is_cuda:
-                h_out = out.cpu()
-                h_in = inp.cpu()
-                ga.reduce_scatter(self._ctx, h_out.data_ptr(), h_in.data_ptr(),
-                                  h_out.numel(), _gdtype(out), op, tag=tag)
-                out.copy_(h_out)
+                self._rs(out.get_device()).run(
+                    inp.data_ptr(), out.data_ptr(), out.numel(),
+                    _gdtype(out), op)
             else:
                 ga.reduce_scatter(self._ctx, out.data_ptr(), inp.data_ptr(),
                                   out.numel(), _gdtype(out), op, tag=tag)
@@ -244,13 +258,19 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
                       input_split_sizes, opts=None):
         out = output.detach()
         inp = input.detach().contiguous()
+        if (out.is_cuda and not output_split_sizes
+                and not input_split_sizes):
+            with self._lock:
+                self._a2a(out.get_device()).run(
+                    inp.data_ptr(), out.data_ptr(),
+                    inp.numel() // self.size(), inp.element_size())
+            return _ret_work(output)
         h_out = out.cpu() if out.is_cuda else out
         h_in = inp.cpu() if inp.is_cuda else inp
         with self._lock:
             tag = self._tag()
             if not output_split_sizes and not input_split_sizes:
                 per = inp.numel() // self.size()
-                row = inp[0].numel() if inp.dim() > 0 else 1
                 ga.alltoall(self._ctx, h_out.data_ptr(), h_in.data_ptr(),
                             per, _gdtype(inp), tag=tag)
             else:

@@ -204,6 +204,144 @@ def test_hip_broadcast_one_to_all():
     assert not errors, errors[0]
 
 
+def _two_rank_generic(builder_and_check):
+    import threading
+
+    store = ga.HashStore()
+    errors = []
+
+    def worker(rank):
+        try:
+            dev = ga.create_tcp_device()
+            ctx = ga.Context(rank, 2)
+            ctx.connect_full_mesh(store, dev)
+            ctx.set_timeout(60000)
+            torch.cuda.set_device(0)
+            builder_and_check(ctx, rank)
+        except Exception:  # noqa: BLE001
+            import traceback
+
+            errors.append(traceback.format_exc())
+
+    ths = [threading.Thread(target=worker, args=(r,)) for r in range(2)]
+    [t.start() for t in ths]
+    [t.join(180) for t in ths]
+    assert not errors, errors[0]
+
+
+def test_hip_allgather_ring():
+    def fn(ctx, rank):
+        n = 1_000_000
+        inp = torch.full((n,), float(rank + 1), device="cuda")
+        out = torch.zeros(2 * n, device="cuda")
+        algo = ga._C.HipAllgatherRing(ctx, 0)
+        for _ in range(2):
+            algo.run(inp.data_ptr(), out.data_ptr(), n, 4)
+            torch.cuda.synchronize()
+            assert torch.all(out[:n] == 1.0) and torch.all(out[n:] == 2.0)
+
+    _two_rank_generic(fn)
+
+
+def test_hip_reduce_scatter_ring():
+    def fn(ctx, rank):
+        n = 500_000
+        g = torch.Generator("cpu").manual_seed(rank)
+        inp = torch.rand(2 * n, generator=g).cuda()
+        out = torch.zeros(n, device="cuda")
+        ref = sum(
+            torch.rand(2 * n, generator=torch.Generator("cpu").manual_seed(r))
+            for r in range(2))
+        algo = ga._C.HipReduceScatterRing(ctx, 0)
+        for _ in range(2):
+            algo.run(inp.data_ptr(), out.data_ptr(), n, ga.DType.f32,
+                     ga.ReduceOp.sum)
+            torch.cuda.synchronize()
+            expect = ref[rank * n:(rank + 1) * n]
+            assert torch.allclose(out.cpu(), expect, atol=1e-5)
+
+    _two_rank_generic(fn)
+
+
+def test_hip_alltoall():
+    def fn(ctx, rank):
+        n = 300_000
+        inp = torch.cat([
+            torch.full((n,), float(rank * 2 + d), device="cuda")
+            for d in range(2)
+        ])
+        out = torch.zeros(2 * n, device="cuda")
+        algo = ga._C.HipAlltoall(ctx, 0)
+        for _ in range(2):
+            algo.run(inp.data_ptr(), out.data_ptr(), n, 4)
+            torch.cuda.synchronize()
+            for s in range(2):
+                assert torch.all(out[s * n:(s + 1) * n] == s * 2 + rank)
+
+    _two_rank_generic(fn)
+
+
+def test_pg_cuda_collectives():
+    """ProcessGroup with CUDA tensors: device-native paths."""
+    from gloo_amd.pg import ProcessGroupGlooAmd
+
+    import threading
+
+    store = ga.HashStore()
+    errors = []
+
+    def worker(rank):
+        try:
+            torch.cuda.set_device(0)
+            pg = ProcessGroupGlooAmd(store, rank, 2)
+            t = torch.full((700_000,), float(rank + 1), device="cuda")
+            pg.allreduce([t]).wait()
+            torch.cuda.synchronize()
+            assert torch.all(t == 3.0)
+
+            inp = torch.full((100_000,), float(rank), device="cuda")
+            out = torch.zeros(200_000, device="cuda")
+            pg._allgather_base(out, inp).wait()
+            torch.cuda.synchronize()
+            assert torch.all(out[:100_000] == 0) and torch.all(
+                out[100_000:] == 1)
+
+            rs_in = torch.arange(200_000, dtype=torch.float32,
+                                 device="cuda") + rank
+            rs_out = torch.zeros(100_000, device="cuda")
+            pg._reduce_scatter_base(rs_out, rs_in).wait()
+            torch.cuda.synchronize()
+            expect = (torch.arange(200_000, dtype=torch.float32) * 2 +
+                      1)[rank * 100_000:(rank + 1) * 100_000]
+            assert torch.allclose(rs_out.cpu(), expect)
+
+            a2a_in = torch.cat([
+                torch.full((50_000,), float(rank * 2 + d), device="cuda")
+                for d in range(2)])
+            a2a_out = torch.zeros(100_000, device="cuda")
+            pg.alltoall_base(a2a_out, a2a_in, [], []).wait()
+            torch.cuda.synchronize()
+            for s in range(2):
+                assert torch.all(
+                    a2a_out[s * 50_000:(s + 1) * 50_000] == s * 2 + rank)
+
+            b = (torch.arange(50_000, dtype=torch.float32, device="cuda")
+                 if rank == 0 else torch.zeros(50_000, device="cuda"))
+            pg.broadcast([b]).wait()
+            torch.cuda.synchronize()
+            assert torch.allclose(
+                b.cpu(), torch.arange(50_000, dtype=torch.float32))
+        except Exception:  # noqa: BLE001
+            import traceback
+
+            errors.append(traceback.format_exc())
+
+    ths = [threading.Thread(target=worker, args=(r,)) for r in range(2)]
+    [t.start() for t in ths]
+    [t.join(180) for t in ths]
+    assert not errors, errors[0]
+
+
 def test_host_staged_small():
     """Below the on-device threshold the ring takes the host-staged path."""
     _two_rank_device_test("ring", 1000)  # 4 KB << 256 KB threshold
