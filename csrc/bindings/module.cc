@@ -13,6 +13,7 @@
 #include "algorithms/factory.h"
 #include "collectives/collectives.h"
 #include "collectives/reduce_fns.h"
+#include "common/linux.h"
 #include "common/store.h"
 #include "context.h"
 #include "hip/algorithms.h"
@@ -87,6 +88,12 @@ PYBIND11_MODULE(_C, m) {
       .value("max", ReduceOp::MAX);
 
   m.def("dtype_size", &dtypeSize);
+
+  // topology helpers (common/linux)
+  m.def("list_interfaces", &listInterfaces);
+  m.def("interface_speed", &getInterfaceSpeedByName);
+  m.def("interface_to_bus_id", &interfaceToBusID);
+  m.def("pci_distance", &pciDistance);
 
   // --- stores ---------------------------------------------------------------
   // Trampoline so Python classes (e.g. a torch.distributed Store adapter)
