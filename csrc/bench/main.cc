@@ -501,6 +501,9 @@ void runOne(
 
 int main(int argc, char** argv) {
   Options o = parse(argc, argv);
+  if (o.benchmark.rfind("hip_", 0) == 0) {
+    o.gpu = true; // hip_* benchmarks imply the device path
+  }
   GA_ENFORCE_GE(o.rank, 0);
   GA_ENFORCE_LT(o.rank, o.size);
 
