@@ -126,14 +126,16 @@ def _two_rank_device_test(algo_name, elements, dtype=torch.float32,
                 algo = ga._C.HipAllreduceHalvingDoubling(ctx, 0, inbox_cap)
             gdt = ga.dtype_from_torch(dtype)
             algo.run(x.data_ptr(), elements, gdt, ga.ReduceOp.sum)
-            torch.cuda.synchronize()
+            # NOTE: no torch.cuda.synchronize() in threaded 2-ranks-on-one-
+            # GPU tests: a device-wide sync would also wait on the OTHER
+            # rank's in-flight doorbell wait kernels (deadlock). run() is
+            # blocking (its streams are synced).
             got = x.float().cpu()
             tol = 2e-2 if dtype in (torch.bfloat16, torch.float16) else 1e-4
             assert torch.allclose(got, expect, rtol=tol, atol=tol), (
                 rank, (got - expect).abs().max().item())
             # run twice: cross-run seq/flag reuse
             algo.run(x.data_ptr(), elements, gdt, ga.ReduceOp.sum)
-            torch.cuda.synchronize()
             results[rank] = True
         except Exception:  # noqa: BLE001
             import traceback
@@ -191,7 +193,6 @@ def test_hip_broadcast_one_to_all():
                 x = torch.zeros(n, dtype=torch.float32).cuda()
             algo = ga._C.HipBroadcastOneToAll(ctx, 0, 0)
             algo.run(x.data_ptr(), x.numel() * 4)
-            torch.cuda.synchronize()
             assert torch.equal(x.cpu(), torch.arange(n, dtype=torch.float32))
         except Exception:  # noqa: BLE001
             import traceback
@@ -237,7 +238,6 @@ def test_hip_allgather_ring():
         algo = ga._C.HipAllgatherRing(ctx, 0)
         for _ in range(2):
             algo.run(inp.data_ptr(), out.data_ptr(), n, 4)
-            torch.cuda.synchronize()
             assert torch.all(out[:n] == 1.0) and torch.all(out[n:] == 2.0)
 
     _two_rank_generic(fn)
@@ -256,7 +256,6 @@ def test_hip_reduce_scatter_ring():
         for _ in range(2):
             algo.run(inp.data_ptr(), out.data_ptr(), n, ga.DType.f32,
                      ga.ReduceOp.sum)
-            torch.cuda.synchronize()
             expect = ref[rank * n:(rank + 1) * n]
             assert torch.allclose(out.cpu(), expect, atol=1e-5)
 
@@ -274,7 +273,6 @@ def test_hip_alltoall():
         algo = ga._C.HipAlltoall(ctx, 0)
         for _ in range(2):
             algo.run(inp.data_ptr(), out.data_ptr(), n, 4)
-            torch.cuda.synchronize()
             for s in range(2):
                 assert torch.all(out[s * n:(s + 1) * n] == s * 2 + rank)
 
@@ -296,13 +294,11 @@ def test_pg_cuda_collectives():
             pg = ProcessGroupGlooAmd(store, rank, 2)
             t = torch.full((700_000,), float(rank + 1), device="cuda")
             pg.allreduce([t]).wait()
-            torch.cuda.synchronize()
             assert torch.all(t == 3.0)
 
             inp = torch.full((100_000,), float(rank), device="cuda")
             out = torch.zeros(200_000, device="cuda")
             pg._allgather_base(out, inp).wait()
-            torch.cuda.synchronize()
             assert torch.all(out[:100_000] == 0) and torch.all(
                 out[100_000:] == 1)
 
@@ -310,7 +306,6 @@ def test_pg_cuda_collectives():
                                  device="cuda") + rank
             rs_out = torch.zeros(100_000, device="cuda")
             pg._reduce_scatter_base(rs_out, rs_in).wait()
-            torch.cuda.synchronize()
             expect = (torch.arange(200_000, dtype=torch.float32) * 2 +
                       1)[rank * 100_000:(rank + 1) * 100_000]
             assert torch.allclose(rs_out.cpu(), expect)
@@ -320,7 +315,6 @@ def test_pg_cuda_collectives():
                 for d in range(2)])
             a2a_out = torch.zeros(100_000, device="cuda")
             pg.alltoall_base(a2a_out, a2a_in, [], []).wait()
-            torch.cuda.synchronize()
             for s in range(2):
                 assert torch.all(
                     a2a_out[s * 50_000:(s + 1) * 50_000] == s * 2 + rank)
@@ -328,7 +322,6 @@ def test_pg_cuda_collectives():
             b = (torch.arange(50_000, dtype=torch.float32, device="cuda")
                  if rank == 0 else torch.zeros(50_000, device="cuda"))
             pg.broadcast([b]).wait()
-            torch.cuda.synchronize()
             assert torch.allclose(
                 b.cpu(), torch.arange(50_000, dtype=torch.float32))
         except Exception:  # noqa: BLE001
