@@ -673,19 +673,21 @@ PYBIND11_MODULE(_C, m) {
   m.def(
       "hip_allreduce_local",
       [](std::vector<uintptr_t> ptrs, size_t n, DType dt, ReduceOp op,
-         int device) {
+         int device, uintptr_t stream) {
         std::vector<void*> p;
         for (auto v : ptrs) {
           p.push_back(reinterpret_cast<void*>(v));
         }
         py::gil_scoped_release rel;
-        hip::hipAllreduceLocal(p, n, dt, op, device);
+        hip::hipAllreduceLocal(p, n, dt, op, device,
+                               reinterpret_cast<hipStream_t>(stream));
       },
       py::arg("ptrs"),
       py::arg("n"),
       py::arg("dtype"),
       py::arg("op"),
-      py::arg("device") = 0);
+      py::arg("device") = 0,
+      py::arg("stream") = 0);
 
   py::class_<hip::HipAllreduceRing>(m, "HipAllreduceRing")
       .def(
@@ -702,14 +704,16 @@ PYBIND11_MODULE(_C, m) {
       .def(
           "run",
           [](hip::HipAllreduceRing& a, uintptr_t ptr, size_t n, DType dt,
-             ReduceOp op) {
+             ReduceOp op, uintptr_t stream) {
             py::gil_scoped_release rel;
-            a.run(reinterpret_cast<void*>(ptr), n, dt, op);
+            a.run(reinterpret_cast<void*>(ptr), n, dt, op,
+                  reinterpret_cast<hipStream_t>(stream));
           },
           py::arg("ptr"),
           py::arg("elements"),
           py::arg("dtype") = DType::F32,
-          py::arg("op") = ReduceOp::SUM);
+          py::arg("op") = ReduceOp::SUM,
+          py::arg("stream") = 0);
 
   py::class_<hip::HipAllreduceHalvingDoubling>(m, "HipAllreduceHalvingDoubling")
       .def(
@@ -725,14 +729,16 @@ PYBIND11_MODULE(_C, m) {
       .def(
           "run",
           [](hip::HipAllreduceHalvingDoubling& a, uintptr_t ptr, size_t n,
-             DType dt, ReduceOp op) {
+             DType dt, ReduceOp op, uintptr_t stream) {
             py::gil_scoped_release rel;
-            a.run(reinterpret_cast<void*>(ptr), n, dt, op);
+            a.run(reinterpret_cast<void*>(ptr), n, dt, op,
+                  reinterpret_cast<hipStream_t>(stream));
           },
           py::arg("ptr"),
           py::arg("elements"),
           py::arg("dtype") = DType::F32,
-          py::arg("op") = ReduceOp::SUM);
+          py::arg("op") = ReduceOp::SUM,
+          py::arg("stream") = 0);
 
   py::class_<hip::HipAllgatherRing>(m, "HipAllgatherRing")
       .def(
@@ -748,15 +754,17 @@ PYBIND11_MODULE(_C, m) {
       .def(
           "run",
           [](hip::HipAllgatherRing& a, uintptr_t in, uintptr_t out,
-             size_t inElements, size_t es) {
+             size_t inElements, size_t es, uintptr_t stream) {
             py::gil_scoped_release rel;
             a.run(reinterpret_cast<const void*>(in),
-                  reinterpret_cast<void*>(out), inElements, es);
+                  reinterpret_cast<void*>(out), inElements, es,
+                  reinterpret_cast<hipStream_t>(stream));
           },
           py::arg("in_ptr"),
           py::arg("out_ptr"),
           py::arg("in_elements"),
-          py::arg("element_size") = 4);
+          py::arg("element_size") = 4,
+          py::arg("stream") = 0);
 
   py::class_<hip::HipReduceScatterRing>(m, "HipReduceScatterRing")
       .def(
@@ -772,16 +780,18 @@ PYBIND11_MODULE(_C, m) {
       .def(
           "run",
           [](hip::HipReduceScatterRing& a, uintptr_t in, uintptr_t out,
-             size_t recvElements, DType dt, ReduceOp op) {
+             size_t recvElements, DType dt, ReduceOp op, uintptr_t stream) {
             py::gil_scoped_release rel;
             a.run(reinterpret_cast<const void*>(in),
-                  reinterpret_cast<void*>(out), recvElements, dt, op);
+                  reinterpret_cast<void*>(out), recvElements, dt, op,
+                  reinterpret_cast<hipStream_t>(stream));
           },
           py::arg("in_ptr"),
           py::arg("out_ptr"),
           py::arg("recv_elements"),
           py::arg("dtype") = DType::F32,
-          py::arg("op") = ReduceOp::SUM);
+          py::arg("op") = ReduceOp::SUM,
+          py::arg("stream") = 0);
 
   py::class_<hip::HipAlltoall>(m, "HipAlltoall")
       .def(
@@ -797,15 +807,17 @@ PYBIND11_MODULE(_C, m) {
       .def(
           "run",
           [](hip::HipAlltoall& a, uintptr_t in, uintptr_t out,
-             size_t perRankElements, size_t es) {
+             size_t perRankElements, size_t es, uintptr_t stream) {
             py::gil_scoped_release rel;
             a.run(reinterpret_cast<const void*>(in),
-                  reinterpret_cast<void*>(out), perRankElements, es);
+                  reinterpret_cast<void*>(out), perRankElements, es,
+                  reinterpret_cast<hipStream_t>(stream));
           },
           py::arg("in_ptr"),
           py::arg("out_ptr"),
           py::arg("per_rank_elements"),
-          py::arg("element_size") = 4);
+          py::arg("element_size") = 4,
+          py::arg("stream") = 0);
 
   py::class_<hip::HipBroadcastOneToAll>(m, "HipBroadcastOneToAll")
       .def(
@@ -821,10 +833,13 @@ PYBIND11_MODULE(_C, m) {
           py::arg("num_streams") = 4)
       .def(
           "run",
-          [](hip::HipBroadcastOneToAll& a, uintptr_t ptr, size_t bytes) {
+          [](hip::HipBroadcastOneToAll& a, uintptr_t ptr, size_t bytes,
+             uintptr_t stream) {
             py::gil_scoped_release rel;
-            a.run(reinterpret_cast<void*>(ptr), bytes);
+            a.run(reinterpret_cast<void*>(ptr), bytes,
+                  reinterpret_cast<hipStream_t>(stream));
           },
           py::arg("ptr"),
-          py::arg("bytes"));
+          py::arg("bytes"),
+          py::arg("stream") = 0);
 }

@@ -36,12 +36,37 @@ void watchdogWait(
   auto deadline = std::chrono::steady_clock::now() + timeout;
   while (!ev.query()) {
     if (timeout.count() > 0 && std::chrono::steady_clock::now() > deadline) {
+      std::string flags;
+      if (getEnvFlag("GLOO_AMD_FLAG_DEBUG")) {
+        uint64_t vals[16] = {0};
+        (void)hipMemcpy(vals, mesh.flag(0), sizeof(vals),
+                        hipMemcpyDeviceToHost);
+        for (int i = 0; i < 16; i++) {
+          flags += " " + std::to_string(vals[i]);
+        }
+      }
       mesh.poisonFlags();
       (void)hipDeviceSynchronize();
-      GA_THROW_IO("hip collective timed out in ", what);
+      GA_THROW_IO("hip collective timed out in ", what, " flags:", flags);
     }
     std::this_thread::yield();
   }
+}
+
+// Order our streams after the caller's stream (where the input tensors
+// were produced). nullptr = legacy default stream, which covers torch's
+// default current stream. The transient event is destroyed immediately:
+// HIP frees it once the recorded work and waits complete.
+void gateStreams(
+    hipStream_t caller,
+    std::initializer_list<hipStream_t> gated) {
+  hipEvent_t ev = nullptr;
+  GA_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+  GA_HIP_CHECK(hipEventRecord(ev, caller));
+  for (auto s : gated) {
+    GA_HIP_CHECK(hipStreamWaitEvent(s, ev, 0));
+  }
+  GA_HIP_CHECK(hipEventDestroy(ev));
 }
 } // namespace
 
@@ -72,13 +97,15 @@ void HipAllreduceRing::run(
     void* devPtr,
     size_t elements,
     DType dtype,
-    ReduceOp op) {
+    ReduceOp op,
+    hipStream_t callerStream) {
   GA_HIP_CHECK(hipSetDevice(device_));
   const size_t es = dtypeSize(dtype);
   const size_t bytes = elements * es;
   if (ctx_->size == 1 || elements == 0) {
     return;
   }
+  gateStreams(callerStream, {cs_->stream(), ks_->stream()});
   if (bytes < onDeviceThreshold()) {
     runHostStaged(static_cast<char*>(devPtr), bytes, elements, dtype, op);
   } else {
@@ -281,7 +308,8 @@ void HipAllreduceHalvingDoubling::run(
     void* devPtr,
     size_t elements,
     DType dtype,
-    ReduceOp op) {
+    ReduceOp op,
+    hipStream_t callerStream) {
   GA_HIP_CHECK(hipSetDevice(device_));
   const int P = ctx_->size;
   const int r = ctx_->rank;
@@ -290,6 +318,7 @@ void HipAllreduceHalvingDoubling::run(
   if (P == 1 || elements == 0) {
     return;
   }
+  gateStreams(callerStream, {cs_->stream(), ks_->stream()});
   char* buf = static_cast<char*>(devPtr);
   const int T = log2P_;
   mesh_->ensureCapacity(bytes, inboxCap_);
@@ -444,7 +473,10 @@ HipBroadcastOneToAll::HipBroadcastOneToAll(
   fBACK_ = mesh_->allocFlags(ctx_->size);
 }
 
-void HipBroadcastOneToAll::run(void* devPtr, size_t bytes) {
+void HipBroadcastOneToAll::run(
+    void* devPtr,
+    size_t bytes,
+    hipStream_t callerStream) {
   GA_HIP_CHECK(hipSetDevice(device_));
   const int P = ctx_->size;
   const int r = ctx_->rank;
@@ -452,10 +484,12 @@ void HipBroadcastOneToAll::run(void* devPtr, size_t bytes) {
     return;
   }
   mesh_->ensureCapacity(bytes, 4096);
+  gateStreams(callerStream, {cs_->stream()});
   char* work = mesh_->work();
   char* buf = static_cast<char*>(devPtr);
   seq_++;
 
+  auto timeout = ctx_->getTimeout();
   if (r == root_) {
     GA_HIP_CHECK(hipMemcpyAsync(
         work, buf, bytes, hipMemcpyDeviceToDevice, cs_->stream()));
@@ -481,6 +515,9 @@ void HipBroadcastOneToAll::run(void* devPtr, size_t bytes) {
     }
     cs_->synchronize();
     for (auto& st : fanout_) {
+      HipEvent done(device_);
+      done.record(st->stream());
+      watchdogWait(done, *mesh_, timeout, "hip_broadcast (fanout)");
       st->synchronize();
     }
   } else {
@@ -488,6 +525,9 @@ void HipBroadcastOneToAll::run(void* devPtr, size_t bytes) {
     GA_HIP_CHECK(hipMemcpyAsync(
         buf, work, bytes, hipMemcpyDeviceToDevice, cs_->stream()));
     launchWriteFlag(mesh_->peerFlag(root_, fBACK_ + r), seq_, cs_->stream());
+    HipEvent done(device_);
+    done.record(cs_->stream());
+    watchdogWait(done, *mesh_, timeout, "hip_broadcast (recv)");
     cs_->synchronize();
   }
 }
@@ -517,7 +557,8 @@ void HipAllgatherRing::run(
     const void* devIn,
     void* devOut,
     size_t inElements,
-    size_t es) {
+    size_t es,
+    hipStream_t callerStream) {
   GA_HIP_CHECK(hipSetDevice(device_));
   const int P = ctx_->size;
   const int r = ctx_->rank;
@@ -534,6 +575,7 @@ void HipAllgatherRing::run(
       1, static_cast<int>((blockBytes + inboxCap_ - 1) / inboxCap_));
   const size_t segCapBytes = ((inElements + S - 1) / S) * es;
   mesh_->ensureCapacity(totalBytes, segCapBytes);
+  gateStreams(callerStream, {cs_->stream(), ks_->stream()});
 
   const int right = (r + 1) % P;
   const int left = (r - 1 + P) % P;
@@ -633,7 +675,8 @@ void HipReduceScatterRing::run(
     void* devOut,
     size_t recvElements,
     DType dtype,
-    ReduceOp op) {
+    ReduceOp op,
+    hipStream_t callerStream) {
   GA_HIP_CHECK(hipSetDevice(device_));
   const int P = ctx_->size;
   const int r = ctx_->rank;
@@ -651,6 +694,7 @@ void HipReduceScatterRing::run(
       1, static_cast<int>((blockBytes + inboxCap_ - 1) / inboxCap_));
   const size_t segCapBytes = ((recvElements + S - 1) / S) * es;
   mesh_->ensureCapacity(totalBytes, segCapBytes);
+  gateStreams(callerStream, {cs_->stream(), ks_->stream()});
 
   const int right = (r + 1) % P;
   const int left = (r - 1 + P) % P;
@@ -750,7 +794,8 @@ void HipAlltoall::run(
     const void* devIn,
     void* devOut,
     size_t perRankElements,
-    size_t es) {
+    size_t es,
+    hipStream_t callerStream) {
   GA_HIP_CHECK(hipSetDevice(device_));
   const int P = ctx_->size;
   const int r = ctx_->rank;
@@ -764,6 +809,19 @@ void HipAlltoall::run(
     return;
   }
   mesh_->ensureCapacity(blockBytes * P, 4096);
+  {
+    std::vector<hipStream_t> gated{cs_->stream()};
+    for (auto& st : fanout_) {
+      gated.push_back(st->stream());
+    }
+    hipEvent_t ev = nullptr;
+    GA_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+    GA_HIP_CHECK(hipEventRecord(ev, callerStream));
+    for (auto g : gated) {
+      GA_HIP_CHECK(hipStreamWaitEvent(g, ev, 0));
+    }
+    GA_HIP_CHECK(hipEventDestroy(ev));
+  }
   char* work = mesh_->work();
   const uint64_t seq = ++seq_;
 
@@ -817,6 +875,9 @@ void HipAlltoall::run(
   }
   cs_->synchronize();
   for (auto& st : fanout_) {
+    HipEvent done(device_);
+    done.record(st->stream());
+    watchdogWait(done, *mesh_, ctx_->getTimeout(), "hip_alltoall (fanout)");
     st->synchronize();
   }
 }
@@ -830,7 +891,8 @@ void hipAllreduceLocal(
     size_t elements,
     DType dtype,
     ReduceOp op,
-    int device) {
+    int device,
+    hipStream_t callerStream) {
   GA_ENFORCE(!ptrs.empty());
   if (ptrs.size() == 1 || elements == 0) {
     return;
@@ -843,6 +905,7 @@ void hipAllreduceLocal(
     cached = std::make_unique<HipStream>(device);
   }
   HipStream& s = *cached;
+  gateStreams(callerStream, {s.stream()});
   const size_t es = dtypeSize(dtype);
   if (ptrs.size() <= 8) {
     // Fused: one pass reduces and broadcasts (every pointer gets the

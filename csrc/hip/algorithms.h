@@ -49,7 +49,14 @@ class HipAllreduceRing {
       size_t inboxCap = 0 /*0 -> default 4 MiB*/);
 
   // In-place allreduce of a device buffer. Blocking (streams synced).
-  void run(void* devPtr, size_t elements, DType dtype, ReduceOp op);
+  // callerStream: the stream on which the caller produced devPtr (e.g.
+  // torch.cuda.current_stream()); our streams are event-ordered after it.
+  void run(
+      void* devPtr,
+      size_t elements,
+      DType dtype,
+      ReduceOp op,
+      hipStream_t callerStream = nullptr);
 
   XgmiMesh* mesh() {
     return mesh_.get();
@@ -83,7 +90,12 @@ class HipAllreduceHalvingDoubling {
       int device,
       size_t inboxCap = 0);
 
-  void run(void* devPtr, size_t elements, DType dtype, ReduceOp op);
+  void run(
+      void* devPtr,
+      size_t elements,
+      DType dtype,
+      ReduceOp op,
+      hipStream_t callerStream = nullptr);
 
  private:
   std::shared_ptr<Context> ctx_;
@@ -114,7 +126,7 @@ class HipBroadcastOneToAll {
       int root = 0,
       int numStreams = 4);
 
-  void run(void* devPtr, size_t bytes);
+  void run(void* devPtr, size_t bytes, hipStream_t callerStream = nullptr);
 
  private:
   std::shared_ptr<Context> ctx_;
@@ -137,7 +149,12 @@ class HipAllgatherRing {
       int device,
       size_t inboxCap = 0);
   // out must hold size * inElements elements.
-  void run(const void* devIn, void* devOut, size_t inElements, size_t es);
+  void run(
+      const void* devIn,
+      void* devOut,
+      size_t inElements,
+      size_t es,
+      hipStream_t callerStream = nullptr);
 
  private:
   std::shared_ptr<Context> ctx_;
@@ -167,7 +184,8 @@ class HipReduceScatterRing {
       void* devOut,
       size_t recvElements,
       DType dtype,
-      ReduceOp op);
+      ReduceOp op,
+      hipStream_t callerStream = nullptr);
 
  private:
   std::shared_ptr<Context> ctx_;
@@ -197,7 +215,8 @@ class HipAlltoall {
       const void* devIn,
       void* devOut,
       size_t perRankElements,
-      size_t es);
+      size_t es,
+      hipStream_t callerStream = nullptr);
 
  private:
   std::shared_ptr<Context> ctx_;
@@ -217,7 +236,8 @@ void hipAllreduceLocal(
     size_t elements,
     DType dtype,
     ReduceOp op,
-    int device);
+    int device,
+    hipStream_t callerStream = nullptr);
 
 } // namespace hip
 } // namespace glooamd

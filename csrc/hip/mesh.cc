@@ -142,7 +142,21 @@ void XgmiMesh::ensureCapacity(size_t workCap, size_t inboxCap) {
   }
   workCap_ = std::max(workCap_, workCap);
   inboxCap_ = std::max(inboxCap_, inboxCap);
+  // Every rank must close its mapping of our old buffer BEFORE we free
+  // it: freeing memory a peer still has IPC-mapped leaves the backing
+  // pages pinned and the replacement allocation can fail to export
+  // (hipIpcGetMemHandle: invalid argument).
+  {
+    BarrierOptions bar(ctx_);
+    bar.tag = ctx_->nextSlot();
+    barrier(bar);
+  }
   releasePeers();
+  {
+    BarrierOptions bar(ctx_);
+    bar.tag = ctx_->nextSlot();
+    barrier(bar);
+  }
   {
     std::lock_guard<std::mutex> lock(allocMutex());
     (void)hipFree(data_);

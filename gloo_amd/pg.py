@@ -79,6 +79,12 @@ def _gdtype(t):
     return ga.dtype_from_torch(t.dtype)
 
 
+def _cur_stream(t):
+    """The caller stream that produced tensor t (our device algorithms
+    order themselves after it)."""
+    return torch.cuda.current_stream(t.get_device()).cuda_stream
+
+
 class ProcessGroupGlooAmd(dist.ProcessGroup):
     def __init__(self, store, rank, size, timeout=timedelta(seconds=300)):
         super().__init__(rank, size)
@@ -148,7 +154,8 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
                 assert t_.is_contiguous(), "gloo_amd needs contiguous tensors"
                 if t_.is_cuda:
                     self._ring(t_.get_device()).run(
-                        t_.data_ptr(), t_.numel(), _gdtype(t_), gop)
+                        t_.data_ptr(), t_.numel(), _gdtype(t_), gop,
+                        stream=_cur_stream(t_))
                 else:
                     ga.allreduce(
                         self._ctx, [t_.data_ptr()], t_.numel(), _gdtype(t_),
@@ -165,7 +172,8 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
                 assert t_.is_contiguous()
                 if t_.is_cuda:
                     self._bcaster(t_.get_device(), root).run(
-                        t_.data_ptr(), t_.numel() * t_.element_size())
+                        t_.data_ptr(), t_.numel() * t_.element_size(),
+                        stream=_cur_stream(t_))
                 else:
                     ga.broadcast(
                         self._ctx, t_.data_ptr(), 0, t_.numel(), _gdtype(t_),
@@ -181,7 +189,7 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
             if out.is_cuda:
                 self._ag(out.get_device()).run(
                     inp.data_ptr(), out.data_ptr(), inp.numel(),
-                    inp.element_size())
+                    inp.element_size(), stream=_cur_stream(out))
             else:
                 ga.allgather(self._ctx, out.data_ptr(), inp.data_ptr(),
                              inp.numel(), _gdtype(inp), tag=tag)
@@ -232,7 +240,7 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
             if out.is_cuda:
                 self._rs(out.get_device()).run(
                     inp.data_ptr(), out.data_ptr(), out.numel(),
-                    _gdtype(out), op)
+                    _gdtype(out), op, stream=_cur_stream(out))
             else:
                 ga.reduce_scatter(self._ctx, out.data_ptr(), inp.data_ptr(),
                                   out.numel(), _gdtype(out), op, tag=tag)
@@ -263,7 +271,8 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
             with self._lock:
                 self._a2a(out.get_device()).run(
                     inp.data_ptr(), out.data_ptr(),
-                    inp.numel() // self.size(), inp.element_size())
+                    inp.numel() // self.size(), inp.element_size(),
+                    stream=_cur_stream(out))
             return _ret_work(output)
         h_out = out.cpu() if out.is_cuda else out
         h_in = inp.cpu() if inp.is_cuda else inp
