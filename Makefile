@@ -26,7 +26,7 @@ CC_OBJS := $(patsubst csrc/%.cc,build/%.o,$(CC_SRCS))
 HIP_OBJS := $(patsubst csrc/%.hip,build/%.hip.o,$(HIP_SRCS))
 BIND_OBJS := $(patsubst csrc/%.cc,build/%.o,$(BIND_SRCS))
 
-LDFLAGS := -L$(ROCM)/lib -lamdhip64 -pthread
+LDFLAGS := -L$(ROCM)/lib -lamdhip64 -lssl -lcrypto -pthread
 
 all: $(TARGET)
 

@@ -21,6 +21,7 @@
 #include "rendezvous/stores.h"
 #include "transport/tcp/context.h"
 #include "transport/tcp/device.h"
+#include "transport/tcp/tls.h"
 
 namespace py = pybind11;
 using namespace glooamd;
@@ -185,6 +186,23 @@ PYBIND11_MODULE(_C, m) {
             tcp::createTcpDevice(attr));
       },
       py::arg("hostname") = std::string());
+
+  m.def(
+      "create_tls_device",
+      [](const std::string& hostname, const std::string& pkey,
+         const std::string& cert, const std::string& caFile) {
+        tcp::tls::TlsAttr attr;
+        attr.tcp.hostname = hostname;
+        attr.pkeyFile = pkey;
+        attr.certFile = cert;
+        attr.caFile = caFile;
+        return std::static_pointer_cast<transport::Device>(
+            tcp::tls::createTlsDevice(attr));
+      },
+      py::arg("hostname") = std::string(),
+      py::arg("pkey") = std::string(),
+      py::arg("cert") = std::string(),
+      py::arg("ca_file") = std::string());
 
   // --- unbound + bound buffers ---------------------------------------------
   py::class_<transport::UnboundBuffer>(m, "UnboundBuffer")

@@ -73,6 +73,12 @@ class TcpContext : public transport::Context,
     return device_.get();
   }
 
+ protected:
+  // For transport subclasses (tls) overriding createPair.
+  void setPair(int rank, std::unique_ptr<TcpPair> pair) {
+    pairs_[rank] = std::move(pair);
+  }
+
  private:
   friend class TcpPair;
   friend class TcpUnboundBuffer;

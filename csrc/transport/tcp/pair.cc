@@ -82,11 +82,20 @@ std::string TcpPair::str() const {
       std::to_string(peerRank_) + " " + self_.str() + "]";
 }
 
+ssize_t TcpPair::ioRead(char* buf, size_t len) {
+  return ::read(fd_, buf, len);
+}
+
+ssize_t TcpPair::ioWritev(const struct iovec* iov, int iovcnt) {
+  return ::writev(fd_, iov, iovcnt);
+}
+
 void TcpPair::connect(const std::vector<char>& peerAddressBytes) {
   peer_ = TcpAddress::fromBytes(peerAddressBytes);
   auto timeout = ctx_->getTimeout();
+  bool initiator = ctx_->rank < peerRank_;
 
-  if (ctx_->rank < peerRank_) {
+  if (initiator) {
     // Initiator: dial the peer's listener and write the peer pair's seq.
     int fd = -1;
     int attempts = 0;
@@ -120,13 +129,15 @@ void TcpPair::connect(const std::vector<char>& peerAddressBytes) {
       GA_ENFORCE_GT(n, 0, "seq write: ", strerror(errno));
       written += n;
     }
-    setNonBlocking(fd);
     setSocketOptions(fd);
     fd_ = fd;
   } else {
     // Listener: wait for the peer to dial with our seq.
     fd_ = dev_->waitForConnection(self_.seq(), timeout);
   }
+
+  ioHandshake(initiator); // e.g. TLS; fd may be blocking here
+  setNonBlocking(fd_);
 
   {
     std::lock_guard<std::mutex> lock(ctx_->mu_);
@@ -148,6 +159,7 @@ void TcpPair::close() {
   }
   if (fd >= 0) {
     dev_->loop().unregisterDescriptor(fd);
+    ioClose();
     ::close(fd);
   }
 }
@@ -346,7 +358,7 @@ void TcpPair::flushTxLocked() {
       tx_.pop_front();
       continue;
     }
-    ssize_t n = writev(fd_, iov, iovcnt);
+    ssize_t n = ioWritev(iov, iovcnt);
     if (n < 0) {
       if (errno == EAGAIN || errno == EWOULDBLOCK) {
         armEpollOutLocked();
@@ -410,8 +422,7 @@ void TcpPair::handleEvents(uint32_t events) {
 
 bool TcpPair::readPreamble() {
   while (rxPreRead_ < sizeof(Preamble)) {
-    ssize_t n = read(
-        fd_,
+    ssize_t n = ioRead(
         reinterpret_cast<char*>(&rxPre_) + rxPreRead_,
         sizeof(Preamble) - rxPreRead_);
     if (n > 0) {
@@ -449,7 +460,7 @@ void TcpPair::readLoop() {
     }
     // Payload phase: read into the destination chosen by dispatch.
     while (rxRead_ < rxLen_) {
-      ssize_t n = read(fd_, rxDst_ + rxRead_, rxLen_ - rxRead_);
+      ssize_t n = ioRead(rxDst_ + rxRead_, rxLen_ - rxRead_);
       if (n > 0) {
         rxRead_ += n;
         continue;

@@ -17,6 +17,8 @@
 //    whichever thread holds the lock (inline fast path + EPOLLOUT drain).
 #pragma once
 
+#include <sys/uio.h>
+
 #include <deque>
 #include <string>
 #include <unordered_map>
@@ -115,6 +117,21 @@ class TcpPair : public transport::Pair, public Handler {
 
   // loop thread entry
   void handleEvents(uint32_t events) override;
+
+ protected:
+  // Byte-level I/O, overridable by the TLS pair. Semantics match
+  // read(2)/writev(2): >0 bytes, 0 EOF (read), -1 with errno
+  // (EAGAIN/EWOULDBLOCK = retry later).
+  virtual ssize_t ioRead(char* buf, size_t len);
+  virtual ssize_t ioWritev(const struct iovec* iov, int iovcnt);
+  // Called from connect() once the raw socket is established (blocking
+  // allowed; fd_ is still in blocking mode for the initiator path).
+  virtual void ioHandshake(bool initiator) {}
+  virtual void ioClose() {}
+
+  int fd() const {
+    return fd_;
+  }
 
  private:
   struct TxOp {

@@ -41,6 +41,7 @@ from gloo_amd._C import (  # noqa: F401
     barrier,
     broadcast,
     create_tcp_device,
+    create_tls_device,
     dtype_size,
     gather,
     gatherv,
