@@ -66,7 +66,8 @@ struct Options {
           "  new_allreduce_ring, new_allreduce_bcube, sendrecv_roundtrip,\n"
           "  hip_allreduce_ring, hip_allreduce_ring_chunked,\n"
           "  hip_allreduce_halving_doubling, hip_broadcast_one_to_all,\n"
-          "  hip_allgather_ring, hip_reduce_scatter, hip_alltoall\n");
+          "  hip_allgather_ring, hip_reduce_scatter, hip_alltoall,\n"
+          "  hip_allreduce_direct\n");
   exit(1);
 }
 
@@ -332,6 +333,12 @@ Bench makeHipBench(
   if (name == "hip_allreduce_ring" || name == "hip_allreduce_ring_chunked") {
     auto algo = std::make_shared<hip::HipAllreduceRing>(
         ctx, device, name == "hip_allreduce_ring_chunked");
+    DType dt = dtype;
+    b.run = [algo, devPtr, elements, dt] {
+      algo->run(devPtr, elements, dt, ReduceOp::SUM);
+    };
+  } else if (name == "hip_allreduce_direct") {
+    auto algo = std::make_shared<hip::HipAllreduceDirect>(ctx, device);
     DType dt = dtype;
     b.run = [algo, devPtr, elements, dt] {
       algo->run(devPtr, elements, dt, ReduceOp::SUM);

@@ -533,6 +533,157 @@ void HipBroadcastOneToAll::run(
 }
 
 // ===========================================================================
+// HipAllreduceDirect
+// ===========================================================================
+
+HipAllreduceDirect::HipAllreduceDirect(
+    std::shared_ptr<Context> ctx,
+    int device,
+    int numStreams)
+    : ctx_(std::move(ctx)), device_(device) {
+  GA_ENFORCE_LE(ctx_->size, 8, "direct allreduce supports <= 8 ranks");
+  GA_HIP_CHECK(hipSetDevice(device_));
+  mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, kDefaultInboxCap);
+  cs_ = std::make_unique<HipStream>(device_, true);
+  const int nf = std::min(numStreams, std::max(1, ctx_->size - 1));
+  for (int i = 0; i < nf; i++) {
+    fanout_.push_back(std::make_unique<HipStream>(device_));
+  }
+  doneEvent_ = std::make_unique<HipEvent>(device_);
+  fRS_ = mesh_->allocFlags(ctx_->size);
+  fAG_ = mesh_->allocFlags(ctx_->size);
+  fACK_ = mesh_->allocFlags(ctx_->size);
+}
+
+void HipAllreduceDirect::run(
+    void* devPtr,
+    size_t elements,
+    DType dtype,
+    ReduceOp op,
+    hipStream_t callerStream) {
+  GA_HIP_CHECK(hipSetDevice(device_));
+  const int P = ctx_->size;
+  const int r = ctx_->rank;
+  const size_t es = dtypeSize(dtype);
+  if (P == 1 || elements == 0) {
+    return;
+  }
+  char* user = static_cast<char*>(devPtr);
+  const size_t perRank = (elements + P - 1) / P;
+  const size_t blockCapBytes = perRank * es;
+  // inbox holds one scatter slot per source rank.
+  mesh_->ensureCapacity(
+      elements * es, (P * blockCapBytes + 1) / 2 + 64);
+  char* work = mesh_->work();
+  auto subInbox = [&](char* base, int s) {
+    return base + static_cast<size_t>(s) * blockCapBytes;
+  };
+  auto blockOff = [&](int b) {
+    return std::min(static_cast<size_t>(b) * perRank, elements) * es;
+  };
+  auto blockLen = [&](int b) {
+    return std::min(static_cast<size_t>(b + 1) * perRank, elements) * es -
+        blockOff(b);
+  };
+
+  {
+    std::vector<hipStream_t> gated{cs_->stream()};
+    for (auto& st : fanout_) {
+      gated.push_back(st->stream());
+    }
+    hipEvent_t ev = nullptr;
+    GA_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+    GA_HIP_CHECK(hipEventRecord(ev, callerStream));
+    for (auto g : gated) {
+      GA_HIP_CHECK(hipStreamWaitEvent(g, ev, 0));
+    }
+    GA_HIP_CHECK(hipEventDestroy(ev));
+  }
+
+  const uint64_t seq = ++seq_;
+
+  // --- scatter: my block d -> d's inbox slot r, all links concurrently ---
+  for (int j = 1; j < P; j++) {
+    const int d = (r + j) % P;
+    auto& st = *fanout_[(j - 1) % fanout_.size()];
+    if (seq > 1) {
+      launchWaitFlagGte(mesh_->flag(fACK_ + d), seq - 1, st.stream());
+    }
+    if (blockLen(d) > 0) {
+      GA_HIP_CHECK(hipMemcpyAsync(
+          subInbox(mesh_->peerInbox(d, 0), r),
+          user + blockOff(d),
+          blockLen(d),
+          hipMemcpyDeviceToDevice,
+          st.stream()));
+    }
+    launchWriteFlag(mesh_->peerFlag(d, fRS_ + r), seq, st.stream());
+  }
+
+  // --- fused local reduction of my block (cs) ---
+  for (int s = 0; s < P; s++) {
+    if (s != r) {
+      launchWaitFlagGte(mesh_->flag(fRS_ + s), seq, cs_->stream());
+    }
+  }
+  if (blockLen(r) > 0) {
+    const void* srcs[8];
+    srcs[0] = user + blockOff(r);
+    int k = 1;
+    for (int s = 0; s < P; s++) {
+      if (s != r) {
+        srcs[k++] = subInbox(mesh_->inbox(0), s);
+      }
+    }
+    launchReduceN(
+        work + blockOff(r), srcs, P, blockLen(r) / es, dtype, op,
+        cs_->stream());
+  }
+  HipEvent reduced(device_);
+  reduced.record(cs_->stream());
+
+  // --- broadcast my reduced block over all links concurrently ---
+  for (int j = 1; j < P; j++) {
+    const int d = (r + j) % P;
+    auto& st = *fanout_[(j - 1) % fanout_.size()];
+    reduced.streamWait(st.stream());
+    if (blockLen(r) > 0) {
+      GA_HIP_CHECK(hipMemcpyAsync(
+          mesh_->peerWork(d) + blockOff(r),
+          work + blockOff(r),
+          blockLen(r),
+          hipMemcpyDeviceToDevice,
+          st.stream()));
+    }
+    launchWriteFlag(mesh_->peerFlag(d, fAG_ + r), seq, st.stream());
+  }
+
+  // --- collect all reduced blocks, stage out, ack everyone (cs) ---
+  for (int s = 0; s < P; s++) {
+    if (s != r) {
+      launchWaitFlagGte(mesh_->flag(fAG_ + s), seq, cs_->stream());
+    }
+  }
+  GA_HIP_CHECK(hipMemcpyAsync(
+      user, work, elements * es, hipMemcpyDeviceToDevice, cs_->stream()));
+  for (int s = 0; s < P; s++) {
+    if (s != r) {
+      launchWriteFlag(mesh_->peerFlag(s, fACK_ + r), seq, cs_->stream());
+    }
+  }
+  doneEvent_->record(cs_->stream());
+  auto timeout = ctx_->getTimeout();
+  watchdogWait(*doneEvent_, *mesh_, timeout, "hip_allreduce_direct (cs)");
+  cs_->synchronize();
+  for (auto& st : fanout_) {
+    HipEvent done(device_);
+    done.record(st->stream());
+    watchdogWait(done, *mesh_, timeout, "hip_allreduce_direct (fanout)");
+    st->synchronize();
+  }
+}
+
+// ===========================================================================
 // HipAllgatherRing
 // ===========================================================================
 

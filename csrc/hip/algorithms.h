@@ -140,6 +140,41 @@ class HipBroadcastOneToAll {
   uint64_t seq_{0};
 };
 
+// Fully-connected "direct" allreduce: every MI355X pair shares an xGMI
+// link, so the optimal schedule is one-shot — each rank scatters its P
+// blocks to their owners over P-1 DIFFERENT links concurrently (fanout
+// streams -> separate SDMA engines), fuses a (P)-way reduction of its
+// own block, then broadcasts the result over all links at once. Wire
+// time ~ 2*S*(P-1)/P divided by ~(P-1) concurrent links, vs the ring's
+// per-link serialization. No reference counterpart: NVSwitch hides this
+// choice; point-to-point xGMI rewards it.
+class HipAllreduceDirect {
+ public:
+  HipAllreduceDirect(
+      std::shared_ptr<Context> ctx,
+      int device,
+      int numStreams = 7);
+
+  void run(
+      void* devPtr,
+      size_t elements,
+      DType dtype,
+      ReduceOp op,
+      hipStream_t callerStream = nullptr);
+
+ private:
+  std::shared_ptr<Context> ctx_;
+  int device_;
+  std::unique_ptr<XgmiMesh> mesh_;
+  std::unique_ptr<HipStream> cs_;
+  std::vector<std::unique_ptr<HipStream>> fanout_;
+  std::unique_ptr<HipEvent> doneEvent_;
+  int fRS_; // [src] scatter-block arrived
+  int fAG_; // [src] reduced-block arrived
+  int fACK_; // [src] peer finished reading my work / inbox reusable
+  uint64_t seq_{0};
+};
+
 // Device-native allgather ring: out (size*inElements) assembled via the
 // same segmented inbox pipeline as the allreduce ring's allgather phase.
 class HipAllgatherRing {

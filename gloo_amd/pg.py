@@ -109,7 +109,21 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
 
     def _ring(self, device):
         if device not in self._hip_ring:
-            self._hip_ring[device] = ga._C.HipAllreduceRing(self._ctx, device)
+            import os
+
+            algo = os.environ.get("GLOO_AMD_ALLREDUCE", "auto")
+            size = self.size()
+            if algo == "ring" or size > 8 or (algo == "auto" and size <= 2):
+                self._hip_ring[device] = ga._C.HipAllreduceRing(
+                    self._ctx, device)
+            elif algo == "hd" and size & (size - 1) == 0:
+                self._hip_ring[device] = ga._C.HipAllreduceHalvingDoubling(
+                    self._ctx, device)
+            else:
+                # fully-connected xGMI: one-shot direct allreduce uses all
+                # links concurrently
+                self._hip_ring[device] = ga._C.HipAllreduceDirect(
+                    self._ctx, device)
         return self._hip_ring[device]
 
     def _bcaster(self, device, root):

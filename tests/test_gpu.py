@@ -230,6 +230,23 @@ def _two_rank_generic(builder_and_check):
     assert not errors, errors[0]
 
 
+def test_hip_allreduce_direct():
+    def fn(ctx, rank):
+        n = 1_000_001  # odd: block tails
+        g = torch.Generator("cpu").manual_seed(rank)
+        x = torch.rand(n, generator=g).cuda()
+        ref = sum(
+            torch.rand(n, generator=torch.Generator("cpu").manual_seed(r))
+            for r in range(2))
+        algo = ga._C.HipAllreduceDirect(ctx, 0)
+        for _ in range(3):
+            y = x.clone()
+            algo.run(y.data_ptr(), n, ga.DType.f32, ga.ReduceOp.sum)
+            assert torch.allclose(y.cpu(), ref, atol=1e-5)
+
+    _two_rank_generic(fn)
+
+
 def test_hip_allgather_ring():
     def fn(ctx, rank):
         n = 1_000_000
