@@ -54,8 +54,10 @@ XgmiMesh::XgmiMesh(
     GA_HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&data_), total));
   }
   flags_ = static_cast<uint64_t*>(allocFineGrained(kNumFlags * 8));
+  // hipMemset (sync form) completes before returning; no device-wide
+  // sync here (it could block on other ranks' in-flight kernels when
+  // several ranks share one GPU).
   GA_HIP_CHECK(hipMemset(flags_, 0, kNumFlags * 8));
-  GA_HIP_CHECK(hipDeviceSynchronize());
   exchange();
 }
 
@@ -140,12 +142,13 @@ void XgmiMesh::ensureCapacity(size_t workCap, size_t inboxCap) {
     // the same collective arguments, so either all grow or none do.
     return;
   }
-  workCap_ = std::max(workCap_, workCap);
+  // Over-allocate on growth so regrowth is rare.
+  workCap_ = std::max(std::max(workCap_ * 2, workCap), size_t(8) << 20);
   inboxCap_ = std::max(inboxCap_, inboxCap);
-  // Every rank must close its mapping of our old buffer BEFORE we free
-  // it: freeing memory a peer still has IPC-mapped leaves the backing
-  // pages pinned and the replacement allocation can fail to export
-  // (hipIpcGetMemHandle: invalid argument).
+  // Close every mapping of the old buffers collectively, but do NOT free
+  // the old allocation: hipIpcGetMemHandle on an allocation recycled
+  // from freed-while-peer-mapped pages fails with invalid argument, so
+  // retired buffers live until the mesh is destroyed.
   {
     BarrierOptions bar(ctx_);
     bar.tag = ctx_->nextSlot();
@@ -157,9 +160,10 @@ void XgmiMesh::ensureCapacity(size_t workCap, size_t inboxCap) {
     bar.tag = ctx_->nextSlot();
     barrier(bar);
   }
+  retired_.push_back(data_);
+  data_ = nullptr;
   {
     std::lock_guard<std::mutex> lock(allocMutex());
-    (void)hipFree(data_);
     GA_HIP_CHECK(hipMalloc(
         reinterpret_cast<void**>(&data_), workCap_ + 2 * inboxCap_));
   }
@@ -174,17 +178,32 @@ int XgmiMesh::allocFlags(int count) {
 }
 
 void XgmiMesh::poisonFlags() {
+  // Release OUR GPU-side waits AND every peer's: a rank that failed may
+  // never write the flags its peers are spinning on, and their stuck
+  // kernels would wedge device-wide syncs.
   std::vector<uint64_t> poison(kNumFlags, ~uint64_t(0) >> 1);
   (void)hipMemcpy(flags_, poison.data(), kNumFlags * 8,
                   hipMemcpyHostToDevice);
+  for (size_t r = 0; r < peerFlags_.size(); r++) {
+    if (static_cast<int>(r) != ctx_->rank && peerFlags_[r] != nullptr) {
+      (void)hipMemcpy(peerFlags_[r], poison.data(), kNumFlags * 8,
+                      hipMemcpyHostToDevice);
+    }
+  }
 }
 
 XgmiMesh::~XgmiMesh() {
-  (void)hipDeviceSynchronize();
+  // No device-wide sync here: the owning algorithm's streams are synced
+  // by their own destructors (declared after the mesh, destroyed first);
+  // a device-wide wait could block on OTHER ranks' in-flight kernels
+  // when several ranks share one GPU.
   releasePeers();
+  std::lock_guard<std::mutex> lock(allocMutex());
   if (data_ != nullptr) {
-    std::lock_guard<std::mutex> lock(allocMutex());
     (void)hipFree(data_);
+  }
+  for (auto* p : retired_) {
+    (void)hipFree(p);
   }
   if (flags_ != nullptr) {
     (void)hipFree(flags_);

@@ -91,6 +91,7 @@ class XgmiMesh {
   size_t inboxCap_;
 
   char* data_{nullptr}; // coarse-grained: work + 2 inboxes
+  std::vector<char*> retired_; // outgrown buffers (freed at destruction)
   uint64_t* flags_{nullptr}; // fine-grained page, peer-writable
   std::vector<char*> peerData_;
   std::vector<uint64_t*> peerFlags_;
