@@ -28,7 +28,7 @@ BIND_OBJS := $(patsubst csrc/%.cc,build/%.o,$(BIND_SRCS))
 
 LDFLAGS := -L$(ROCM)/lib -lamdhip64 -lssl -lcrypto -pthread
 
-all: $(TARGET)
+all: $(TARGET) $(BENCH)
 
 build/%.o: csrc/%.cc
 	@mkdir -p $(dir $@)
