@@ -15,8 +15,8 @@ TARGET := gloo_amd/_C$(EXT_SUFFIX)
 BENCH := bin/gloo_amd_bench
 
 CXXFLAGS := -O3 -g -std=c++17 -fPIC -Wall -Wextra -Wno-unused-parameter \
-  -pthread -Icsrc -I$(ROCM)/include -D__HIP_PLATFORM_AMD__=1
-HIPCCFLAGS := -O3 -std=c++17 -fPIC --offload-arch=gfx950 -Icsrc
+  -MMD -MP -pthread -Icsrc -I$(ROCM)/include -D__HIP_PLATFORM_AMD__=1
+HIPCCFLAGS := -O3 -std=c++17 -fPIC -MMD -MP --offload-arch=gfx950 -Icsrc
 
 CC_SRCS := $(shell find csrc -name '*.cc' ! -path 'csrc/bindings/*' ! -path 'csrc/bench/*')
 HIP_SRCS := $(shell find csrc -name '*.hip' 2>/dev/null)
@@ -43,6 +43,8 @@ $(TARGET): $(CC_OBJS) $(HIP_OBJS) $(BIND_OBJS)
 
 clean:
 	rm -rf build $(TARGET) $(BENCH)
+
+-include $(shell find build -name "*.d" 2>/dev/null)
 
 .PHONY: all clean
 
