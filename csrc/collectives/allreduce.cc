@@ -138,6 +138,57 @@ void ringAllreduce(AllreduceOptions& opts) {
   }
 }
 
+// Latency-optimized path for small payloads: full-buffer recursive
+// doubling (log2 P rounds of concurrent exchange) instead of the ring's
+// 4(P-1) serialized segment hops. Non-power-of-2 sizes fold the extra
+// ranks into partners before/after the exchange.
+void smallAllreduce(AllreduceOptions& opts) {
+  auto& ctx = opts.context;
+  const int P = ctx->size;
+  const int r = ctx->rank;
+  const size_t bytes = opts.elements * opts.elementSize;
+  char* out = static_cast<char*>(opts.outputs[0]);
+  const uint64_t slot = Slot::build(SlotPrefix::kAllreduce, opts.tag);
+  const auto timeout = opts.timeout.count() > 0 ? opts.timeout
+                                                : ctx->getTimeout();
+  int pow2 = 1, steps = 0;
+  while (pow2 * 2 <= P) {
+    pow2 *= 2;
+    steps++;
+  }
+  const int extras = P - pow2;
+
+  auto tmp = makeAligned(bytes + 64);
+  auto outBuf = ctx->createUnboundBuffer(out, bytes);
+  auto tmpBuf = ctx->createUnboundBuffer(tmp.get(), bytes + 64);
+
+  if (r >= pow2) {
+    outBuf->send(r - pow2, slot, 0, bytes);
+    outBuf->waitSend(timeout);
+  } else if (r < extras) {
+    tmpBuf->recv(r + pow2, slot, 0, bytes);
+    tmpBuf->waitRecv(timeout);
+    opts.reduce(out, out, tmp.get(), opts.elements);
+  }
+  if (r < pow2) {
+    for (int t = 0; t < steps; t++) {
+      const int peer = r ^ (1 << t);
+      tmpBuf->recv(peer, slot + 1 + t, 0, bytes);
+      outBuf->send(peer, slot + 1 + t, 0, bytes);
+      tmpBuf->waitRecv(timeout);
+      opts.reduce(out, out, tmp.get(), opts.elements);
+      outBuf->waitSend(timeout);
+    }
+  }
+  if (r < extras) {
+    outBuf->send(r + pow2, slot + 1 + steps, 0, bytes);
+    outBuf->waitSend(timeout);
+  } else if (r >= pow2) {
+    outBuf->recv(r - pow2, slot + 1 + steps, 0, bytes);
+    outBuf->waitRecv(timeout);
+  }
+}
+
 void bcubeAllreduce(AllreduceOptions& opts) {
   auto& ctx = opts.context;
   const int P = ctx->size;
@@ -275,7 +326,12 @@ void allreduce(AllreduceOptions& opts) {
       }
       return p == P;
     }();
-    if (bcubeOk) {
+    static const size_t smallThreshold = static_cast<size_t>(
+        getEnvInt("GLOO_AMD_SMALL_ALLREDUCE", 16384));
+    if (opts.elements * opts.elementSize <= smallThreshold &&
+        opts.algorithm == AllreduceOptions::Algorithm::RING) {
+      smallAllreduce(opts);
+    } else if (bcubeOk) {
       bcubeAllreduce(opts);
     } else {
       ringAllreduce(opts);
