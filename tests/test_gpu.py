@@ -24,6 +24,22 @@ if not GPU:
     pytest.skip("no GPU", allow_module_level=True)
 
 
+def _dev(t):
+    """Pinned stream-local H2D (see _host)."""
+    out = torch.empty(t.shape, dtype=t.dtype, device="cuda")
+    out.copy_(t.pin_memory())
+    return out
+
+
+def _host(t):
+    """Pinned stream-local D2H for threaded multi-rank tests: pageable
+    .cpu() has device-wide sync semantics and can deadlock with the peer
+    rank's in-flight doorbell wait kernels on a shared GPU+process."""
+    out = torch.empty(t.shape, dtype=t.dtype, pin_memory=True)
+    out.copy_(t)
+    return out
+
+
 DTYPES = [
     (torch.float32, ga.DType.f32, 1e-6),
     (torch.float64, ga.DType.f64, 1e-12),
@@ -111,8 +127,8 @@ def _two_rank_device_test(algo_name, elements, dtype=torch.float32,
             ctx.set_timeout(60000)
             torch.cuda.set_device(0)
             g = torch.Generator(device="cpu").manual_seed(rank)
-            x = torch.rand(elements, generator=g, dtype=torch.float32)
-            x = x.to(dtype).cuda()
+            x = _dev(torch.rand(elements, generator=g,
+                               dtype=torch.float32).to(dtype))
             ref_inputs = [
                 torch.rand(elements,
                            generator=torch.Generator("cpu").manual_seed(r),
@@ -130,7 +146,7 @@ def _two_rank_device_test(algo_name, elements, dtype=torch.float32,
             # GPU tests: a device-wide sync would also wait on the OTHER
             # rank's in-flight doorbell wait kernels (deadlock). run() is
             # blocking (its streams are synced).
-            got = x.float().cpu()
+            got = _host(x).float()
             tol = 2e-2 if dtype in (torch.bfloat16, torch.float16) else 1e-4
             assert torch.allclose(got, expect, rtol=tol, atol=tol), (
                 rank, (got - expect).abs().max().item())
@@ -188,12 +204,12 @@ def test_hip_broadcast_one_to_all():
             torch.cuda.set_device(0)
             n = 1_000_000
             if rank == 0:
-                x = torch.arange(n, dtype=torch.float32).cuda()
+                x = torch.arange(n, dtype=torch.float32, device="cuda")
             else:
-                x = torch.zeros(n, dtype=torch.float32).cuda()
+                x = torch.zeros(n, dtype=torch.float32, device="cuda")
             algo = ga._C.HipBroadcastOneToAll(ctx, 0, 0)
             algo.run(x.data_ptr(), x.numel() * 4)
-            assert torch.equal(x.cpu(), torch.arange(n, dtype=torch.float32))
+            assert torch.equal(_host(x), torch.arange(n, dtype=torch.float32))
         except Exception:  # noqa: BLE001
             import traceback
 
@@ -234,7 +250,7 @@ def test_hip_allreduce_direct():
     def fn(ctx, rank):
         n = 1_000_001  # odd: block tails
         g = torch.Generator("cpu").manual_seed(rank)
-        x = torch.rand(n, generator=g).cuda()
+        x = _dev(torch.rand(n, generator=g))
         ref = sum(
             torch.rand(n, generator=torch.Generator("cpu").manual_seed(r))
             for r in range(2))
@@ -242,7 +258,7 @@ def test_hip_allreduce_direct():
         for _ in range(3):
             y = x.clone()
             algo.run(y.data_ptr(), n, ga.DType.f32, ga.ReduceOp.sum)
-            assert torch.allclose(y.cpu(), ref, atol=1e-5)
+            assert torch.allclose(_host(y), ref, atol=1e-5)
 
     _two_rank_generic(fn)
 
@@ -264,7 +280,7 @@ def test_hip_reduce_scatter_ring():
     def fn(ctx, rank):
         n = 500_000
         g = torch.Generator("cpu").manual_seed(rank)
-        inp = torch.rand(2 * n, generator=g).cuda()
+        inp = _dev(torch.rand(2 * n, generator=g))
         out = torch.zeros(n, device="cuda")
         ref = sum(
             torch.rand(2 * n, generator=torch.Generator("cpu").manual_seed(r))
@@ -274,7 +290,7 @@ def test_hip_reduce_scatter_ring():
             algo.run(inp.data_ptr(), out.data_ptr(), n, ga.DType.f32,
                      ga.ReduceOp.sum)
             expect = ref[rank * n:(rank + 1) * n]
-            assert torch.allclose(out.cpu(), expect, atol=1e-5)
+            assert torch.allclose(_host(out), expect, atol=1e-5)
 
     _two_rank_generic(fn)
 
@@ -325,7 +341,7 @@ def test_pg_cuda_collectives():
             pg._reduce_scatter_base(rs_out, rs_in).wait()
             expect = (torch.arange(200_000, dtype=torch.float32) * 2 +
                       1)[rank * 100_000:(rank + 1) * 100_000]
-            assert torch.allclose(rs_out.cpu(), expect)
+            assert torch.allclose(_host(rs_out), expect)
 
             a2a_in = torch.cat([
                 torch.full((50_000,), float(rank * 2 + d), device="cuda")
@@ -340,7 +356,7 @@ def test_pg_cuda_collectives():
                  if rank == 0 else torch.zeros(50_000, device="cuda"))
             pg.broadcast([b]).wait()
             assert torch.allclose(
-                b.cpu(), torch.arange(50_000, dtype=torch.float32))
+                _host(b), torch.arange(50_000, dtype=torch.float32))
         except Exception:  # noqa: BLE001
             import traceback
 
