@@ -884,5 +884,6 @@ PYBIND11_MODULE(_C, m) {
           },
           py::arg("ptr"),
           py::arg("bytes"),
-          py::arg("stream") = 0);
+          py::arg("stream") = 0)
+      .def("debug_flags", &hip::HipBroadcastOneToAll::debugFlags);
 }

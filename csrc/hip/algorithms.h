@@ -128,6 +128,10 @@ class HipBroadcastOneToAll {
 
   void run(void* devPtr, size_t bytes, hipStream_t callerStream = nullptr);
 
+  std::vector<uint64_t> debugFlags() {
+    return mesh_->readFlags();
+  }
+
  private:
   std::shared_ptr<Context> ctx_;
   int device_;

@@ -6,6 +6,7 @@
 
 #include "collectives/collectives.h"
 #include "common/logging.h"
+#include "common/utils.h"
 
 namespace glooamd {
 namespace hip {
@@ -114,6 +115,17 @@ void XgmiMesh::exchange() {
     }
   }
 
+  if (getEnvFlag("GLOO_AMD_FLAG_DEBUG")) {
+    std::string msg = "mesh exchange rank " + std::to_string(ctx_->rank) +
+        " data=" + std::to_string(reinterpret_cast<uintptr_t>(data_)) +
+        " flags=" + std::to_string(reinterpret_cast<uintptr_t>(flags_));
+    for (int r = 0; r < P; r++) {
+      msg += " peer" + std::to_string(r) + "=(" +
+          std::to_string(reinterpret_cast<uintptr_t>(peerData_[r])) + "," +
+          std::to_string(reinterpret_cast<uintptr_t>(peerFlags_[r])) + ")";
+    }
+    GA_ERROR << msg;
+  }
   // Everyone has mapped everyone before first use.
   BarrierOptions bar(ctx_);
   bar.tag = ctx_->nextSlot();
@@ -190,6 +202,12 @@ void XgmiMesh::poisonFlags() {
                       hipMemcpyHostToDevice);
     }
   }
+}
+
+std::vector<uint64_t> XgmiMesh::readFlags(int n) {
+  std::vector<uint64_t> out(n, 0);
+  (void)hipMemcpy(out.data(), flags_, n * 8, hipMemcpyDeviceToHost);
+  return out;
 }
 
 XgmiMesh::~XgmiMesh() {

@@ -81,6 +81,9 @@ class XgmiMesh {
   // run() watchdog on timeout so streams can drain before throwing).
   void poisonFlags();
 
+  // Debug: host copy of the first n local flag values.
+  std::vector<uint64_t> readFlags(int n = 16);
+
  private:
   void exchange(); // allgather + open peer handles
   void releasePeers();

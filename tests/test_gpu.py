@@ -150,7 +150,9 @@ def _two_rank_device_test(algo_name, elements, dtype=torch.float32,
             tol = 2e-2 if dtype in (torch.bfloat16, torch.float16) else 1e-4
             assert torch.allclose(got, expect, rtol=tol, atol=tol), (
                 rank, (got - expect).abs().max().item())
-            # run twice: cross-run seq/flag reuse
+            # run twice: cross-run seq/flag reuse (barrier keeps the other
+            # rank's torch ops out of our spin window on a shared GPU)
+            ga.barrier(ctx, tag=901)
             algo.run(x.data_ptr(), elements, gdt, ga.ReduceOp.sum)
             results[rank] = True
         except Exception:  # noqa: BLE001
@@ -255,10 +257,12 @@ def test_hip_allreduce_direct():
             torch.rand(n, generator=torch.Generator("cpu").manual_seed(r))
             for r in range(2))
         algo = ga._C.HipAllreduceDirect(ctx, 0)
-        for _ in range(3):
-            y = x.clone()
+        y = x.clone()
+        for it in range(3):
+            y.copy_(x)
             algo.run(y.data_ptr(), n, ga.DType.f32, ga.ReduceOp.sum)
             assert torch.allclose(_host(y), ref, atol=1e-5)
+            ga.barrier(ctx, tag=911 + it)
 
     _two_rank_generic(fn)
 
@@ -269,9 +273,10 @@ def test_hip_allgather_ring():
         inp = torch.full((n,), float(rank + 1), device="cuda")
         out = torch.zeros(2 * n, device="cuda")
         algo = ga._C.HipAllgatherRing(ctx, 0)
-        for _ in range(2):
+        for it in range(2):
             algo.run(inp.data_ptr(), out.data_ptr(), n, 4)
             assert torch.all(out[:n] == 1.0) and torch.all(out[n:] == 2.0)
+            ga.barrier(ctx, tag=902 + it)
 
     _two_rank_generic(fn)
 
@@ -286,11 +291,12 @@ def test_hip_reduce_scatter_ring():
             torch.rand(2 * n, generator=torch.Generator("cpu").manual_seed(r))
             for r in range(2))
         algo = ga._C.HipReduceScatterRing(ctx, 0)
-        for _ in range(2):
+        for it in range(2):
             algo.run(inp.data_ptr(), out.data_ptr(), n, ga.DType.f32,
                      ga.ReduceOp.sum)
             expect = ref[rank * n:(rank + 1) * n]
             assert torch.allclose(_host(out), expect, atol=1e-5)
+            ga.barrier(ctx, tag=905 + it)
 
     _two_rank_generic(fn)
 
@@ -304,10 +310,11 @@ def test_hip_alltoall():
         ])
         out = torch.zeros(2 * n, device="cuda")
         algo = ga._C.HipAlltoall(ctx, 0)
-        for _ in range(2):
+        for it in range(2):
             algo.run(inp.data_ptr(), out.data_ptr(), n, 4)
             for s in range(2):
                 assert torch.all(out[s * n:(s + 1) * n] == s * 2 + rank)
+            ga.barrier(ctx, tag=908 + it)
 
     _two_rank_generic(fn)
 
