@@ -160,10 +160,12 @@ def _two_rank_device_test(algo_name, elements, dtype=torch.float32,
 
             errors.append(traceback.format_exc())
 
-    ths = [threading.Thread(target=worker, args=(r,)) for r in range(2)]
+    ths = [threading.Thread(target=worker, args=(r,), daemon=True)
+           for r in range(2)]
     [t.start() for t in ths]
-    [t.join(180) for t in ths]
+    [t.join(100) for t in ths]
     assert not errors, errors[0]
+    assert not any(t.is_alive() for t in ths), "worker hung"
     assert results == {0: True, 1: True}
 
 
@@ -222,9 +224,10 @@ def test_hip_broadcast_one_to_all():
 
             errors.append(traceback.format_exc())
 
-    ths = [threading.Thread(target=worker, args=(r,)) for r in range(2)]
+    ths = [threading.Thread(target=worker, args=(r,), daemon=True)
+           for r in range(2)]
     [t.start() for t in ths]
-    [t.join(120) for t in ths]
+    [t.join(100) for t in ths]
     assert not errors, errors[0]
 
 
@@ -247,10 +250,12 @@ def _two_rank_generic(builder_and_check):
 
             errors.append(traceback.format_exc())
 
-    ths = [threading.Thread(target=worker, args=(r,)) for r in range(2)]
+    ths = [threading.Thread(target=worker, args=(r,), daemon=True)
+           for r in range(2)]
     [t.start() for t in ths]
-    [t.join(180) for t in ths]
+    [t.join(100) for t in ths]
     assert not errors, errors[0]
+    assert not any(t.is_alive() for t in ths), "worker hung"
 
 
 def test_hip_allreduce_direct():
@@ -374,9 +379,10 @@ def test_pg_cuda_collectives():
 
             errors.append(traceback.format_exc())
 
-    ths = [threading.Thread(target=worker, args=(r,)) for r in range(2)]
+    ths = [threading.Thread(target=worker, args=(r,), daemon=True)
+           for r in range(2)]
     [t.start() for t in ths]
-    [t.join(180) for t in ths]
+    [t.join(100) for t in ths]
     assert not errors, errors[0]
 
 
