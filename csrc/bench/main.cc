@@ -507,6 +507,8 @@ void runOne(
 } // namespace
 
 int main(int argc, char** argv) {
+  // One HW queue per pooled stream (see csrc/hip/core.h pooledStream).
+  setenv("GPU_MAX_HW_QUEUES", "8", 0);
   Options o = parse(argc, argv);
   if (o.benchmark.rfind("hip_", 0) == 0) {
     o.gpu = true; // hip_* benchmarks imply the device path

@@ -85,8 +85,8 @@ HipAllreduceRing::HipAllreduceRing(
       inboxCap_(inboxCap == 0 ? kDefaultInboxCap : inboxCap) {
   GA_HIP_CHECK(hipSetDevice(device_));
   mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, inboxCap_);
-  cs_ = std::make_unique<HipStream>(device_, /*highPriority=*/true);
-  ks_ = std::make_unique<HipStream>(device_);
+  cs_ = pooledStream(ctx_.get(), device_, 0);
+  ks_ = pooledStream(ctx_.get(), device_, 1);
   initEvent_ = std::make_unique<HipEvent>(device_);
   doneEvent_ = std::make_unique<HipEvent>(device_);
   fDATA_ = mesh_->allocFlags(2);
@@ -293,8 +293,8 @@ HipAllreduceHalvingDoubling::HipAllreduceHalvingDoubling(
   }
   GA_HIP_CHECK(hipSetDevice(device_));
   mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, inboxCap_);
-  cs_ = std::make_unique<HipStream>(device_, true);
-  ks_ = std::make_unique<HipStream>(device_);
+  cs_ = pooledStream(ctx_.get(), device_, 0);
+  ks_ = pooledStream(ctx_.get(), device_, 1);
   stepEvent_ = std::make_unique<HipEvent>(device_);
   initEvent_ = std::make_unique<HipEvent>(device_);
   doneEvent_ = std::make_unique<HipEvent>(device_);
@@ -465,9 +465,11 @@ HipBroadcastOneToAll::HipBroadcastOneToAll(
     : ctx_(std::move(ctx)), device_(device), root_(root) {
   GA_HIP_CHECK(hipSetDevice(device_));
   mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, 4096);
-  cs_ = std::make_unique<HipStream>(device_);
-  for (int i = 0; i < numStreams; i++) {
-    fanout_.push_back(std::make_unique<HipStream>(device_));
+  cs_ = pooledStream(ctx_.get(), device_, 0);
+  const int nf = std::min(
+      {numStreams, std::max(1, ctx_->size - 1), kStreamPoolSize - 1});
+  for (int i = 0; i < nf; i++) {
+    fanout_.push_back(pooledStream(ctx_.get(), device_, 1 + i));
   }
   fBDATA_ = mesh_->allocFlags(1);
   fBACK_ = mesh_->allocFlags(ctx_->size);
@@ -544,10 +546,11 @@ HipAllreduceDirect::HipAllreduceDirect(
   GA_ENFORCE_LE(ctx_->size, 8, "direct allreduce supports <= 8 ranks");
   GA_HIP_CHECK(hipSetDevice(device_));
   mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, kDefaultInboxCap);
-  cs_ = std::make_unique<HipStream>(device_, true);
-  const int nf = std::min(numStreams, std::max(1, ctx_->size - 1));
+  cs_ = pooledStream(ctx_.get(), device_, 0);
+  const int nf = std::min(
+      {numStreams, std::max(1, ctx_->size - 1), kStreamPoolSize - 1});
   for (int i = 0; i < nf; i++) {
-    fanout_.push_back(std::make_unique<HipStream>(device_));
+    fanout_.push_back(pooledStream(ctx_.get(), device_, 1 + i));
   }
   doneEvent_ = std::make_unique<HipEvent>(device_);
   fRS_ = mesh_->allocFlags(ctx_->size);
@@ -696,8 +699,8 @@ HipAllgatherRing::HipAllgatherRing(
       inboxCap_(inboxCap == 0 ? kDefaultInboxCap : inboxCap) {
   GA_HIP_CHECK(hipSetDevice(device_));
   mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, inboxCap_);
-  cs_ = std::make_unique<HipStream>(device_, true);
-  ks_ = std::make_unique<HipStream>(device_);
+  cs_ = pooledStream(ctx_.get(), device_, 0);
+  ks_ = pooledStream(ctx_.get(), device_, 1);
   initEvent_ = std::make_unique<HipEvent>(device_);
   doneEvent_ = std::make_unique<HipEvent>(device_);
   fDATA_ = mesh_->allocFlags(2);
@@ -813,8 +816,8 @@ HipReduceScatterRing::HipReduceScatterRing(
       inboxCap_(inboxCap == 0 ? kDefaultInboxCap : inboxCap) {
   GA_HIP_CHECK(hipSetDevice(device_));
   mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, inboxCap_);
-  cs_ = std::make_unique<HipStream>(device_, true);
-  ks_ = std::make_unique<HipStream>(device_);
+  cs_ = pooledStream(ctx_.get(), device_, 0);
+  ks_ = pooledStream(ctx_.get(), device_, 1);
   initEvent_ = std::make_unique<HipEvent>(device_);
   doneEvent_ = std::make_unique<HipEvent>(device_);
   fDATA_ = mesh_->allocFlags(2);
@@ -933,9 +936,11 @@ HipAlltoall::HipAlltoall(
     : ctx_(std::move(ctx)), device_(device) {
   GA_HIP_CHECK(hipSetDevice(device_));
   mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, 4096);
-  cs_ = std::make_unique<HipStream>(device_, true);
-  for (int i = 0; i < numStreams; i++) {
-    fanout_.push_back(std::make_unique<HipStream>(device_));
+  cs_ = pooledStream(ctx_.get(), device_, 0);
+  const int nf = std::min(
+      {numStreams, std::max(1, ctx_->size - 1), kStreamPoolSize - 1});
+  for (int i = 0; i < nf; i++) {
+    fanout_.push_back(pooledStream(ctx_.get(), device_, 1 + i));
   }
   fDATA_ = mesh_->allocFlags(ctx_->size);
   fACK_ = mesh_->allocFlags(ctx_->size);

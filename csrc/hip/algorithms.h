@@ -71,8 +71,8 @@ class HipAllreduceRing {
   bool chunked_;
   size_t inboxCap_;
   std::unique_ptr<XgmiMesh> mesh_;
-  std::unique_ptr<HipStream> cs_; // compute: reduces, local copies, acks
-  std::unique_ptr<HipStream> ks_; // copies out to peers + data doorbells
+  HipStream* cs_; // compute: reduces, local copies, acks (pooled)
+  HipStream* ks_; // copies out to peers + data doorbells (pooled)
   std::vector<std::unique_ptr<HipEvent>> events_;
   std::unique_ptr<HipEvent> initEvent_;
   std::unique_ptr<HipEvent> doneEvent_;
@@ -103,8 +103,8 @@ class HipAllreduceHalvingDoubling {
   size_t inboxCap_;
   int log2P_;
   std::unique_ptr<XgmiMesh> mesh_;
-  std::unique_ptr<HipStream> cs_;
-  std::unique_ptr<HipStream> ks_;
+  HipStream* cs_;
+  HipStream* ks_;
   std::unique_ptr<HipEvent> stepEvent_;
   std::unique_ptr<HipEvent> initEvent_;
   std::unique_ptr<HipEvent> doneEvent_;
@@ -137,8 +137,8 @@ class HipBroadcastOneToAll {
   int device_;
   int root_;
   std::unique_ptr<XgmiMesh> mesh_;
-  std::unique_ptr<HipStream> cs_;
-  std::vector<std::unique_ptr<HipStream>> fanout_;
+  HipStream* cs_;
+  std::vector<HipStream*> fanout_;
   int fBDATA_;
   int fBACK_;
   uint64_t seq_{0};
@@ -170,8 +170,8 @@ class HipAllreduceDirect {
   std::shared_ptr<Context> ctx_;
   int device_;
   std::unique_ptr<XgmiMesh> mesh_;
-  std::unique_ptr<HipStream> cs_;
-  std::vector<std::unique_ptr<HipStream>> fanout_;
+  HipStream* cs_;
+  std::vector<HipStream*> fanout_;
   std::unique_ptr<HipEvent> doneEvent_;
   int fRS_; // [src] scatter-block arrived
   int fAG_; // [src] reduced-block arrived
@@ -200,8 +200,8 @@ class HipAllgatherRing {
   int device_;
   size_t inboxCap_;
   std::unique_ptr<XgmiMesh> mesh_;
-  std::unique_ptr<HipStream> cs_;
-  std::unique_ptr<HipStream> ks_;
+  HipStream* cs_;
+  HipStream* ks_;
   std::vector<std::unique_ptr<HipEvent>> events_;
   std::unique_ptr<HipEvent> initEvent_;
   std::unique_ptr<HipEvent> doneEvent_;
@@ -231,8 +231,8 @@ class HipReduceScatterRing {
   int device_;
   size_t inboxCap_;
   std::unique_ptr<XgmiMesh> mesh_;
-  std::unique_ptr<HipStream> cs_;
-  std::unique_ptr<HipStream> ks_;
+  HipStream* cs_;
+  HipStream* ks_;
   std::vector<std::unique_ptr<HipEvent>> events_;
   std::unique_ptr<HipEvent> initEvent_;
   std::unique_ptr<HipEvent> doneEvent_;
@@ -261,8 +261,8 @@ class HipAlltoall {
   std::shared_ptr<Context> ctx_;
   int device_;
   std::unique_ptr<XgmiMesh> mesh_;
-  std::unique_ptr<HipStream> cs_;
-  std::vector<std::unique_ptr<HipStream>> fanout_;
+  HipStream* cs_;
+  std::vector<HipStream*> fanout_;
   int fDATA_; // [src]
   int fACK_; // [src]
   uint64_t seq_{0};

@@ -1,5 +1,9 @@
 #include "hip/core.h"
 
+#include <map>
+#include <memory>
+#include <tuple>
+
 namespace glooamd {
 namespace hip {
 
@@ -59,6 +63,25 @@ void HipStream::synchronize() {
 void HipStream::recordAndWait(HipStream& other) {
   GA_HIP_CHECK(hipEventRecord(event_, stream_));
   GA_HIP_CHECK(hipStreamWaitEvent(other.stream_, event_, 0));
+}
+
+HipStream* pooledStream(const void* key, int device, int idx) {
+  GA_ENFORCE_GE(idx, 0);
+  GA_ENFORCE_LT(idx, kStreamPoolSize, "stream pool exhausted");
+  static std::mutex mu;
+  // Intentionally leaked: destroying streams during static teardown (after
+  // the HIP runtime unloads) is unsafe.
+  static auto* pool = new std::map<std::tuple<const void*, int, int>,
+                                   std::unique_ptr<HipStream>>();
+  std::lock_guard<std::mutex> lock(mu);
+  auto k = std::make_tuple(key, device, idx);
+  auto it = pool->find(k);
+  if (it == pool->end()) {
+    it = pool->emplace(k, std::make_unique<HipStream>(
+                              device, /*highPriority=*/idx == 0))
+             .first;
+  }
+  return it->second.get();
 }
 
 HipEvent::HipEvent(int device) {

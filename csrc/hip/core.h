@@ -63,6 +63,18 @@ class HipStream {
   hipEvent_t event_{nullptr};
 };
 
+// Process-wide stream pool. ROCm maps HIP streams onto at most
+// GPU_MAX_HW_QUEUES (default 4, we request 8) hardware queues per
+// process+device; streams beyond that SHARE a queue, and a doorbell
+// spin kernel then head-of-line-blocks whatever shares its queue —
+// instant deadlock for flag-wait protocols. Capping each communicator
+// at 8 pooled streams (index 0 = high priority) keeps every stream on
+// its own hardware queue. Pool key: one per (communicator, device) —
+// different algorithm objects of one communicator intentionally reuse
+// the same streams (collectives on a communicator are serialized).
+constexpr int kStreamPoolSize = 8;
+HipStream* pooledStream(const void* key, int device, int idx);
+
 class HipEvent {
  public:
   explicit HipEvent(int device);

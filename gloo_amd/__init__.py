@@ -6,6 +6,13 @@ CPU collectives over an epoll TCP transport, and HIP/CDNA4 device
 collectives over xGMI (hipIpcMemHandle peer transport) for GPU tensors.
 """
 
+import os as _os
+
+# 8 hardware queues per process so each pooled stream owns one (the
+# ROCm default of 4 would multiplex streams onto shared queues, where a
+# doorbell spin kernel head-of-line-blocks the stream behind it).
+_os.environ.setdefault("GPU_MAX_HW_QUEUES", "8")
+
 # Load torch (and its bundled HIP runtime) BEFORE our extension: _C's
 # libamdhip64.so.7 dependency then resolves by SONAME onto torch's
 # already-loaded runtime, so one process has exactly one HIP runtime.
