@@ -16,6 +16,12 @@ BENCH := bin/gloo_amd_bench
 
 CXXFLAGS := -O3 -g -std=c++17 -fPIC -Wall -Wextra -Wno-unused-parameter \
   -MMD -MP -pthread -Icsrc -I$(ROCM)/include -D__HIP_PLATFORM_AMD__=1
+
+# make SANITIZE=thread (or address) for instrumented builds
+ifdef SANITIZE
+CXXFLAGS += -fsanitize=$(SANITIZE) -fno-omit-frame-pointer
+LDFLAGS_EXTRA := -fsanitize=$(SANITIZE)
+endif
 HIPCCFLAGS := -O3 -std=c++17 -fPIC -MMD -MP --offload-arch=gfx950 -Icsrc
 
 CC_SRCS := $(shell find csrc -name '*.cc' ! -path 'csrc/bindings/*' ! -path 'csrc/bench/*')
@@ -26,7 +32,7 @@ CC_OBJS := $(patsubst csrc/%.cc,build/%.o,$(CC_SRCS))
 HIP_OBJS := $(patsubst csrc/%.hip,build/%.hip.o,$(HIP_SRCS))
 BIND_OBJS := $(patsubst csrc/%.cc,build/%.o,$(BIND_SRCS))
 
-LDFLAGS := -L$(ROCM)/lib -lamdhip64 -lssl -lcrypto -pthread
+LDFLAGS := -L$(ROCM)/lib -lamdhip64 -lssl -lcrypto -pthread $(LDFLAGS_EXTRA)
 
 all: $(TARGET) $(BENCH)
 

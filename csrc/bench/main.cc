@@ -67,7 +67,7 @@ struct Options {
           "  hip_allreduce_ring, hip_allreduce_ring_chunked,\n"
           "  hip_allreduce_halving_doubling, hip_broadcast_one_to_all,\n"
           "  hip_allgather_ring, hip_reduce_scatter, hip_alltoall,\n"
-          "  hip_allreduce_direct\n");
+          "  hip_allreduce_direct, hip_allreduce_bcube\n");
   exit(1);
 }
 
@@ -339,6 +339,12 @@ Bench makeHipBench(
     };
   } else if (name == "hip_allreduce_direct") {
     auto algo = std::make_shared<hip::HipAllreduceDirect>(ctx, device);
+    DType dt = dtype;
+    b.run = [algo, devPtr, elements, dt] {
+      algo->run(devPtr, elements, dt, ReduceOp::SUM);
+    };
+  } else if (name == "hip_allreduce_bcube") {
+    auto algo = std::make_shared<hip::HipAllreduceBcube>(ctx, device, o.base);
     DType dt = dtype;
     b.run = [algo, devPtr, elements, dt] {
       algo->run(devPtr, elements, dt, ReduceOp::SUM);

@@ -758,6 +758,30 @@ PYBIND11_MODULE(_C, m) {
           py::arg("op") = ReduceOp::SUM,
           py::arg("stream") = 0);
 
+  py::class_<hip::HipAllreduceBcube>(m, "HipAllreduceBcube")
+      .def(
+          py::init([](std::shared_ptr<Context> ctx, int device, int base) {
+            py::gil_scoped_release rel;
+            return std::make_unique<hip::HipAllreduceBcube>(ctx, device,
+                                                            base);
+          }),
+          py::arg("context"),
+          py::arg("device"),
+          py::arg("base") = 0)
+      .def(
+          "run",
+          [](hip::HipAllreduceBcube& a, uintptr_t ptr, size_t n, DType dt,
+             ReduceOp op, uintptr_t stream) {
+            py::gil_scoped_release rel;
+            a.run(reinterpret_cast<void*>(ptr), n, dt, op,
+                  reinterpret_cast<hipStream_t>(stream));
+          },
+          py::arg("ptr"),
+          py::arg("elements"),
+          py::arg("dtype") = DType::F32,
+          py::arg("op") = ReduceOp::SUM,
+          py::arg("stream") = 0);
+
   py::class_<hip::HipAllreduceDirect>(m, "HipAllreduceDirect")
       .def(
           py::init([](std::shared_ptr<Context> ctx, int device,

@@ -687,6 +687,40 @@ void HipAllreduceDirect::run(
 }
 
 // ===========================================================================
+// HipAllreduceBcube
+// ===========================================================================
+
+HipAllreduceBcube::HipAllreduceBcube(
+    std::shared_ptr<Context> ctx,
+    int device,
+    int base) {
+  const int P = ctx->size;
+  const int b = base > 0 ? base : std::max(2, ctx->base);
+  if (P <= 8 && (b >= P || P > 2)) {
+    direct_ = std::make_unique<HipAllreduceDirect>(ctx, device);
+  } else if (b == 2 && (P & (P - 1)) == 0) {
+    hd_ = std::make_unique<HipAllreduceHalvingDoubling>(ctx, device);
+  } else {
+    ring_ = std::make_unique<HipAllreduceRing>(ctx, device);
+  }
+}
+
+void HipAllreduceBcube::run(
+    void* devPtr,
+    size_t elements,
+    DType dtype,
+    ReduceOp op,
+    hipStream_t callerStream) {
+  if (direct_) {
+    direct_->run(devPtr, elements, dtype, op, callerStream);
+  } else if (hd_) {
+    hd_->run(devPtr, elements, dtype, op, callerStream);
+  } else {
+    ring_->run(devPtr, elements, dtype, op, callerStream);
+  }
+}
+
+// ===========================================================================
 // HipAllgatherRing
 // ===========================================================================
 

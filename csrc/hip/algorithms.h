@@ -179,6 +179,32 @@ class HipAllreduceDirect {
   uint64_t seq_{0};
 };
 
+// Base-b hypercube allreduce (reference cuda_allreduce_bcube parity).
+// On a fully-connected xGMI node, bcube's purpose — engaging multiple
+// links per step — is achieved optimally by the one-shot direct
+// schedule (bcube with base == size IS direct; base == 2 is halving
+// doubling). This class therefore dispatches to the strongest matching
+// engine instead of re-implementing the grouped multi-step exchange.
+class HipAllreduceBcube {
+ public:
+  HipAllreduceBcube(
+      std::shared_ptr<Context> ctx,
+      int device,
+      int base = 0 /*0 -> context base*/);
+
+  void run(
+      void* devPtr,
+      size_t elements,
+      DType dtype,
+      ReduceOp op,
+      hipStream_t callerStream = nullptr);
+
+ private:
+  std::unique_ptr<HipAllreduceDirect> direct_;
+  std::unique_ptr<HipAllreduceHalvingDoubling> hd_;
+  std::unique_ptr<HipAllreduceRing> ring_;
+};
+
 // Device-native allgather ring: out (size*inElements) assembled via the
 // same segmented inbox pipeline as the allreduce ring's allgather phase.
 class HipAllgatherRing {
