@@ -78,19 +78,50 @@ HipAllreduceRing::HipAllreduceRing(
     std::shared_ptr<Context> ctx,
     int device,
     bool chunked,
-    size_t inboxCap)
+    size_t inboxCap,
+    int numRings)
     : ctx_(std::move(ctx)),
       device_(device),
       chunked_(chunked),
       inboxCap_(inboxCap == 0 ? kDefaultInboxCap : inboxCap) {
   GA_HIP_CHECK(hipSetDevice(device_));
   mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, inboxCap_);
-  cs_ = pooledStream(ctx_.get(), device_, 0);
-  ks_ = pooledStream(ctx_.get(), device_, 1);
-  initEvent_ = std::make_unique<HipEvent>(device_);
-  doneEvent_ = std::make_unique<HipEvent>(device_);
-  fDATA_ = mesh_->allocFlags(2);
-  fACK_ = mesh_->allocFlags(2);
+
+  // Ring strides coprime to size: each ring's neighbor hop uses a
+  // different xGMI link, so the rings' wire traffic runs concurrently.
+  const int P = ctx_->size;
+  const int maxRings =
+      std::max(1, std::min(numRings == 0 ? 4 : numRings,
+                           kStreamPoolSize / 2));
+  for (int st = 1; st < std::max(P, 2) &&
+       static_cast<int>(strides_.size()) < maxRings;
+       st++) {
+    int a = st, b = P;
+    while (b != 0) {
+      int t = a % b;
+      a = b;
+      b = t;
+    }
+    if (a == 1) { // gcd(st, P) == 1 -> single cycle
+      strides_.push_back(st);
+    }
+  }
+  if (strides_.empty()) {
+    strides_.push_back(1);
+  }
+  const int R = static_cast<int>(strides_.size());
+  cs_.resize(R);
+  ks_.resize(R);
+  events_.resize(R);
+  for (int j = 0; j < R; j++) {
+    cs_[j] = pooledStream(ctx_.get(), device_, 2 * j);
+    ks_[j] = pooledStream(ctx_.get(), device_, 2 * j + 1);
+    initEvent_.push_back(std::make_unique<HipEvent>(device_));
+    doneEvent_.push_back(std::make_unique<HipEvent>(device_));
+    fDATA_.push_back(mesh_->allocFlags(2));
+    fACK_.push_back(mesh_->allocFlags(2));
+  }
+  lastAckPerRing_.assign(R, {0, 0});
 }
 
 void HipAllreduceRing::run(
@@ -105,7 +136,20 @@ void HipAllreduceRing::run(
   if (ctx_->size == 1 || elements == 0) {
     return;
   }
-  gateStreams(callerStream, {cs_->stream(), ks_->stream()});
+  {
+    std::vector<hipStream_t> gated;
+    for (size_t j = 0; j < cs_.size(); j++) {
+      gated.push_back(cs_[j]->stream());
+      gated.push_back(ks_[j]->stream());
+    }
+    hipEvent_t ev = nullptr;
+    GA_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+    GA_HIP_CHECK(hipEventRecord(ev, callerStream));
+    for (auto g : gated) {
+      GA_HIP_CHECK(hipStreamWaitEvent(g, ev, 0));
+    }
+    GA_HIP_CHECK(hipEventDestroy(ev));
+  }
   if (bytes < onDeviceThreshold()) {
     runHostStaged(static_cast<char*>(devPtr), bytes, elements, dtype, op);
   } else {
@@ -127,8 +171,8 @@ void HipAllreduceRing::runHostStaged(
     hostStageCap_ = bytes;
   }
   GA_HIP_CHECK(hipMemcpyAsync(
-      hostStage_, buf, bytes, hipMemcpyDeviceToHost, cs_->stream()));
-  cs_->synchronize();
+      hostStage_, buf, bytes, hipMemcpyDeviceToHost, cs_[0]->stream()));
+  cs_[0]->synchronize();
   AllreduceOptions opts(ctx_);
   opts.outputs = {hostStage_};
   opts.elements = n;
@@ -137,31 +181,43 @@ void HipAllreduceRing::runHostStaged(
   opts.tag = 0x7fff0000u; // reserved tag band for hip host staging
   allreduce(opts);
   GA_HIP_CHECK(hipMemcpyAsync(
-      buf, hostStage_, bytes, hipMemcpyHostToDevice, cs_->stream()));
-  cs_->synchronize();
+      buf, hostStage_, bytes, hipMemcpyHostToDevice, cs_[0]->stream()));
+  cs_[0]->synchronize();
 }
 
-void HipAllreduceRing::runDevice(
-    char* buf,
-    size_t bytes,
+void HipAllreduceRing::enqueueRing(
+    int j,
+    char* work,
+    size_t elemOff,
     size_t n,
+    size_t es,
     DType dt,
     ReduceOp op) {
   const int P = ctx_->size;
   const int r = ctx_->rank;
-  const size_t es = bytes / n;
-  const int right = (r + 1) % P;
-  const int left = (r - 1 + P) % P;
+  const int stride = strides_[j];
+  const int right = (r + stride) % P;
+  const int left = (r - stride + P) % P;
+  // Virtual position of this rank along the stride-ring.
+  int vr = 0;
+  for (int t = 0; t < P; t++) {
+    if ((t * stride) % P == r) {
+      vr = t;
+      break;
+    }
+  }
 
+  const int R = static_cast<int>(strides_.size());
+  const size_t subCap = inboxCap_ / R; // per-(ring,parity) inbox bytes
   const size_t perRank = (n + P - 1) / P;
   const int S = chunked_
       ? std::max<int>(
-            2,
-            static_cast<int>(
-                (perRank * es + inboxCap_ - 1) / inboxCap_))
-      : 1;
-  const size_t segCapBytes = ((perRank + S - 1) / S) * es;
-  mesh_->ensureCapacity(bytes, segCapBytes);
+            2, static_cast<int>((perRank * es + subCap - 1) / subCap))
+      : std::max<int>(
+            1, static_cast<int>((perRank * es + subCap - 1) / subCap));
+  auto ringInbox = [&](char* base, int par) {
+    return base + (static_cast<size_t>(j) * 2 + par) * subCap;
+  };
 
   struct Step {
     Seg send;
@@ -172,102 +228,131 @@ void HipAllreduceRing::runDevice(
   std::vector<Step> steps;
   steps.reserve(2 * K1);
   for (int k = 0; k < K1; k++) {
-    int i = k / S, s = k % S;
-    steps.push_back({segmentOf(n, P, (r - i + P) % P, s, S),
-                     segmentOf(n, P, (r - i - 1 + 2 * P) % P, s, S),
+    int i = k / S, q = k % S;
+    steps.push_back({segmentOf(n, P, (vr - i + P) % P, q, S),
+                     segmentOf(n, P, (vr - i - 1 + 2 * P) % P, q, S),
                      true});
   }
   for (int k = 0; k < K1; k++) {
-    int i = k / S, s = k % S;
-    steps.push_back({segmentOf(n, P, (r + 1 - i + P) % P, s, S),
-                     segmentOf(n, P, (r - i + P) % P, s, S),
+    int i = k / S, q = k % S;
+    steps.push_back({segmentOf(n, P, (vr + 1 - i + P) % P, q, S),
+                     segmentOf(n, P, (vr - i + P) % P, q, S),
                      false});
   }
   const int K = static_cast<int>(steps.size());
+  ringSteps_ = std::max<uint64_t>(ringSteps_, K);
 
   const int pool = S + 2;
-  while (static_cast<int>(events_.size()) < pool) {
-    events_.push_back(std::make_unique<HipEvent>(device_));
+  auto& evs = events_[j];
+  while (static_cast<int>(evs.size()) < pool) {
+    evs.push_back(std::make_unique<HipEvent>(device_));
   }
-
-  char* work = mesh_->work();
-  auto csm = cs_->stream();
-  auto ksm = ks_->stream();
-
-  // Stage in.
-  GA_HIP_CHECK(
-      hipMemcpyAsync(work, buf, bytes, hipMemcpyDeviceToDevice, csm));
-  initEvent_->record(csm);
-
+  auto csm = cs_[j]->stream();
+  auto ksm = ks_[j]->stream();
+  char* base = work + elemOff * es;
   auto seqOf = [&](int k) { return seqBase_ + k + 1; };
 
   for (int k = 0; k < K; k++) {
     const Step& st = steps[k];
     const int par = k & 1;
-
-    // --- sender side (ks): work[send] -> right's inbox[par] ---
-    // Inbox-reuse gate: right must have consumed what we last put in
-    // inbox[par] (covers both the k-2 step of this run and the tail of
-    // the previous run).
-    const uint64_t prevAck = (k >= 2) ? seqOf(k - 2)
-                                      : (seqBase_ >= 2 ? seqBase_ - 2 + par + 1
-                                                       : 0);
+    const uint64_t prevAck =
+        (k >= 2) ? seqOf(k - 2) : lastAckPerRing_[j][par];
     if (prevAck > 0) {
-      launchWaitFlagGte(mesh_->flag(fACK_ + par), prevAck, ksm);
+      launchWaitFlagGte(mesh_->flag(fACK_[j] + par), prevAck, ksm);
     }
     if (k >= S) {
-      events_[(k - S) % pool]->streamWait(ksm);
+      evs[(k - S) % pool]->streamWait(ksm);
     } else {
-      initEvent_->streamWait(ksm);
+      initEvent_[j]->streamWait(ksm);
     }
     if (st.send.len > 0) {
       GA_HIP_CHECK(hipMemcpyAsync(
-          mesh_->peerInbox(right, par),
-          work + st.send.off * es,
+          ringInbox(mesh_->peerInbox(right, 0), par),
+          base + st.send.off * es,
           st.send.len * es,
           hipMemcpyDeviceToDevice,
           ksm));
     }
-    launchWriteFlag(mesh_->peerFlag(right, fDATA_ + par), seqOf(k), ksm);
+    launchWriteFlag(mesh_->peerFlag(right, fDATA_[j] + par), seqOf(k), ksm);
 
-    // --- receiver side (cs): inbox[par] -> work[recv] ---
-    launchWaitFlagGte(mesh_->flag(fDATA_ + par), seqOf(k), csm);
+    launchWaitFlagGte(mesh_->flag(fDATA_[j] + par), seqOf(k), csm);
     if (st.recv.len > 0) {
       if (st.reduceStep) {
         launchReduce2(
-            work + st.recv.off * es,
-            work + st.recv.off * es,
-            mesh_->inbox(par),
+            base + st.recv.off * es,
+            base + st.recv.off * es,
+            ringInbox(mesh_->inbox(0), par),
             st.recv.len,
             dt,
             op,
             csm);
       } else {
         GA_HIP_CHECK(hipMemcpyAsync(
-            work + st.recv.off * es,
-            mesh_->inbox(par),
+            base + st.recv.off * es,
+            ringInbox(mesh_->inbox(0), par),
             st.recv.len * es,
             hipMemcpyDeviceToDevice,
             csm));
       }
     }
-    launchWriteFlag(mesh_->peerFlag(left, fACK_ + par), seqOf(k), csm);
-    events_[k % pool]->record(csm);
+    launchWriteFlag(mesh_->peerFlag(left, fACK_[j] + par), seqOf(k), csm);
+    evs[k % pool]->record(csm);
+  }
+  for (int k = std::max(0, K - 2); k < K; k++) {
+    lastAckPerRing_[j][k & 1] = seqOf(k);
+  }
+}
+
+void HipAllreduceRing::runDevice(
+    char* buf,
+    size_t bytes,
+    size_t n,
+    DType dt,
+    ReduceOp op) {
+  const size_t es = bytes / n;
+  mesh_->ensureCapacity(bytes, inboxCap_);
+  char* work = mesh_->work();
+  const int R = static_cast<int>(strides_.size());
+
+  // Stage in once; every ring is ordered after it.
+  GA_HIP_CHECK(hipMemcpyAsync(
+      work, buf, bytes, hipMemcpyDeviceToDevice, cs_[0]->stream()));
+  HipEvent staged(device_);
+  staged.record(cs_[0]->stream());
+  for (int j = 0; j < R; j++) {
+    staged.streamWait(cs_[j]->stream());
+    staged.streamWait(ks_[j]->stream());
+    initEvent_[j]->record(cs_[j]->stream());
   }
 
-  // Stage out.
-  GA_HIP_CHECK(
-      hipMemcpyAsync(buf, work, bytes, hipMemcpyDeviceToDevice, csm));
-  doneEvent_->record(csm);
+  // Partition the buffer across the rings and enqueue each schedule.
+  ringSteps_ = 0;
+  for (int j = 0; j < R; j++) {
+    Seg part = subspanOf({0, n}, j, R);
+    if (part.len > 0) {
+      enqueueRing(j, work, part.off, part.len, es, dt, op);
+    }
+  }
+
+  // Copy-out after every ring's compute stream finished its part.
+  for (int j = 1; j < R; j++) {
+    doneEvent_[j]->record(cs_[j]->stream());
+    doneEvent_[j]->streamWait(cs_[0]->stream());
+  }
+  GA_HIP_CHECK(hipMemcpyAsync(
+      buf, work, bytes, hipMemcpyDeviceToDevice, cs_[0]->stream()));
+  doneEvent_[0]->record(cs_[0]->stream());
 
   auto timeout = ctx_->getTimeout();
-  watchdogWait(*doneEvent_, *mesh_, timeout, "hip_allreduce_ring (cs)");
-  // ks tail: the final data-flag writes; drains once peers ack.
-  initEvent_->record(ksm);
-  watchdogWait(*initEvent_, *mesh_, timeout, "hip_allreduce_ring (ks)");
-  cs_->synchronize();
-  ks_->synchronize();
-  seqBase_ += K;
+  watchdogWait(*doneEvent_[0], *mesh_, timeout, "hip_allreduce_ring (cs)");
+  for (int j = 0; j < R; j++) {
+    HipEvent drain(device_);
+    drain.record(ks_[j]->stream());
+    watchdogWait(drain, *mesh_, timeout, "hip_allreduce_ring (ks)");
+    ks_[j]->synchronize();
+    cs_[j]->synchronize();
+  }
+  seqBase_ += ringSteps_;
 }
 
 // ===========================================================================

@@ -28,6 +28,7 @@
 // with MI355X-tuned default (256 KiB, env GLOO_AMD_DEVICE_THRESHOLD).
 #pragma once
 
+#include <array>
 #include <memory>
 #include <vector>
 
@@ -42,11 +43,16 @@ size_t onDeviceThreshold(); // bytes; below this, host-staged path
 
 class HipAllreduceRing {
  public:
+  // numRings 0 -> auto: partition the buffer across rings whose neighbor
+  // strides are coprime to size (8 ranks -> strides 1,3,5,7), so several
+  // xGMI links carry traffic concurrently instead of one. Each ring gets
+  // its own flag slots, inbox region and stream pair from the pool.
   HipAllreduceRing(
       std::shared_ptr<Context> ctx,
       int device,
       bool chunked = true,
-      size_t inboxCap = 0 /*0 -> default 4 MiB*/);
+      size_t inboxCap = 0 /*0 -> default 4 MiB*/,
+      int numRings = 0);
 
   // In-place allreduce of a device buffer. Blocking (streams synced).
   // callerStream: the stream on which the caller produced devPtr (e.g.
@@ -65,19 +71,34 @@ class HipAllreduceRing {
  private:
   void runDevice(char* buf, size_t bytes, size_t n, DType dt, ReduceOp op);
   void runHostStaged(char* buf, size_t bytes, size_t n, DType dt, ReduceOp op);
+  // Enqueue one ring's full schedule (reduce-scatter + allgather over
+  // the element range [elemOff, elemOff+elems) with neighbor stride
+  // strides_[j]) on that ring's stream pair.
+  void enqueueRing(
+      int j,
+      char* work,
+      size_t elemOff,
+      size_t elems,
+      size_t es,
+      DType dt,
+      ReduceOp op);
 
   std::shared_ptr<Context> ctx_;
   int device_;
   bool chunked_;
   size_t inboxCap_;
   std::unique_ptr<XgmiMesh> mesh_;
-  HipStream* cs_; // compute: reduces, local copies, acks (pooled)
-  HipStream* ks_; // copies out to peers + data doorbells (pooled)
-  std::vector<std::unique_ptr<HipEvent>> events_;
-  std::unique_ptr<HipEvent> initEvent_;
-  std::unique_ptr<HipEvent> doneEvent_;
-  int fDATA_, fACK_; // flag indices (x2 parity each)
+  // per-ring resources (ring j uses pooled streams 2j / 2j+1)
+  std::vector<int> strides_;
+  std::vector<HipStream*> cs_;
+  std::vector<HipStream*> ks_;
+  std::vector<std::vector<std::unique_ptr<HipEvent>>> events_;
+  std::vector<std::unique_ptr<HipEvent>> initEvent_;
+  std::vector<std::unique_ptr<HipEvent>> doneEvent_;
+  std::vector<int> fDATA_, fACK_; // flag indices (x2 parity each, per ring)
+  std::vector<std::array<uint64_t, 2>> lastAckPerRing_;
   uint64_t seqBase_{0};
+  uint64_t ringSteps_{0}; // steps per ring in the last run
   // pinned staging for the host path
   void* hostStage_{nullptr};
   size_t hostStageCap_{0};

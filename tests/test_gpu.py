@@ -391,6 +391,28 @@ def test_host_staged_small():
     _two_rank_device_test("ring", 1000)  # 4 KB << 256 KB threshold
 
 
+def test_cross_process_ipc_4rank():
+    """Four processes on one GPU: exercises the multi-ring (stride 1,3)
+    chunked allreduce over IPC."""
+    worker = os.path.join(os.path.dirname(__file__), "ipc_worker.py")
+    tmp = "/tmp/ga_ipc4_%d" % os.getpid()
+    os.makedirs(tmp, exist_ok=True)
+    env = dict(os.environ)
+    env["HSA_ENABLE_IPC_MODE_LEGACY"] = "0"
+    procs = [
+        subprocess.Popen(
+            [sys.executable, worker, str(r), "4", tmp],
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, env=env)
+        for r in range(4)
+    ]
+    outs = []
+    for p in procs:
+        out, _ = p.communicate(timeout=300)
+        outs.append(out.decode())
+    assert all(p.returncode == 0 for p in procs), "\n".join(outs)
+    assert all("IPC-OK" in o for o in outs), "\n".join(outs)
+
+
 def test_cross_process_ipc():
     """Two real processes on one GPU: hipIpcMemHandle + doorbells."""
     worker = os.path.join(os.path.dirname(__file__), "ipc_worker.py")
