@@ -65,3 +65,10 @@ $(BENCH): $(CC_OBJS) $(HIP_OBJS) $(BENCH_OBJS)
 build/bench/%.o: csrc/bench/%.cc
 	@mkdir -p $(dir $@)
 	$(CXX) $(CXXFLAGS) -c $< -o $@
+
+examples: $(CC_OBJS) $(HIP_OBJS)
+	@mkdir -p bin
+	$(CXX) $(CXXFLAGS) examples/example_allreduce.cc $(CC_OBJS) $(HIP_OBJS) -o bin/example_allreduce $(LDFLAGS)
+	$(CXX) $(CXXFLAGS) examples/example_v2.cc $(CC_OBJS) $(HIP_OBJS) -o bin/example_v2 $(LDFLAGS)
+
+.PHONY: examples

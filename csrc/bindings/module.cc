@@ -18,6 +18,7 @@
 #include "context.h"
 #include "hip/algorithms.h"
 #include "hip/kernels.h"
+#include "rendezvous/context_factory.h"
 #include "rendezvous/stores.h"
 #include "transport/tcp/context.h"
 #include "transport/tcp/device.h"
@@ -334,6 +335,16 @@ PYBIND11_MODULE(_C, m) {
                                            size);
           },
           py::keep_alive<0, 1>());
+
+  py::class_<ContextFactory>(m, "ContextFactory")
+      .def(py::init<std::shared_ptr<Context>>(), py::arg("backing_context"))
+      .def(
+          "make_context",
+          [](ContextFactory& f, std::shared_ptr<transport::Device> dev) {
+            py::gil_scoped_release rel;
+            return f.makeContext(dev);
+          },
+          py::arg("device"));
 
   // --- collectives ----------------------------------------------------------
   m.def(
