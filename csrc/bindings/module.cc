@@ -180,13 +180,15 @@ PYBIND11_MODULE(_C, m) {
 
   m.def(
       "create_tcp_device",
-      [](const std::string& hostname) {
+      [](const std::string& hostname, bool useLibuv) {
         tcp::TcpAttr attr;
         attr.hostname = hostname;
+        attr.useLibuv = useLibuv;
         return std::static_pointer_cast<transport::Device>(
             tcp::createTcpDevice(attr));
       },
-      py::arg("hostname") = std::string());
+      py::arg("hostname") = std::string(),
+      py::arg("use_libuv") = false);
 
   m.def(
       "create_tls_device",

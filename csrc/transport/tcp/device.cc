@@ -11,6 +11,7 @@
 
 #include "common/logging.h"
 #include "transport/tcp/context.h"
+#include "transport/tcp/uv_loop.h"
 
 namespace glooamd {
 namespace tcp {
@@ -37,6 +38,11 @@ std::shared_ptr<TcpDevice> createTcpDevice(const TcpAttr& attr) {
 }
 
 TcpDevice::TcpDevice(const TcpAttr& attr) {
+  if (attr.useLibuv) {
+    loop_ = makeUvLoop();
+  } else {
+    loop_ = std::make_unique<EpollLoop>();
+  }
   std::string host = attr.hostname.empty() ? "127.0.0.1" : attr.hostname;
 
   struct addrinfo hints;
@@ -79,17 +85,17 @@ TcpDevice::TcpDevice(const TcpAttr& attr) {
   GA_ENFORCE_EQ(rv, 0, "getsockname: ", strerror(errno));
 
   setNonBlocking(listenFd_);
-  loop_.registerDescriptor(listenFd_, EPOLLIN, this);
+  loop_->registerDescriptor(listenFd_, EPOLLIN, this);
 }
 
 TcpDevice::~TcpDevice() {
   if (listenFd_ >= 0) {
-    loop_.unregisterDescriptor(listenFd_);
+    loop_->unregisterDescriptor(listenFd_);
   }
   // Drop any still-pending accepted connections.
-  loop_.defer([this] {
+  loop_->defer([this] {
     for (auto& kv : readers_) {
-      loop_.unregisterDescriptor(kv.first);
+      loop_->unregisterDescriptor(kv.first);
       close(kv.first);
     }
     readers_.clear();
@@ -137,7 +143,7 @@ void TcpDevice::handleEvents(uint32_t /*events*/) {
       std::lock_guard<std::mutex> lock(mu_);
       readers_[fd] = std::move(reader);
     }
-    loop_.registerDescriptor(fd, EPOLLIN, raw);
+    loop_->registerDescriptor(fd, EPOLLIN, raw);
   }
 }
 

@@ -26,6 +26,9 @@ struct TcpAttr {
   // Interface to bind (optional, unused when hostname given).
   std::string iface;
   int port = 0; // 0 -> ephemeral
+  // Drive the transport with a libuv loop instead of raw epoll
+  // (reference uv-transport parity; same wire protocol).
+  bool useLibuv = false;
 };
 
 class TcpDevice;
@@ -44,7 +47,7 @@ class TcpDevice : public transport::Device,
       override;
 
   Loop& loop() {
-    return loop_;
+    return *loop_;
   }
 
   uint64_t nextSeq() {
@@ -88,7 +91,8 @@ class TcpDevice : public transport::Device,
   std::unordered_map<uint64_t, int> arrived_; // seq -> connected fd
   std::unordered_map<int, std::unique_ptr<SeqReader>> readers_;
 
-  Loop loop_; // declared last: destroyed first, joining the thread
+  // declared last: destroyed first, joining the loop thread
+  std::unique_ptr<Loop> loop_;
 };
 
 // Socket helpers shared by pair/device.

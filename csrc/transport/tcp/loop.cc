@@ -12,7 +12,7 @@
 namespace glooamd {
 namespace tcp {
 
-Loop::Loop() {
+EpollLoop::EpollLoop() {
   epfd_ = epoll_create1(EPOLL_CLOEXEC);
   GA_ENFORCE_GE(epfd_, 0, "epoll_create1: ", strerror(errno));
   evfd_ = eventfd(0, EFD_NONBLOCK | EFD_CLOEXEC);
@@ -28,7 +28,7 @@ Loop::Loop() {
   threadId_ = thread_.get_id();
 }
 
-Loop::~Loop() {
+EpollLoop::~EpollLoop() {
   done_ = true;
   uint64_t one = 1;
   (void)!write(evfd_, &one, sizeof(one));
@@ -39,7 +39,7 @@ Loop::~Loop() {
   close(epfd_);
 }
 
-void Loop::registerDescriptor(int fd, uint32_t events, Handler* h) {
+void EpollLoop::registerDescriptor(int fd, uint32_t events, Handler* h) {
   struct epoll_event ev;
   std::memset(&ev, 0, sizeof(ev));
   ev.events = events;
@@ -48,7 +48,7 @@ void Loop::registerDescriptor(int fd, uint32_t events, Handler* h) {
   GA_ENFORCE_EQ(rv, 0, "epoll_ctl ADD: ", strerror(errno));
 }
 
-void Loop::modifyDescriptor(int fd, uint32_t events, Handler* h) {
+void EpollLoop::modifyDescriptor(int fd, uint32_t events, Handler* h) {
   struct epoll_event ev;
   std::memset(&ev, 0, sizeof(ev));
   ev.events = events;
@@ -57,12 +57,12 @@ void Loop::modifyDescriptor(int fd, uint32_t events, Handler* h) {
   GA_ENFORCE_EQ(rv, 0, "epoll_ctl MOD: ", strerror(errno));
 }
 
-void Loop::unregisterNoWait(int fd) {
+void EpollLoop::unregisterNoWait(int fd) {
   int rv = epoll_ctl(epfd_, EPOLL_CTL_DEL, fd, nullptr);
   GA_ENFORCE_EQ(rv, 0, "epoll_ctl DEL: ", strerror(errno));
 }
 
-void Loop::unregisterDescriptor(int fd) {
+void EpollLoop::unregisterDescriptor(int fd) {
   int rv = epoll_ctl(epfd_, EPOLL_CTL_DEL, fd, nullptr);
   GA_ENFORCE_EQ(rv, 0, "epoll_ctl DEL: ", strerror(errno));
   // If called off the loop thread, wait until the loop has moved past the
@@ -72,7 +72,7 @@ void Loop::unregisterDescriptor(int fd) {
   }
 }
 
-void Loop::waitForTick() {
+void EpollLoop::waitForTick() {
   std::unique_lock<std::mutex> lock(mu_);
   uint64_t current = tick_;
   uint64_t one = 1;
@@ -80,7 +80,7 @@ void Loop::waitForTick() {
   cv_.wait(lock, [&] { return tick_ > current || done_.load(); });
 }
 
-void Loop::defer(std::function<void()> fn) {
+void EpollLoop::defer(std::function<void()> fn) {
   {
     std::lock_guard<std::mutex> lock(mu_);
     deferred_.push_back(std::move(fn));
@@ -89,7 +89,7 @@ void Loop::defer(std::function<void()> fn) {
   (void)!write(evfd_, &one, sizeof(one));
 }
 
-void Loop::run() {
+void EpollLoop::run() {
   std::array<struct epoll_event, 64> events;
   while (!done_.load()) {
     int n = epoll_wait(epfd_, events.data(), events.size(), 100);

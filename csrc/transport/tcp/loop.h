@@ -1,8 +1,9 @@
-// Epoll event loop: one thread per tcp Device.
+// Event loops driving the tcp transport.
 // Capability parity with reference gloo/transport/tcp/loop.cc:103-220
-// (epoll_wait + handler dispatch + deferred-function wakeup + safe
-// off-thread unregister). Re-designed: eventfd wakeup instead of a pipe,
-// tick-counter barrier for teardown safety.
+// (epoll) and gloo/transport/uv/ (libuv): the SAME pair protocol runs
+// over either loop; `Loop` is the interface, `EpollLoop` the primary
+// Linux implementation, `UvLoop` (csrc/transport/tcp/uv_loop.*) the
+// libuv-driven one.
 #pragma once
 
 #include <atomic>
@@ -24,25 +25,38 @@ class Handler {
 
 class Loop {
  public:
-  Loop();
-  ~Loop();
+  virtual ~Loop() = default;
 
   // Register fd with EPOLLIN/EPOLLOUT mask; handler invoked on loop thread.
-  void registerDescriptor(int fd, uint32_t events, Handler* h);
-  void modifyDescriptor(int fd, uint32_t events, Handler* h);
+  virtual void registerDescriptor(int fd, uint32_t events, Handler* h) = 0;
+  virtual void modifyDescriptor(int fd, uint32_t events, Handler* h) = 0;
 
   // After return, the handler for fd is guaranteed not running and will
-  // never run again (blocks one dispatch tick when called off-thread).
-  void unregisterDescriptor(int fd);
+  // never run again (may block one dispatch tick when called off-thread).
+  virtual void unregisterDescriptor(int fd) = 0;
 
-  // epoll DEL only — no tick wait. Safe to call while holding locks the
-  // loop thread may contend on; pair with defer() for the close().
-  void unregisterNoWait(int fd);
+  // Detach only — no tick wait. Safe while holding locks the loop thread
+  // may contend on; pair with defer() for the close().
+  virtual void unregisterNoWait(int fd) = 0;
 
   // Run fn on the loop thread soon.
-  void defer(std::function<void()> fn);
+  virtual void defer(std::function<void()> fn) = 0;
 
-  bool inLoopThread() const {
+  virtual bool inLoopThread() const = 0;
+};
+
+class EpollLoop : public Loop {
+ public:
+  EpollLoop();
+  ~EpollLoop() override;
+
+  void registerDescriptor(int fd, uint32_t events, Handler* h) override;
+  void modifyDescriptor(int fd, uint32_t events, Handler* h) override;
+  void unregisterDescriptor(int fd) override;
+  void unregisterNoWait(int fd) override;
+  void defer(std::function<void()> fn) override;
+
+  bool inLoopThread() const override {
     return std::this_thread::get_id() == threadId_;
   }
 
