@@ -14,6 +14,8 @@
 //   [--gpu] [--half-precision] [--iteration-time-ms 2000]
 #include <getopt.h>
 
+#include <thread>
+
 #include <cstdio>
 #include <cstring>
 #include <functional>
@@ -29,6 +31,7 @@
 #include "context.h"
 #include "hip/algorithms.h"
 #include "hip/kernels.h"
+#include "rendezvous/context_factory.h"
 #include "rendezvous/stores.h"
 #include "transport/tcp/device.h"
 
@@ -46,6 +49,7 @@ struct Options {
   long iterationTimeMs = 2000;
   int warmupIters = 5;
   int inputs = 1;
+  int threads = 1;
   bool gpu = false;
   bool halfPrecision = false;
   int base = 2;
@@ -83,6 +87,7 @@ Options parse(int argc, char** argv) {
       {"iteration-time-ms", required_argument, nullptr, 'i'},
       {"warmup-iters", required_argument, nullptr, 'w'},
       {"inputs", required_argument, nullptr, 'n'},
+      {"threads", required_argument, nullptr, 'T'},
       {"gpu", no_argument, nullptr, 'g'},
       {"half-precision", no_argument, nullptr, 'h'},
       {"base", required_argument, nullptr, 'B'},
@@ -118,6 +123,9 @@ Options parse(int argc, char** argv) {
         break;
       case 'n':
         o.inputs = atoi(optarg);
+        break;
+      case 'T':
+        o.threads = atoi(optarg);
         break;
       case 'g':
         o.gpu = true;
@@ -541,18 +549,45 @@ int main(int argc, char** argv) {
   auto ctx = std::make_shared<Context>(o.rank, o.size, o.base);
   ctx->connectFullMesh(*store, device);
 
+  // Per-thread contexts via store-less re-rendezvous (reference
+  // Runner/ContextFactory parity, runner.cc:151-156). Thread 0 reports.
+  std::vector<std::shared_ptr<Context>> ctxs{ctx};
+  if (o.threads > 1) {
+    ContextFactory factory(ctx);
+    for (int t = 1; t < o.threads; t++) {
+      ctxs.push_back(factory.makeContext(device));
+    }
+  }
+
   if (o.rank == 0) {
     printHeader(o);
   }
+  auto runAll = [&](size_t n) {
+    if (o.threads == 1) {
+      runOne(o, ctx, n);
+      return;
+    }
+    std::vector<std::thread> ths;
+    for (int t = 0; t < o.threads; t++) {
+      Options ot = o;
+      if (t != 0) {
+        ot.rank = -1; // only thread 0 of rank 0 prints
+      }
+      ths.emplace_back([ot, &ctxs, t, n] { runOne(ot, ctxs[t], n); });
+    }
+    for (auto& th : ths) {
+      th.join();
+    }
+  };
   if (o.elements >= 0) {
-    runOne(o, ctx, o.elements);
+    runAll(o.elements);
   } else {
     // Reference README sweep: 1 ... 5,000,000.
     const long sweep[] = {1, 2, 5, 10, 20, 50, 100, 200, 500,
                           1000, 2000, 5000, 10000, 20000, 50000,
                           100000, 200000, 500000, 1000000, 2000000, 5000000};
     for (long n : sweep) {
-      runOne(o, ctx, n);
+      runAll(n);
     }
   }
   // Final sync so no rank tears down while a peer is mid-collective.
