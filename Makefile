@@ -32,7 +32,7 @@ CC_OBJS := $(patsubst csrc/%.cc,build/%.o,$(CC_SRCS))
 HIP_OBJS := $(patsubst csrc/%.hip,build/%.hip.o,$(HIP_SRCS))
 BIND_OBJS := $(patsubst csrc/%.cc,build/%.o,$(BIND_SRCS))
 
-LDFLAGS := -L$(ROCM)/lib -lamdhip64 -lssl -lcrypto -luv -pthread $(LDFLAGS_EXTRA)
+LDFLAGS := -L$(ROCM)/lib -lamdhip64 -lroctx64 -lssl -lcrypto -luv -pthread $(LDFLAGS_EXTRA)
 
 all: $(TARGET) $(BENCH)
 

@@ -7,6 +7,7 @@
 #include "collectives/schedule.h"
 #include "common/utils.h"
 #include "hip/kernels.h"
+#include "hip/trace.h"
 
 namespace glooamd {
 namespace hip {
@@ -130,6 +131,7 @@ void HipAllreduceRing::run(
     DType dtype,
     ReduceOp op,
     hipStream_t callerStream) {
+  TraceRange tr("gloo_amd::hip_allreduce_ring");
   GA_HIP_CHECK(hipSetDevice(device_));
   const size_t es = dtypeSize(dtype);
   const size_t bytes = elements * es;
@@ -395,6 +397,7 @@ void HipAllreduceHalvingDoubling::run(
     DType dtype,
     ReduceOp op,
     hipStream_t callerStream) {
+  TraceRange tr("gloo_amd::hip_allreduce_halving_doubling");
   GA_HIP_CHECK(hipSetDevice(device_));
   const int P = ctx_->size;
   const int r = ctx_->rank;
@@ -564,6 +567,7 @@ void HipBroadcastOneToAll::run(
     void* devPtr,
     size_t bytes,
     hipStream_t callerStream) {
+  TraceRange tr("gloo_amd::hip_broadcast_one_to_all");
   GA_HIP_CHECK(hipSetDevice(device_));
   const int P = ctx_->size;
   const int r = ctx_->rank;
@@ -649,6 +653,7 @@ void HipAllreduceDirect::run(
     DType dtype,
     ReduceOp op,
     hipStream_t callerStream) {
+  TraceRange tr("gloo_amd::hip_allreduce_direct");
   GA_HIP_CHECK(hipSetDevice(device_));
   const int P = ctx_->size;
   const int r = ctx_->rank;
@@ -832,6 +837,7 @@ void HipAllgatherRing::run(
     size_t inElements,
     size_t es,
     hipStream_t callerStream) {
+  TraceRange tr("gloo_amd::hip_allgather_ring");
   GA_HIP_CHECK(hipSetDevice(device_));
   const int P = ctx_->size;
   const int r = ctx_->rank;
@@ -950,6 +956,7 @@ void HipReduceScatterRing::run(
     DType dtype,
     ReduceOp op,
     hipStream_t callerStream) {
+  TraceRange tr("gloo_amd::hip_reduce_scatter_ring");
   GA_HIP_CHECK(hipSetDevice(device_));
   const int P = ctx_->size;
   const int r = ctx_->rank;
@@ -1071,6 +1078,7 @@ void HipAlltoall::run(
     size_t perRankElements,
     size_t es,
     hipStream_t callerStream) {
+  TraceRange tr("gloo_amd::hip_alltoall");
   GA_HIP_CHECK(hipSetDevice(device_));
   const int P = ctx_->size;
   const int r = ctx_->rank;
