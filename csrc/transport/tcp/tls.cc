@@ -132,6 +132,7 @@ void TlsPair::ioClose() {
 }
 
 ssize_t TlsPair::ioRead(char* buf, size_t len) {
+  std::lock_guard<std::mutex> lock(sslMu_);
   int n = SSL_read(ssl_, buf, static_cast<int>(std::min<size_t>(len, 1 << 30)));
   if (n > 0) {
     return n;
@@ -156,6 +157,7 @@ ssize_t TlsPair::ioRead(char* buf, size_t len) {
 }
 
 ssize_t TlsPair::ioWritev(const struct iovec* iov, int iovcnt) {
+  std::lock_guard<std::mutex> lock(sslMu_);
   ssize_t total = 0;
   for (int i = 0; i < iovcnt; i++) {
     size_t off = 0;

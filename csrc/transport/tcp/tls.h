@@ -9,6 +9,8 @@
 
 #include <openssl/ssl.h>
 
+#include <mutex>
+
 #include "transport/tcp/context.h"
 #include "transport/tcp/device.h"
 
@@ -61,6 +63,12 @@ class TlsPair : public TcpPair {
  private:
   SSL_CTX* sslCtx_;
   SSL* ssl_{nullptr};
+  // OpenSSL SSL objects are not thread-safe: the loop thread reads while
+  // a user thread may be flushing writes, so every SSL_* data call is
+  // serialized per pair. Lock order is ctx-mutex -> sslMu_ (writes) or
+  // sslMu_ alone (payload reads); the read path never takes the ctx
+  // mutex while holding sslMu_, so the order is consistent.
+  std::mutex sslMu_;
 };
 
 class TlsContext : public TcpContext {
