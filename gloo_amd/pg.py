@@ -360,6 +360,19 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
         out.copy_(h_out.view_as(out))
         return _ret_work(output_tensors)
 
+    def allreduce_coalesced(self, tensors, opts=None):
+        return self.allreduce(tensors, opts)
+
+    def allgather_into_tensor_coalesced(self, outputs, inputs, opts=None):
+        for out, inp in zip(outputs, inputs):
+            self._allgather_base(out, inp, opts).wait()
+        return _ret_work(outputs)
+
+    def reduce_scatter_tensor_coalesced(self, outputs, inputs, opts=None):
+        for out, inp in zip(outputs, inputs):
+            self._reduce_scatter_base(out, inp, opts).wait()
+        return _ret_work(outputs)
+
     def barrier(self, opts=None):
         with self._lock:
             ga.barrier(self._ctx, tag=self._tag())
