@@ -642,9 +642,14 @@ HipAllreduceDirect::HipAllreduceDirect(
     fanout_.push_back(pooledStream(ctx_.get(), device_, 1 + i));
   }
   doneEvent_ = std::make_unique<HipEvent>(device_);
+  for (int i = 0; i < 4; i++) {
+    chunkEvents_.push_back(std::make_unique<HipEvent>(device_));
+  }
   fRS_ = mesh_->allocFlags(ctx_->size);
   fAG_ = mesh_->allocFlags(ctx_->size);
   fACK_ = mesh_->allocFlags(ctx_->size);
+  fDONE_ = mesh_->allocFlags(ctx_->size);
+  lastAck_.assign(ctx_->size, {0, 0});
 }
 
 void HipAllreduceDirect::run(
@@ -663,13 +668,19 @@ void HipAllreduceDirect::run(
   }
   char* user = static_cast<char*>(devPtr);
   const size_t perRank = (elements + P - 1) / P;
-  const size_t blockCapBytes = perRank * es;
-  // inbox holds one scatter slot per source rank.
-  mesh_->ensureCapacity(
-      elements * es, (P * blockCapBytes + 1) / 2 + 64);
+  // Chunk the per-rank block so scatter / reduce / broadcast pipeline
+  // for large payloads. Inbox layout: P sources x 2 parities x chunkCap.
+  const size_t chunkCap = std::min<size_t>(
+      std::max<size_t>(perRank * es, 1), size_t(8) << 20);
+  const int C =
+      std::max<int>(1, static_cast<int>((perRank * es + chunkCap - 1) /
+                                        chunkCap));
+  mesh_->ensureCapacity(elements * es, P * chunkCap);
   char* work = mesh_->work();
-  auto subInbox = [&](char* base, int s) {
-    return base + static_cast<size_t>(s) * blockCapBytes;
+  const size_t slotStride = chunkCap; // one chunk per (src, parity) slot
+  auto subInbox = [&](char* base, int src, int par) {
+    return base +
+        (static_cast<size_t>(src) * 2 + par) * slotStride;
   };
   auto blockOff = [&](int b) {
     return std::min(static_cast<size_t>(b) * perRank, elements) * es;
@@ -678,10 +689,18 @@ void HipAllreduceDirect::run(
     return std::min(static_cast<size_t>(b + 1) * perRank, elements) * es -
         blockOff(b);
   };
+  auto chunkOffIn = [&](int b, int c) {
+    return std::min<size_t>(static_cast<size_t>(c) * chunkCap, blockLen(b));
+  };
+  auto chunkLen = [&](int b, int c) {
+    return std::min<size_t>(
+               static_cast<size_t>(c + 1) * chunkCap, blockLen(b)) -
+        chunkOffIn(b, c);
+  };
 
   {
     std::vector<hipStream_t> gated{cs_->stream()};
-    for (auto& st : fanout_) {
+    for (auto* st : fanout_) {
       gated.push_back(st->stream());
     }
     hipEvent_t ev = nullptr;
@@ -693,87 +712,116 @@ void HipAllreduceDirect::run(
     GA_HIP_CHECK(hipEventDestroy(ev));
   }
 
-  const uint64_t seq = ++seq_;
+  const uint64_t runSeq = ++runSeq_;
+  auto cseq = [&](int c) { return chunkSeqBase_ + c + 1; };
 
-  // --- scatter: my block d -> d's inbox slot r, all links concurrently ---
+  // --- chunk-major pipeline: for each chunk, enqueue scatter (fanout),
+  // fused reduce (cs), broadcast (fanout). Flags gate cross-rank deps,
+  // so chunk c+1's wire traffic overlaps chunk c's reduction. ---
   for (int j = 1; j < P; j++) {
     const int d = (r + j) % P;
     auto& st = *fanout_[(j - 1) % fanout_.size()];
-    if (seq > 1) {
-      launchWaitFlagGte(mesh_->flag(fACK_ + d), seq - 1, st.stream());
-    }
-    if (blockLen(d) > 0) {
-      GA_HIP_CHECK(hipMemcpyAsync(
-          subInbox(mesh_->peerInbox(d, 0), r),
-          user + blockOff(d),
-          blockLen(d),
-          hipMemcpyDeviceToDevice,
-          st.stream()));
-    }
-    launchWriteFlag(mesh_->peerFlag(d, fRS_ + r), seq, st.stream());
-  }
-
-  // --- fused local reduction of my block (cs) ---
-  for (int s = 0; s < P; s++) {
-    if (s != r) {
-      launchWaitFlagGte(mesh_->flag(fRS_ + s), seq, cs_->stream());
+    if (runSeq > 1) {
+      // d finished its previous copy-out before we touch its memory.
+      launchWaitFlagGte(mesh_->flag(fDONE_ + d), runSeq - 1, st.stream());
     }
   }
-  if (blockLen(r) > 0) {
-    const void* srcs[8];
-    srcs[0] = user + blockOff(r);
-    int k = 1;
-    for (int s = 0; s < P; s++) {
-      if (s != r) {
-        srcs[k++] = subInbox(mesh_->inbox(0), s);
+  for (int c = 0; c < C; c++) {
+    const int par = c & 1;
+    // scatter chunk c of every destination block
+    for (int j = 1; j < P; j++) {
+      const int d = (r + j) % P;
+      auto& st = *fanout_[(j - 1) % fanout_.size()];
+      const uint64_t gate = (c >= 2) ? cseq(c - 2) : lastAck_[d][par];
+      if (gate > 0) {
+        launchWaitFlagGte(mesh_->flag(fACK_ + d), gate, st.stream());
+      }
+      const size_t len = chunkLen(d, c);
+      if (len > 0) {
+        GA_HIP_CHECK(hipMemcpyAsync(
+            subInbox(mesh_->peerInbox(d, 0), r, par),
+            user + blockOff(d) + chunkOffIn(d, c),
+            len,
+            hipMemcpyDeviceToDevice,
+            st.stream()));
+      }
+      launchWriteFlag(mesh_->peerFlag(d, fRS_ + r), cseq(c), st.stream());
+    }
+    // fused reduction of chunk c of my block
+    for (int src = 0; src < P; src++) {
+      if (src != r) {
+        launchWaitFlagGte(mesh_->flag(fRS_ + src), cseq(c), cs_->stream());
       }
     }
-    launchReduceN(
-        work + blockOff(r), srcs, P, blockLen(r) / es, dtype, op,
-        cs_->stream());
-  }
-  HipEvent reduced(device_);
-  reduced.record(cs_->stream());
-
-  // --- broadcast my reduced block over all links concurrently ---
-  for (int j = 1; j < P; j++) {
-    const int d = (r + j) % P;
-    auto& st = *fanout_[(j - 1) % fanout_.size()];
-    reduced.streamWait(st.stream());
-    if (blockLen(r) > 0) {
-      GA_HIP_CHECK(hipMemcpyAsync(
-          mesh_->peerWork(d) + blockOff(r),
-          work + blockOff(r),
-          blockLen(r),
-          hipMemcpyDeviceToDevice,
-          st.stream()));
+    const size_t rlen = chunkLen(r, c);
+    if (rlen > 0) {
+      const void* srcs[8];
+      srcs[0] = user + blockOff(r) + chunkOffIn(r, c);
+      int k = 1;
+      for (int src = 0; src < P; src++) {
+        if (src != r) {
+          srcs[k++] = subInbox(mesh_->inbox(0), src, par);
+        }
+      }
+      launchReduceN(
+          work + blockOff(r) + chunkOffIn(r, c), srcs, P, rlen / es, dtype,
+          op, cs_->stream());
     }
-    launchWriteFlag(mesh_->peerFlag(d, fAG_ + r), seq, st.stream());
+    for (int src = 0; src < P; src++) {
+      if (src != r) {
+        launchWriteFlag(mesh_->peerFlag(src, fACK_ + r), cseq(c),
+                        cs_->stream());
+      }
+    }
+    chunkEvents_[c % 4]->record(cs_->stream());
+    // broadcast reduced chunk c over all links
+    for (int j = 1; j < P; j++) {
+      const int d = (r + j) % P;
+      auto& st = *fanout_[(j - 1) % fanout_.size()];
+      chunkEvents_[c % 4]->streamWait(st.stream());
+      if (rlen > 0) {
+        GA_HIP_CHECK(hipMemcpyAsync(
+            mesh_->peerWork(d) + blockOff(r) + chunkOffIn(r, c),
+            work + blockOff(r) + chunkOffIn(r, c),
+            rlen,
+            hipMemcpyDeviceToDevice,
+            st.stream()));
+      }
+      launchWriteFlag(mesh_->peerFlag(d, fAG_ + r), cseq(c), st.stream());
+    }
   }
 
-  // --- collect all reduced blocks, stage out, ack everyone (cs) ---
-  for (int s = 0; s < P; s++) {
-    if (s != r) {
-      launchWaitFlagGte(mesh_->flag(fAG_ + s), seq, cs_->stream());
+  // --- collect, stage out, publish copy-out-done (cs) ---
+  for (int src = 0; src < P; src++) {
+    if (src != r) {
+      launchWaitFlagGte(mesh_->flag(fAG_ + src), cseq(C - 1),
+                        cs_->stream());
     }
   }
   GA_HIP_CHECK(hipMemcpyAsync(
       user, work, elements * es, hipMemcpyDeviceToDevice, cs_->stream()));
-  for (int s = 0; s < P; s++) {
-    if (s != r) {
-      launchWriteFlag(mesh_->peerFlag(s, fACK_ + r), seq, cs_->stream());
+  for (int src = 0; src < P; src++) {
+    if (src != r) {
+      launchWriteFlag(mesh_->peerFlag(src, fDONE_ + r), runSeq,
+                      cs_->stream());
     }
   }
   doneEvent_->record(cs_->stream());
   auto timeout = ctx_->getTimeout();
   watchdogWait(*doneEvent_, *mesh_, timeout, "hip_allreduce_direct (cs)");
   cs_->synchronize();
-  for (auto& st : fanout_) {
+  for (auto* st : fanout_) {
     HipEvent done(device_);
     done.record(st->stream());
     watchdogWait(done, *mesh_, timeout, "hip_allreduce_direct (fanout)");
     st->synchronize();
   }
+  for (int d = 0; d < P; d++) {
+    for (int c = std::max(0, C - 2); c < C; c++) {
+      lastAck_[d][c & 1] = cseq(c);
+    }
+  }
+  chunkSeqBase_ += C;
 }
 
 // ===========================================================================

@@ -194,10 +194,14 @@ class HipAllreduceDirect {
   HipStream* cs_;
   std::vector<HipStream*> fanout_;
   std::unique_ptr<HipEvent> doneEvent_;
-  int fRS_; // [src] scatter-block arrived
-  int fAG_; // [src] reduced-block arrived
-  int fACK_; // [src] peer finished reading my work / inbox reusable
-  uint64_t seq_{0};
+  std::vector<std::unique_ptr<HipEvent>> chunkEvents_;
+  int fRS_; // [src] scatter chunk arrived (monotonic per-chunk seq)
+  int fAG_; // [src] reduced chunk arrived
+  int fACK_; // [src] inbox slot consumed (per-chunk seq)
+  int fDONE_; // [src] peer's copy-out done (per-run seq)
+  uint64_t chunkSeqBase_{0};
+  uint64_t runSeq_{0};
+  std::vector<std::array<uint64_t, 2>> lastAck_; // per src, per parity
 };
 
 // Base-b hypercube allreduce (reference cuda_allreduce_bcube parity).
