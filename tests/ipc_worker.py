@@ -34,6 +34,22 @@ def main():
     # second run exercises cross-run flag sequencing over IPC
     algo.run(x.data_ptr(), n, ga.DType.f32, ga.ReduceOp.sum)
     torch.cuda.synchronize()
+
+    # direct engine with >8MiB blocks (chunk pipeline), 3 runs
+    nd = 5_000_001 * 2 // size * size  # keep per-rank blocks > 8 MiB
+    gd = torch.Generator("cpu").manual_seed(100 + rank)
+    y = torch.rand(nd, generator=gd, dtype=torch.float32).cuda()
+    refd = sum(
+        torch.rand(nd, generator=torch.Generator("cpu").manual_seed(100 + r),
+                   dtype=torch.float32)
+        for r in range(size))
+    direct = ga._C.HipAllreduceDirect(ctx, 0)
+    for _ in range(3):
+        z = y.clone()
+        direct.run(z.data_ptr(), nd, ga.DType.f32, ga.ReduceOp.sum)
+        torch.cuda.synchronize()
+        assert torch.allclose(z.cpu(), refd, atol=1e-4), (
+            (z.cpu() - refd).abs().max())
     print("IPC-OK rank", rank)
 
 

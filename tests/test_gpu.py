@@ -260,7 +260,7 @@ def _two_rank_generic(builder_and_check):
 
 def test_hip_allreduce_direct():
     def fn(ctx, rank):
-        n = 1_000_001  # odd: block tails
+        n = 5_000_001  # odd block tails + >8MiB blocks -> chunk pipeline
         g = torch.Generator("cpu").manual_seed(rank)
         x = _dev(torch.rand(n, generator=g))
         ref = sum(
