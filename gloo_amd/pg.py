@@ -165,7 +165,10 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
         with self._lock:
             for t in tensors:
                 t_ = t.detach()
-                assert t_.is_contiguous(), "gloo_amd needs contiguous tensors"
+                staged = None
+                if not t_.is_contiguous():
+                    staged = t_.contiguous()
+                    t_ = staged
                 if t_.is_cuda:
                     self._ring(t_.get_device()).run(
                         t_.data_ptr(), t_.numel(), _gdtype(t_), gop,
@@ -176,6 +179,8 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
                         gop, tag=self._tag())
                 if op == dist.ReduceOp.AVG:
                     t_.div_(self.size())
+                if staged is not None:
+                    t.detach().copy_(staged)
         return _ret_work(tensors)
 
     def broadcast(self, tensors, opts=None):
@@ -183,7 +188,10 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
         with self._lock:
             for t in tensors:
                 t_ = t.detach()
-                assert t_.is_contiguous()
+                staged = None
+                if not t_.is_contiguous():
+                    staged = t_.contiguous()
+                    t_ = staged
                 if t_.is_cuda:
                     self._bcaster(t_.get_device(), root).run(
                         t_.data_ptr(), t_.numel() * t_.element_size(),
@@ -192,6 +200,8 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
                     ga.broadcast(
                         self._ctx, t_.data_ptr(), 0, t_.numel(), _gdtype(t_),
                         root=root, tag=self._tag())
+                if staged is not None:
+                    t.detach().copy_(staged)
         return _ret_work(tensors)
 
     def _allgather_base(self, output, input, opts=None):

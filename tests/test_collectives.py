@@ -371,3 +371,42 @@ def test_reduce_scatter(spawn_threads, size):
         return True
 
     spawn_threads(size, fn)
+
+
+@pytest.mark.parametrize("size", [6, 8])
+def test_allreduce_wide(spawn_threads, size):
+    """Wider rank counts (node-scale)."""
+
+    def fn(ctx, rank, _):
+        x = fixture(rank, 20_000)
+        ga.allreduce(ctx, [x.ctypes.data], x.size, ga.DType.f32,
+                     ga.ReduceOp.sum)
+        assert np.allclose(x, sum(fixture(r, 20_000) for r in range(size)))
+        return True
+
+    spawn_threads(size, fn)
+
+
+def test_allreduce_bcube_8(spawn_threads):
+    def fn(ctx, rank, _):
+        x = fixture(rank, 9_999)
+        ga.allreduce(ctx, [x.ctypes.data], x.size, ga.DType.f32,
+                     ga.ReduceOp.sum, algorithm="bcube")
+        assert np.allclose(x, sum(fixture(r, 9_999) for r in range(8)))
+        return True
+
+    spawn_threads(8, fn, base=2)
+
+
+def test_alltoall_8(spawn_threads):
+    size, n = 8, 64
+
+    def fn(ctx, rank, _):
+        inp = np.concatenate([fixture(rank * size + d, n) for d in range(size)])
+        out = np.zeros(n * size, dtype=np.float32)
+        ga.alltoall(ctx, out.ctypes.data, inp.ctypes.data, n)
+        for s in range(size):
+            assert np.allclose(out[s * n:(s + 1) * n], fixture(s * size + rank, n))
+        return True
+
+    spawn_threads(size, fn)

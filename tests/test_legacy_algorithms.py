@@ -155,3 +155,29 @@ def test_legacy_allreduce_local(spawn_threads):
         return True
 
     spawn_threads(1, fn)
+
+
+def test_legacy_allreduce_hd_8(spawn_threads):
+    def fn(ctx, rank, _):
+        x = fixture(rank, 12_345)
+        a = ga._C.create_algorithm("allreduce_halving_doubling", ctx,
+                                   [x.ctypes.data], 12_345)
+        a.run()
+        assert np.allclose(x, sum(fixture(r, 12_345) for r in range(8)))
+        return True
+
+    spawn_threads(8, fn)
+
+
+def test_legacy_allreduce_hd_6(spawn_threads):
+    """Non-power-of-2: pre/post folding path."""
+
+    def fn(ctx, rank, _):
+        x = fixture(rank, 5_000)
+        a = ga._C.create_algorithm("allreduce_halving_doubling", ctx,
+                                   [x.ctypes.data], 5_000)
+        a.run()
+        assert np.allclose(x, sum(fixture(r, 5_000) for r in range(6)))
+        return True
+
+    spawn_threads(6, fn)
