@@ -66,6 +66,7 @@ class EpollLoop : public Loop {
 
   int epfd_{-1};
   int evfd_{-1};
+  int busyPollUs_{200}; // GLOO_AMD_BUSY_POLL_US; 0 disables
   std::atomic<bool> done_{false};
   std::thread thread_;
   std::thread::id threadId_;
