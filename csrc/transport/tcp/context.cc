@@ -134,14 +134,14 @@ void TcpContext::signalException(const std::string& msg) {
   cv_.notify_all();
 }
 
-bool TcpContext::offerSendReadyLocked(TcpPair* pair, uint64_t slot) {
+bool TcpContext::claimAnyRecvLocked(int srcRank, uint64_t slot, AnyRecv* out) {
   for (auto it = anyRecvs_.begin(); it != anyRecvs_.end(); ++it) {
     if (it->slot != slot) {
       continue;
     }
     bool match = false;
     for (int r : it->srcRanks) {
-      if (r == pair->peerRank()) {
+      if (r == srcRank) {
         match = true;
         break;
       }
@@ -149,12 +149,20 @@ bool TcpContext::offerSendReadyLocked(TcpPair* pair, uint64_t slot) {
     if (!match) {
       continue;
     }
-    AnyRecv ar = *it;
+    *out = *it;
     anyRecvs_.erase(it);
-    pair->postRecv(ar.buf, ar.slot, ar.offset, ar.nbytes, /*fromClaim=*/true);
     return true;
   }
   return false;
+}
+
+bool TcpContext::offerSendReadyLocked(TcpPair* pair, uint64_t slot) {
+  AnyRecv ar;
+  if (!claimAnyRecvLocked(pair->peerRank(), slot, &ar)) {
+    return false;
+  }
+  pair->postRecv(ar.buf, ar.slot, ar.offset, ar.nbytes, /*fromClaim=*/true);
+  return true;
 }
 
 bool TcpContext::consumeUnclaimedLocked(uint64_t slot, int rank) {
@@ -354,6 +362,17 @@ void TcpUnboundBuffer::recv(
         recvCompletions_.push_back(ctx_->rank);
         s.buf->sendCompletions_++;
         ctx_->cv_.notify_all();
+        return;
+      }
+    }
+  }
+  // Already-arrived eager/credit payloads are the oldest available data
+  // on their pair; take them before claiming merely-notified sends.
+  for (int r : srcRanks) {
+    if (r != ctx_->rank) {
+      auto* pair = static_cast<TcpPair*>(ctx_->getPair(r));
+      if (pair != nullptr &&
+          pair->takeStashLocked(this, slot, offset, nbytes)) {
         return;
       }
     }

@@ -84,9 +84,19 @@ class TcpContext : public transport::Context,
   friend class TcpUnboundBuffer;
   friend class TcpBuffer;
 
+  struct AnyRecv {
+    TcpUnboundBuffer* buf;
+    uint64_t slot;
+    size_t offset;
+    size_t nbytes;
+    std::vector<int> srcRanks;
+  };
+
   // Called on SEND_READY arrival creating an excess notified send.
   // Returns true if an any-recv waiter matched (recv posted inside).
   bool offerSendReadyLocked(TcpPair* pair, uint64_t slot);
+  // Pop the oldest any-recv waiter matching (slot, srcRank) into *out.
+  bool claimAnyRecvLocked(int srcRank, uint64_t slot, AnyRecv* out);
   // Consume one unclaimed send-ready entry for (slot, rank) if present.
   bool consumeUnclaimedLocked(uint64_t slot, int rank);
 
@@ -100,13 +110,6 @@ class TcpContext : public transport::Context,
   std::vector<std::unique_ptr<TcpPair>> pairs_;
 
   std::unordered_map<uint64_t, std::deque<int>> unclaimedSendReady_;
-  struct AnyRecv {
-    TcpUnboundBuffer* buf;
-    uint64_t slot;
-    size_t offset;
-    size_t nbytes;
-    std::vector<int> srcRanks;
-  };
   std::list<AnyRecv> anyRecvs_;
 
   struct SelfOp {

@@ -7,6 +7,7 @@
 #include <sys/uio.h>
 #include <unistd.h>
 
+#include <cstdlib>
 #include <cstring>
 
 #include "common/logging.h"
@@ -262,6 +263,15 @@ void TcpPair::ubufSend(
     op.payloadLen = nb;
     op.ubuf = buf;
     enqueueTxLocked(std::move(op));
+  } else if (nb <= eagerMaxBytes() && localPendingSend_[slot].empty()) {
+    // Small message, no credit, and no older notified send whose payload
+    // this one must not overtake: push the data without rendezvous.
+    TxOp op{};
+    op.pre = {EAGER_DATA, static_cast<uint32_t>(ctx_->rank), slot, nb, 0};
+    op.payload = static_cast<const char*>(buf->ptr) + off;
+    op.payloadLen = nb;
+    op.ubuf = buf;
+    enqueueTxLocked(std::move(op));
   } else {
     localPendingSend_[slot].push_back({buf, off, nb});
     TxOp op{};
@@ -270,12 +280,52 @@ void TcpPair::ubufSend(
   }
 }
 
+size_t TcpPair::eagerMaxBytes() {
+  static size_t v = [] {
+    if (const char* e = getenv("GLOO_AMD_EAGER_MAX")) {
+      return static_cast<size_t>(atoll(e));
+    }
+    return static_cast<size_t>(8192);
+  }();
+  return v;
+}
+
+bool TcpPair::takeStashLocked(
+    TcpUnboundBuffer* buf,
+    uint64_t slot,
+    size_t off,
+    size_t nb) {
+  auto it = eagerStash_.find(slot);
+  if (it == eagerStash_.end() || it->second.empty()) {
+    return false;
+  }
+  std::string payload = std::move(it->second.front());
+  it->second.pop_front();
+  if (it->second.empty()) {
+    eagerStash_.erase(it);
+  }
+  GA_ENFORCE_LE(payload.size(), nb, "stashed payload larger than posted recv");
+  std::memcpy(static_cast<char*>(buf->ptr) + off, payload.data(), payload.size());
+  buf->recvCompletions_.push_back(peerRank_);
+  ctx_->cv_.notify_all();
+  return true;
+}
+
 void TcpPair::postRecv(
     TcpUnboundBuffer* buf,
     uint64_t slot,
     size_t off,
     size_t nb,
     bool fromClaim) {
+  if (!fromClaim && takeStashLocked(buf, slot, off, nb)) {
+    // Satisfied by already-arrived eager/credit data: complete locally.
+    // No RECV_READY (the sender never queued this payload) and no counter
+    // changes (the stash never corresponds to a notified send). Checked
+    // before the error check: a peer that eagerly sent its final payload
+    // may have exited (closing the pair) before this recv was posted —
+    // fully-arrived data is still valid.
+    return;
+  }
   if (error_) {
     std::rethrow_exception(error_);
   }
@@ -323,7 +373,8 @@ void TcpPair::enqueueTxLocked(TxOp op) {
 }
 
 void TcpPair::completeTxLocked(TxOp& op) {
-  if (op.ubuf != nullptr && op.pre.opcode == UNBOUND_DATA) {
+  if (op.ubuf != nullptr &&
+      (op.pre.opcode == UNBOUND_DATA || op.pre.opcode == EAGER_DATA)) {
     op.ubuf->sendCompletions_++;
     ctx_->cv_.notify_all();
   }
@@ -514,26 +565,45 @@ void TcpPair::dispatchPreamble() {
       }
       return;
     }
-    case UNBOUND_DATA: {
+    case UNBOUND_DATA:
+    case EAGER_DATA: {
+      // A posted recv takes the data directly; otherwise a matching
+      // any-recv waiter does; otherwise it is stashed until one is
+      // posted. UNBOUND_DATA can hit the stash path too: a credit send
+      // whose RECV_READY's recv was itself satisfied from the stash.
       auto& q = localPendingRecv_[pre.slot];
-      GA_ENFORCE(
-          !q.empty(),
-          "protocol violation: unbound data with no posted recv, slot ",
-          pre.slot);
-      PendingRecv pr = q.front();
-      q.pop_front();
-      GA_ENFORCE_LE(pre.length, pr.nbytes, "unbound payload overflow");
-      rxUbuf_ = pr.buf;
-      rxDst_ = static_cast<char*>(pr.buf->ptr) + pr.offset;
+      if (!q.empty()) {
+        PendingRecv pr = q.front();
+        q.pop_front();
+        GA_ENFORCE_LE(pre.length, pr.nbytes, "unbound payload overflow");
+        rxUbuf_ = pr.buf;
+        rxDst_ = static_cast<char*>(pr.buf->ptr) + pr.offset;
+      } else {
+        TcpContext::AnyRecv ar;
+        if (ctx_->claimAnyRecvLocked(peerRank_, pre.slot, &ar)) {
+          GA_ENFORCE_LE(pre.length, ar.nbytes, "unbound payload overflow");
+          rxUbuf_ = ar.buf;
+          rxDst_ = static_cast<char*>(ar.buf->ptr) + ar.offset;
+        } else {
+          rxSpill_.resize(pre.length);
+          rxDst_ = rxSpill_.empty() ? nullptr : &rxSpill_[0];
+          rxIsEagerSpill_ = true;
+        }
+      }
       rxLen_ = pre.length;
       rxRead_ = 0;
       rxActive_ = true;
       if (rxLen_ == 0) {
         // complete inline (finishRx would re-lock)
-        rxUbuf_->recvCompletions_.push_back(peerRank_);
-        rxUbuf_ = nullptr;
+        if (rxUbuf_ != nullptr) {
+          rxUbuf_->recvCompletions_.push_back(peerRank_);
+          rxUbuf_ = nullptr;
+          ctx_->cv_.notify_all();
+        } else {
+          eagerStash_[pre.slot].emplace_back();
+          rxIsEagerSpill_ = false;
+        }
         rxActive_ = false;
-        ctx_->cv_.notify_all();
       }
       return;
     }
@@ -578,6 +648,32 @@ void TcpPair::finishRx() {
   if (rxUbuf_ != nullptr) {
     rxUbuf_->recvCompletions_.push_back(peerRank_);
     rxUbuf_ = nullptr;
+  } else if (rxIsEagerSpill_) {
+    // A recv may have been posted while the payload was being read;
+    // deliver directly in that case, matching the bound-data replay rule.
+    auto& q = localPendingRecv_[rxPre_.slot];
+    TcpContext::AnyRecv ar;
+    if (!q.empty()) {
+      PendingRecv pr = q.front();
+      q.pop_front();
+      GA_ENFORCE_LE(rxSpill_.size(), pr.nbytes, "unbound payload overflow");
+      std::memcpy(
+          static_cast<char*>(pr.buf->ptr) + pr.offset,
+          rxSpill_.data(),
+          rxSpill_.size());
+      pr.buf->recvCompletions_.push_back(peerRank_);
+    } else if (ctx_->claimAnyRecvLocked(peerRank_, rxPre_.slot, &ar)) {
+      GA_ENFORCE_LE(rxSpill_.size(), ar.nbytes, "unbound payload overflow");
+      std::memcpy(
+          static_cast<char*>(ar.buf->ptr) + ar.offset,
+          rxSpill_.data(),
+          rxSpill_.size());
+      ar.buf->recvCompletions_.push_back(peerRank_);
+    } else {
+      eagerStash_[rxPre_.slot].push_back(std::move(rxSpill_));
+    }
+    rxSpill_ = std::string();
+    rxIsEagerSpill_ = false;
   } else if (rxBbuf_ != nullptr) {
     rxBbuf_->recvCount_++;
     rxBbuf_ = nullptr;
@@ -656,6 +752,10 @@ void TcpPair::failLocked(std::exception_ptr e) {
   rxUbuf_ = nullptr;
   rxBbuf_ = nullptr;
   rxActive_ = false;
+  rxIsSpill_ = false;
+  rxIsEagerSpill_ = false;
+  // eagerStash_ is deliberately kept: it holds fully-received payloads,
+  // which stay deliverable after the peer closes.
 
   // Tear the socket down without waiting for a loop tick: failLocked runs
   // with the context mutex held, and the loop thread may be blocked on

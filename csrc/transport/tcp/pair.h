@@ -5,10 +5,12 @@
 // buffers, bound-buffer one-sided writes, partial read/write state
 // machines, exception fan-out) — re-designed from scratch:
 //
-//  * 40-byte preamble {opcode, srcRank, slot, length, roffset}; opcodes
-//    BOUND_DATA / UNBOUND_DATA / RECV_READY / SEND_READY.
+//  * 32-byte preamble {opcode, srcRank, slot, length, roffset}; opcodes
+//    BOUND_DATA / UNBOUND_DATA / RECV_READY / SEND_READY / EAGER_DATA.
 //  * Unbound rendezvous: recv() posts RECV_READY; send() consumes a
-//    recv-credit (payload immediately) or queues + posts SEND_READY.
+//    recv-credit (payload immediately), pushes the payload eagerly when
+//    small with no older queued send (receiver stashes if no recv is
+//    posted yet), or queues + posts SEND_READY.
 //  * recv-from-any arbitration by cumulative counters: a SEND_READY is
 //    "available" for any-recv matching iff sendReadySeen > recvsPosted
 //    on that (pair, slot) — see TcpContext.
@@ -41,6 +43,10 @@ enum WireOp : uint32_t {
   UNBOUND_DATA = 2,
   RECV_READY = 3,
   SEND_READY = 4,
+  // Small unbound send pushed without rendezvous: payload travels
+  // immediately; receiver delivers to the front pending recv, a matching
+  // any-recv waiter, or stashes until a recv is posted.
+  EAGER_DATA = 5,
 };
 
 struct Preamble {
@@ -110,6 +116,9 @@ class TcpPair : public transport::Pair, public Handler {
       size_t nb,
       bool fromClaim);
   void boundSend(TcpBuffer* buf, size_t off, size_t len, size_t roff);
+  // Deliver the oldest stashed eager/credit payload for slot into buf, if
+  // any. Completes the recv locally (no RECV_READY). Context mutex held.
+  bool takeStashLocked(TcpUnboundBuffer* buf, uint64_t slot, size_t off, size_t nb);
   void failLocked(std::exception_ptr e); // fan exception to all pending ops
   bool referencesBuffer(const void* buf) const; // in-flight rx/tx check
   void detachBuffer(TcpBuffer* buf); // bound buffer destructor support
@@ -143,6 +152,8 @@ class TcpPair : public transport::Pair, public Handler {
     TcpBuffer* bbuf{nullptr};
     TcpUnboundBuffer* ubuf{nullptr};
   };
+
+  static size_t eagerMaxBytes(); // GLOO_AMD_EAGER_MAX, default 8 KiB
 
   void enqueueTxLocked(TxOp op);
   void flushTxLocked();
@@ -184,6 +195,10 @@ class TcpPair : public transport::Pair, public Handler {
   // BOUND_DATA that arrived before createRecvBuffer: slot -> (roffset, data)
   std::unordered_map<uint64_t, std::deque<std::pair<uint64_t, std::string>>>
       earlyBoundData_;
+  // Unbound payloads (eager, or credit sends that overtook a stash-satisfied
+  // recv's RECV_READY) that arrived with no posted recv. FIFO per slot;
+  // paired with localPendingRecv_: at most one of the two is non-empty.
+  std::unordered_map<uint64_t, std::deque<std::string>> eagerStash_;
 
   std::deque<TxOp> tx_;
   bool epollOutArmed_{false};
@@ -199,6 +214,7 @@ class TcpPair : public transport::Pair, public Handler {
   TcpBuffer* rxBbuf_{nullptr};
   std::string rxSpill_; // payload for not-yet-registered bound slots
   bool rxIsSpill_{false};
+  bool rxIsEagerSpill_{false}; // rxSpill_ holds unbound data for eagerStash_
 
   friend class TcpContext;
   friend class TcpBuffer;

@@ -279,3 +279,112 @@ def test_big_transfer(spawn_threads):
         return True
 
     spawn_threads(2, fn)
+
+
+def test_eager_sends_before_recvs(spawn_threads):
+    """Small sends complete and arrive before any recv is posted; the
+    receiver's stash must deliver them in FIFO order."""
+    import threading as th
+
+    barrier = th.Barrier(2)
+
+    def fn(ctx, rank, size):
+        n_msgs = 16
+        if rank == 0:
+            bufs = [np.full(64, i, dtype=np.float32) for i in range(n_msgs)]
+            ubs = [ctx.create_unbound_buffer(b.ctypes.data, b.nbytes)
+                   for b in bufs]
+            for i, ub in enumerate(ubs):
+                ub.send(1, slot=21)
+            for ub in ubs:
+                ub.wait_send()
+            barrier.wait()  # receiver posts recvs only after all sends done
+        else:
+            barrier.wait()
+            for i in range(n_msgs):
+                out = np.zeros(64, dtype=np.float32)
+                ub = ctx.create_unbound_buffer(out.ctypes.data, out.nbytes)
+                ub.recv(0, slot=21)
+                ub.wait_recv()
+                assert np.all(out == i), (i, out[0])
+        return True
+
+    spawn_threads(2, fn)
+
+
+def test_eager_mixed_sizes_fifo(spawn_threads):
+    """Small (eager-eligible) and large (rendezvous) sends interleaved on
+    one slot must be received in send order."""
+
+    def fn(ctx, rank, size):
+        sizes = [16, 100_000, 32, 200_000, 8, 64]  # elements (f32)
+        if rank == 0:
+            bufs = [np.full(s, i + 1.0, dtype=np.float32)
+                    for i, s in enumerate(sizes)]
+            ubs = [ctx.create_unbound_buffer(b.ctypes.data, b.nbytes)
+                   for b in bufs]
+            for ub in ubs:
+                ub.send(1, slot=22)
+            for ub in ubs:
+                ub.wait_send()
+        else:
+            for i, s in enumerate(sizes):
+                out = np.zeros(s, dtype=np.float32)
+                ub = ctx.create_unbound_buffer(out.ctypes.data, out.nbytes)
+                ub.recv(0, slot=22)
+                ub.wait_recv()
+                assert np.all(out == i + 1.0), (i, out[:4])
+        return True
+
+    spawn_threads(2, fn)
+
+
+def test_eager_recv_from_any_stashed(spawn_threads):
+    """recv-from-any must find payloads that arrived eagerly before the
+    any-recv was registered."""
+    import time
+
+    def fn(ctx, rank, size):
+        if rank in (0, 1):
+            val = np.full(32, rank + 5.0, dtype=np.float32)
+            ub = ctx.create_unbound_buffer(val.ctypes.data, val.nbytes)
+            ub.send(2, slot=23)
+            ub.wait_send()
+        else:
+            time.sleep(0.05)  # let both eager payloads arrive and stash
+            seen = set()
+            for _ in range(2):
+                out = np.zeros(32, dtype=np.float32)
+                ub = ctx.create_unbound_buffer(out.ctypes.data, out.nbytes)
+                ub.recv_any([0, 1], slot=23)
+                ok, src = ub.wait_recv()
+                assert ok and np.all(out == src + 5.0)
+                seen.add(src)
+            assert seen == {0, 1}
+        return True
+
+    spawn_threads(3, fn)
+
+
+def test_eager_zero_length(spawn_threads):
+    """Zero-byte eager sends (pure notifications) before and after the
+    recv post."""
+    import threading as th
+
+    barrier = th.Barrier(2)
+
+    def fn(ctx, rank, size):
+        b = np.zeros(1, dtype=np.float32)
+        ub = ctx.create_unbound_buffer(b.ctypes.data, b.nbytes)
+        if rank == 0:
+            ub.send(1, slot=24, nbytes=0)
+            ub.wait_send()
+            barrier.wait()
+        else:
+            barrier.wait()  # payload already stashed
+            ub.recv(0, slot=24, nbytes=0)
+            ok, src = ub.wait_recv()
+            assert ok and src == 0
+        return True
+
+    spawn_threads(2, fn)
