@@ -72,3 +72,11 @@ examples: $(CC_OBJS) $(HIP_OBJS)
 	$(CXX) $(CXXFLAGS) examples/example_v2.cc $(CC_OBJS) $(HIP_OBJS) -o bin/example_v2 $(LDFLAGS)
 
 .PHONY: examples
+
+# Race-detection stress harness (see csrc/bench/race_stress.cc).
+RACE_OBJS := build/bench/race_stress.o
+race_stress: bin/race_stress
+bin/race_stress: $(RACE_OBJS) $(CC_OBJS) $(HIP_OBJS)
+	@mkdir -p bin
+	$(CXX) -o $@ $^ $(LDFLAGS)
+.PHONY: race_stress

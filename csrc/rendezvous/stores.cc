@@ -47,7 +47,9 @@ void HashStore::wait(
     }
     return true;
   };
-  if (!cv_.wait_for(lock, timeout, pred)) {
+  if (timeout.count() < 0) {
+    cv_.wait(lock, pred); // negative timeout = wait forever
+  } else if (!cv_.wait_for(lock, timeout, pred)) {
     throw TimeoutException("HashStore::wait timeout");
   }
 }

@@ -21,6 +21,7 @@
 
 #include <sys/uio.h>
 
+#include <atomic>
 #include <deque>
 #include <string>
 #include <unordered_map>
@@ -169,9 +170,12 @@ class TcpPair : public transport::Pair, public Handler {
   int peerRank_;
   TcpAddress self_;
   TcpAddress peer_;
-  int fd_{-1};
+  // Written under the context mutex; read lock-free by the loop thread's
+  // readLoop fast-path checks, hence atomic (relaxed is sufficient: the
+  // epoll unregister + tick-wait in close() is the real synchronizer).
+  std::atomic<int> fd_{-1};
   enum State { INIT, CONNECTED, CLOSED };
-  State state_{INIT};
+  std::atomic<State> state_{INIT};
   std::exception_ptr error_;
 
   // --- protocol state (context mutex) ---
