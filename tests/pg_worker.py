@@ -29,6 +29,48 @@ def main():
     assert torch.allclose(t, torch.full((8,), 1.0))
     dist.barrier()
 
+    # object collectives (pickle over the tensor collectives)
+    gathered = [None] * size
+    dist.all_gather_object(gathered, {"rank": rank, "v": [rank] * 3})
+    assert [g["rank"] for g in gathered] == list(range(size))
+    olist = [{"x": 1}, "hello"] if rank == 1 else [None, None]
+    dist.broadcast_object_list(olist, src=1)
+    assert olist[1] == "hello" and olist[0] == {"x": 1}
+
+    # point-to-point, blocking and async, with tags
+    if rank == 0:
+        dist.send(torch.arange(10.0), dst=1, tag=5)
+        got = torch.zeros(10)
+        dist.recv(got, src=1, tag=6)
+        assert torch.allclose(got, torch.arange(10.0) + 1)
+        req = dist.isend(torch.full((4,), 7.0), dst=1, tag=9)
+        req.wait()
+    elif rank == 1:
+        got = torch.zeros(10)
+        dist.recv(got, src=0, tag=5)
+        dist.send(got + 1, dst=0, tag=6)
+        g2 = torch.zeros(4)
+        req = dist.irecv(g2, src=0, tag=9)
+        req.wait()
+        assert torch.allclose(g2, torch.full((4,), 7.0))
+
+    # fused-tensor collectives the dispatcher lowers to *_base
+    out = torch.zeros(size * 8)
+    dist.all_gather_into_tensor(out, torch.full((8,), float(rank)))
+    for r in range(size):
+        assert torch.allclose(out[r * 8:(r + 1) * 8], torch.full((8,), float(r)))
+    rs_out = torch.zeros(8)
+    dist.reduce_scatter_tensor(rs_out, torch.arange(float(size * 8)))
+    assert torch.allclose(
+        rs_out, size * (torch.arange(8.0) + 8 * rank))
+    a2a_out = torch.zeros(size * 4)
+    dist.all_to_all_single(a2a_out, torch.arange(float(size * 4)) + rank * 100)
+    for s in range(size):
+        assert torch.allclose(
+            a2a_out[s * 4:(s + 1) * 4],
+            torch.arange(4.0) + rank * 4 + s * 100)
+    dist.barrier()
+
     # DDP: gradient bucket allreduce over the backend
     torch.manual_seed(0)
     model = nn.Sequential(nn.Linear(16, 32), nn.ReLU(), nn.Linear(32, 4))
