@@ -151,7 +151,20 @@ PYBIND11_MODULE(_C, m) {
             s.wait(keys, ms(timeoutMs));
           },
           py::arg("keys"),
-          py::arg("timeout_ms") = 30000);
+          py::arg("timeout_ms") = 30000)
+      .def("has_v2", &IStore::hasV2)
+      .def(
+          "append",
+          [](IStore& s, const std::string& key, py::bytes data) {
+            std::string d = data;
+            s.append(key, std::vector<char>(d.begin(), d.end()));
+          })
+      .def(
+          "add",
+          [](IStore& s, const std::string& key, int64_t delta) {
+            py::gil_scoped_release rel;
+            return s.add(key, delta);
+          });
 
   py::class_<HashStore, IStore, std::shared_ptr<HashStore>>(m, "HashStore")
       .def(py::init<>());

@@ -4,6 +4,7 @@
 #pragma once
 
 #include <chrono>
+#include <stdexcept>
 #include <string>
 #include <vector>
 
@@ -26,6 +27,24 @@ class IStore {
       const std::chrono::milliseconds& timeout) = 0;
   void wait(const std::vector<std::string>& keys) {
     wait(keys, kDefaultTimeout);
+  }
+
+  // --- v2 store API parity (reference gloo/rendezvous/store.h:25-74) ---
+  // Stores that override append/add atomically report true here.
+  virtual bool hasV2() const {
+    return false;
+  }
+  // Atomically concatenate data to the key's value (creates it if absent).
+  virtual void append(const std::string& key, const std::vector<char>& data) {
+    (void)key;
+    (void)data;
+    throw std::logic_error("store does not support append (no v2 API)");
+  }
+  // Atomically add delta to an integer-valued key; returns the new value.
+  virtual int64_t add(const std::string& key, int64_t delta) {
+    (void)key;
+    (void)delta;
+    throw std::logic_error("store does not support add (no v2 API)");
   }
 
   // Batched variants (v2 store API parity); default to loops.

@@ -2,6 +2,7 @@
 
 #include <arpa/inet.h>
 #include <fcntl.h>
+#include <sys/file.h>
 #include <netdb.h>
 #include <netinet/tcp.h>
 #include <sys/socket.h>
@@ -52,6 +53,24 @@ void HashStore::wait(
   } else if (!cv_.wait_for(lock, timeout, pred)) {
     throw TimeoutException("HashStore::wait timeout");
   }
+}
+
+void HashStore::append(const std::string& key, const std::vector<char>& data) {
+  std::lock_guard<std::mutex> lock(mu_);
+  auto& v = map_[key];
+  v.insert(v.end(), data.begin(), data.end());
+  cv_.notify_all();
+}
+
+int64_t HashStore::add(const std::string& key, int64_t delta) {
+  std::lock_guard<std::mutex> lock(mu_);
+  auto& v = map_[key];
+  int64_t cur = v.empty() ? 0 : strtoll(std::string(v.begin(), v.end()).c_str(), nullptr, 10);
+  cur += delta;
+  std::string s = std::to_string(cur);
+  v.assign(s.begin(), s.end());
+  cv_.notify_all();
+  return cur;
 }
 
 // ---------------------------------------------------------------------------
@@ -105,6 +124,56 @@ std::vector<char> FileStore::get(const std::string& key) {
       std::istreambuf_iterator<char>(f), std::istreambuf_iterator<char>());
 }
 
+namespace {
+// RAII flock on a per-store lock file: cross-process atomicity for the
+// read-modify-write v2 ops.
+class FileLock {
+ public:
+  explicit FileLock(const std::string& path) {
+    fd_ = open(path.c_str(), O_CREAT | O_RDWR | O_CLOEXEC, 0666);
+    GA_ENFORCE_GE(fd_, 0, "FileStore lock open: ", strerror(errno));
+    GA_ENFORCE_EQ(flock(fd_, LOCK_EX), 0, "flock: ", strerror(errno));
+  }
+  ~FileLock() {
+    if (fd_ >= 0) {
+      flock(fd_, LOCK_UN);
+      close(fd_);
+    }
+  }
+
+ private:
+  int fd_{-1};
+};
+
+std::vector<char> readFileIfExists(const std::string& path) {
+  std::ifstream f(path, std::ios::binary);
+  if (!f.good()) {
+    return {};
+  }
+  return std::vector<char>(
+      std::istreambuf_iterator<char>(f), std::istreambuf_iterator<char>());
+}
+} // namespace
+
+void FileStore::append(const std::string& key, const std::vector<char>& data) {
+  FileLock lock(basePath_ + "/.lock");
+  auto cur = readFileIfExists(objectPath(key));
+  cur.insert(cur.end(), data.begin(), data.end());
+  set(key, cur); // tmp+rename keeps readers atomic too
+}
+
+int64_t FileStore::add(const std::string& key, int64_t delta) {
+  FileLock lock(basePath_ + "/.lock");
+  auto cur = readFileIfExists(objectPath(key));
+  int64_t v = cur.empty()
+      ? 0
+      : strtoll(std::string(cur.begin(), cur.end()).c_str(), nullptr, 10);
+  v += delta;
+  std::string s = std::to_string(v);
+  set(key, std::vector<char>(s.begin(), s.end()));
+  return v;
+}
+
 bool FileStore::check(const std::vector<std::string>& keys) const {
   for (const auto& k : keys) {
     struct stat st;
@@ -151,6 +220,16 @@ void PrefixStore::wait(
     prefixed.push_back(prefix_ + "/" + k);
   }
   store_->wait(prefixed, timeout);
+}
+
+void PrefixStore::append(
+    const std::string& key,
+    const std::vector<char>& data) {
+  store_->append(prefix_ + "/" + key, data);
+}
+
+int64_t PrefixStore::add(const std::string& key, int64_t delta) {
+  return store_->add(prefix_ + "/" + key, delta);
 }
 
 // ---------------------------------------------------------------------------
@@ -278,7 +357,7 @@ class TcpStore::Server {
       if (!readAll(fd, &key[0], klen)) {
         break;
       }
-      if (op == 'S') {
+      if (op == 'S' || op == 'A') {
         uint32_t vlen;
         if (!readAll(fd, &vlen, 4)) {
           break;
@@ -288,8 +367,33 @@ class TcpStore::Server {
           break;
         }
         std::lock_guard<std::mutex> lock(mu_);
-        map_[key] = std::move(val);
+        if (op == 'S') {
+          map_[key] = std::move(val);
+        } else {
+          auto& cur = map_[key];
+          cur.insert(cur.end(), val.begin(), val.end());
+        }
         cv_.notify_all();
+      } else if (op == 'I') {
+        int64_t delta;
+        if (!readAll(fd, &delta, 8)) {
+          break;
+        }
+        int64_t result;
+        {
+          std::lock_guard<std::mutex> lock(mu_);
+          auto& cur = map_[key];
+          int64_t v = cur.empty()
+              ? 0
+              : strtoll(
+                    std::string(cur.begin(), cur.end()).c_str(), nullptr, 10);
+          v += delta;
+          std::string sv = std::to_string(v);
+          cur.assign(sv.begin(), sv.end());
+          result = v;
+          cv_.notify_all();
+        }
+        writeAll(fd, &result, 8);
       } else if (op == 'G' || op == 'W') {
         std::unique_lock<std::mutex> lock(mu_);
         cv_.wait(lock, [&] { return map_.count(key) > 0 || done_.load(); });
@@ -380,6 +484,31 @@ void TcpStore::set(const std::string& key, const std::vector<char>& data) {
   if (vlen > 0) {
     writeAll(clientFd_, data.data(), vlen);
   }
+}
+
+void TcpStore::append(
+    const std::string& key,
+    const std::vector<char>& data) {
+  std::lock_guard<std::mutex> lock(clientMu_);
+  uint8_t op = 'A';
+  writeAll(clientFd_, &op, 1);
+  writeString(clientFd_, key);
+  uint32_t vlen = data.size();
+  writeAll(clientFd_, &vlen, 4);
+  if (vlen > 0) {
+    writeAll(clientFd_, data.data(), vlen);
+  }
+}
+
+int64_t TcpStore::add(const std::string& key, int64_t delta) {
+  std::lock_guard<std::mutex> lock(clientMu_);
+  uint8_t op = 'I';
+  writeAll(clientFd_, &op, 1);
+  writeString(clientFd_, key);
+  writeAll(clientFd_, &delta, 8);
+  int64_t result;
+  GA_ENFORCE(readAll(clientFd_, &result, 8), "TcpStore: server closed");
+  return result;
 }
 
 std::vector<char> TcpStore::get(const std::string& key) {

@@ -26,6 +26,11 @@ class HashStore : public IStore {
   void wait(
       const std::vector<std::string>& keys,
       const std::chrono::milliseconds& timeout) override;
+  bool hasV2() const override {
+    return true;
+  }
+  void append(const std::string& key, const std::vector<char>& data) override;
+  int64_t add(const std::string& key, int64_t delta) override;
 
  private:
   std::mutex mu_;
@@ -41,6 +46,12 @@ class FileStore : public IStore {
   void wait(
       const std::vector<std::string>& keys,
       const std::chrono::milliseconds& timeout) override;
+  bool hasV2() const override {
+    return true;
+  }
+  // Atomic across processes via flock on a per-store lock file.
+  void append(const std::string& key, const std::vector<char>& data) override;
+  int64_t add(const std::string& key, int64_t delta) override;
 
  private:
   std::string objectPath(const std::string& key) const;
@@ -56,6 +67,11 @@ class PrefixStore : public IStore {
   void wait(
       const std::vector<std::string>& keys,
       const std::chrono::milliseconds& timeout) override;
+  bool hasV2() const override {
+    return store_->hasV2();
+  }
+  void append(const std::string& key, const std::vector<char>& data) override;
+  int64_t add(const std::string& key, int64_t delta) override;
 
  private:
   std::string prefix_;
@@ -64,8 +80,9 @@ class PrefixStore : public IStore {
 
 // TCP key/value store. One process runs the server (isServer=true, usually
 // rank 0); every process (including the server's own) connects as a client.
-// Wire format: u8 op ('S','G','W') + u32 keylen + key [+ u32 vallen + val];
-// GET blocks server-side until the key exists.
+// Wire format: u8 op ('S','G','W','A','I') + u32 keylen + key
+// [+ u32 vallen + val | i64 delta]; GET blocks server-side until the
+// key exists; 'A' appends, 'I' atomically adds and returns the value.
 class TcpStore : public IStore {
  public:
   TcpStore(
@@ -80,6 +97,11 @@ class TcpStore : public IStore {
   void wait(
       const std::vector<std::string>& keys,
       const std::chrono::milliseconds& timeout) override;
+  bool hasV2() const override {
+    return true;
+  }
+  void append(const std::string& key, const std::vector<char>& data) override;
+  int64_t add(const std::string& key, int64_t delta) override;
 
  private:
   class Server;

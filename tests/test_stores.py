@@ -70,3 +70,45 @@ def test_tcp_store():
     client = ga.TcpStore("127.0.0.1", port, is_server=False)
     _exercise(server, client)
     _exercise(client, server)
+
+
+@pytest.mark.parametrize("mk", ["hash", "file", "prefix", "tcp"])
+def test_store_v2_append_add(mk, tmp_path):
+    """v2 store ops: atomic append and counter add (reference
+    rendezvous/store.h v2 API)."""
+    if mk == "hash":
+        s = ga.HashStore()
+    elif mk == "file":
+        s = ga.FileStore(str(tmp_path))
+    elif mk == "prefix":
+        s = ga.PrefixStore("ns", ga.HashStore())
+    else:
+        import random
+
+        port = random.randint(20000, 40000)
+        s = ga.TcpStore("127.0.0.1", port, is_server=True)
+    assert s.has_v2()
+    s.append("k", b"abc")
+    s.append("k", b"def")
+    assert s.get("k") == b"abcdef"
+    assert s.add("n", 5) == 5
+    assert s.add("n", -2) == 3
+    assert s.get("n") == b"3"
+
+
+def test_store_v2_add_concurrent(tmp_path):
+    """Cross-thread atomicity of add on the file store (flock path)."""
+    import threading
+
+    s = ga.FileStore(str(tmp_path))
+    per = 50
+
+    def worker():
+        s2 = ga.FileStore(str(tmp_path))  # separate instance = new fds
+        for _ in range(per):
+            s2.add("ctr", 1)
+
+    ths = [threading.Thread(target=worker) for _ in range(4)]
+    [t.start() for t in ths]
+    [t.join() for t in ths]
+    assert s.add("ctr", 0) == 4 * per
