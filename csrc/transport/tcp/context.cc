@@ -104,6 +104,17 @@ void TcpContext::connectFullMesh(IStore& store) {
     GA_ENFORCE_EQ(peerAddrs.size(), static_cast<size_t>(size));
     pairs_[i]->connect(peerAddrs[rank]);
   }
+
+  // Connectivity report (reference tcp/context.cc:241-256 parity).
+  if (logThreshold() <= LogLevel::DEBUG) {
+    std::string report = "rank " + std::to_string(rank) + " connected to:";
+    for (int i = 0; i < size; i++) {
+      if (i != rank && pairs_[i] && pairs_[i]->isConnected()) {
+        report += " " + pairs_[i]->str();
+      }
+    }
+    GA_DEBUG << report;
+  }
 }
 
 void TcpContext::signalException(const std::string& msg) {
