@@ -391,6 +391,29 @@ def test_host_staged_small():
     _two_rank_device_test("ring", 1000)  # 4 KB << 256 KB threshold
 
 
+def test_cross_process_ipc_8rank():
+    """Eight processes on one GPU (timeshared): the full node-scale
+    multi-ring (strides 1,3,5,7) and P=8 direct engine over IPC — the
+    same schedules the driver's 8-GPU scaling run executes."""
+    worker = os.path.join(os.path.dirname(__file__), "ipc_worker.py")
+    tmp = "/tmp/ga_ipc8_%d" % os.getpid()
+    os.makedirs(tmp, exist_ok=True)
+    env = dict(os.environ)
+    env["HSA_ENABLE_IPC_MODE_LEGACY"] = "0"
+    procs = [
+        subprocess.Popen(
+            [sys.executable, worker, str(r), "8", tmp],
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, env=env)
+        for r in range(8)
+    ]
+    outs = []
+    for p in procs:
+        out, _ = p.communicate(timeout=300)
+        outs.append(out.decode())
+    assert all(p.returncode == 0 for p in procs), "\n".join(outs)
+    assert all("IPC-OK" in o for o in outs), "\n".join(outs)
+
+
 def test_cross_process_ipc_4rank():
     """Four processes on one GPU: exercises the multi-ring (stride 1,3)
     chunked allreduce over IPC."""
