@@ -3,6 +3,8 @@
 #include <sys/epoll.h>
 #include <uv.h>
 
+#include <cstdlib>
+
 #include <atomic>
 #include <condition_variable>
 #include <mutex>
@@ -20,7 +22,16 @@ namespace {
 class UvLoop : public Loop {
  public:
   UvLoop() {
+    if (const char* e = getenv("GLOO_AMD_BUSY_POLL_US")) {
+      busyPollUs_ = atoi(e);
+    }
     GA_ENFORCE_EQ(uv_loop_init(&loop_), 0);
+    loop_.data = this;
+    // Busy-poll window: while the idle handle is active uv_run polls
+    // with zero timeout; activity (re)arms it, the idle callback
+    // disarms once the window has passed (EpollLoop parity).
+    idle_.data = this;
+    GA_ENFORCE_EQ(uv_idle_init(&loop_, &idle_), 0);
     async_.data = this;
     GA_ENFORCE_EQ(
         uv_async_init(&loop_, &async_,
@@ -52,6 +63,11 @@ class UvLoop : public Loop {
                  });
       }
       polls_.clear();
+      if (idleActive_) {
+        uv_idle_stop(&idle_);
+        idleActive_ = false;
+      }
+      uv_close(reinterpret_cast<uv_handle_t*>(&idle_), nullptr);
       uv_prepare_stop(&prepare_);
       uv_close(reinterpret_cast<uv_handle_t*>(&prepare_), nullptr);
       uv_close(reinterpret_cast<uv_handle_t*>(&async_), nullptr);
@@ -131,9 +147,26 @@ class UvLoop : public Loop {
         }
       }
       if (ev != 0 && h != nullptr) {
+        static_cast<UvLoop*>(p->loop->data)->noteActivity();
         h->handleEvents(ev);
       }
     });
+  }
+
+  // loop thread only
+  void noteActivity() {
+    lastActivityNs_ = uv_hrtime();
+    if (busyPollUs_ > 0 && !idleActive_) {
+      idleActive_ = true;
+      uv_idle_start(&idle_, [](uv_idle_t* i) {
+        auto* self = static_cast<UvLoop*>(i->data);
+        if (uv_hrtime() - self->lastActivityNs_ >
+            static_cast<uint64_t>(self->busyPollUs_) * 1000) {
+          uv_idle_stop(i);
+          self->idleActive_ = false;
+        }
+      });
+    }
   }
 
   void removePoll(int fd) {
@@ -205,6 +238,10 @@ class UvLoop : public Loop {
   uv_loop_t loop_;
   uv_async_t async_;
   uv_prepare_t prepare_;
+  uv_idle_t idle_;
+  bool idleActive_{false}; // loop thread only
+  uint64_t lastActivityNs_{0};
+  int busyPollUs_{200};
   std::thread thread_;
   std::thread::id threadId_;
   std::atomic<bool> done_{false};
