@@ -1,5 +1,7 @@
 // Name-based factory over the legacy Algorithm classes (used by the
-// Python bindings and the benchmark tool).
+// Python bindings and the benchmark tool). Mirrors the reference
+// benchmark's name -> Algorithm registration table
+// (gloo/benchmark/main.cc:920-1068) as a reusable component.
 #pragma once
 
 #include <memory>

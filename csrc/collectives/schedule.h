@@ -1,6 +1,9 @@
 // Shared block/segment math for ring and bcube schedules (used by the
 // CPU v2 collectives and the hip_* device algorithms so both sides of a
-// wire agree on every chunk boundary).
+// wire agree on every chunk boundary). Counterpart of the reference's
+// per-collective segment arithmetic (gloo/allreduce.cc:209-218 segment
+// sizing, gloo/allgather.cc chunk walk), centralized instead of
+// re-derived in every algorithm.
 #pragma once
 
 #include <algorithm>
