@@ -7,8 +7,10 @@
 #include <sys/uio.h>
 #include <unistd.h>
 
+#include <algorithm>
 #include <cstdlib>
 #include <cstring>
+#include <limits>
 
 #include "common/logging.h"
 #include "transport/tcp/context.h"
@@ -388,7 +390,25 @@ void TcpPair::flushTxLocked() {
   if (state_ != CONNECTED || fd_ < 0) {
     return;
   }
+  // Cap the bytes pushed per lock acquisition: with a 16 MiB SO_SNDBUF a
+  // single writev of a large payload memcpys it all into the kernel
+  // while holding the context mutex, stalling the loop thread's RX for
+  // every pair of this context. Past the cap, EPOLLOUT continues the
+  // drain on the loop thread in bounded slices.
+  static const size_t kMaxFlushBytes = [] {
+    if (const char* e = getenv("GLOO_AMD_MAX_FLUSH")) {
+      long v = atol(e);
+      return v <= 0 ? std::numeric_limits<size_t>::max()
+                    : static_cast<size_t>(v);
+    }
+    return static_cast<size_t>(1 << 20);
+  }();
+  size_t flushed = 0;
   while (!tx_.empty()) {
+    if (flushed >= kMaxFlushBytes) {
+      armEpollOutLocked();
+      return;
+    }
     TxOp& op = tx_.front();
     struct iovec iov[2];
     int iovcnt = 0;
@@ -401,7 +421,10 @@ void TcpPair::flushTxLocked() {
     if (op.payloadWritten < op.payloadLen) {
       iov[iovcnt].iov_base =
           const_cast<char*>(op.payload) + op.payloadWritten;
-      iov[iovcnt].iov_len = op.payloadLen - op.payloadWritten;
+      // A single writev into a roomy SO_SNDBUF copies everything at
+      // once; clamp it to the per-flush quota too.
+      iov[iovcnt].iov_len = std::min(
+          op.payloadLen - op.payloadWritten, kMaxFlushBytes - flushed);
       iovcnt++;
     }
     if (iovcnt == 0) {
@@ -423,6 +446,7 @@ void TcpPair::flushTxLocked() {
       return;
     }
     // Account written bytes across preamble then payload.
+    flushed += static_cast<size_t>(n);
     size_t rem = static_cast<size_t>(n);
     size_t preLeft = sizeof(Preamble) - op.preWritten;
     size_t take = std::min(rem, preLeft);
