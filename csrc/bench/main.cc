@@ -34,6 +34,7 @@
 #include "rendezvous/context_factory.h"
 #include "rendezvous/stores.h"
 #include "transport/tcp/device.h"
+#include "transport/tcp/tls.h"
 
 using namespace glooamd;
 
@@ -54,11 +55,14 @@ struct Options {
   bool halfPrecision = false;
   int base = 2;
   bool verify = true;
+  std::string transport = "tcp"; // tcp | tls
+  std::string tlsCert, tlsKey, tlsCa;
 };
 
 [[noreturn]] void usage() {
   fprintf(stderr,
           "gloo_amd_bench --size N --rank R [--store-file DIR | "
+      "[--transport tcp|tls --tls-cert F --tls-key F [--tls-ca F]] "
           "--store-tcp HOST:PORT]\n"
           "  --benchmark NAME --elements N|-1 [--gpu] [--half-precision]\n"
           "  [--iteration-time-ms MS] [--warmup-iters N] [--inputs N] "
@@ -92,6 +96,10 @@ Options parse(int argc, char** argv) {
       {"half-precision", no_argument, nullptr, 'h'},
       {"base", required_argument, nullptr, 'B'},
       {"no-verify", no_argument, nullptr, 'V'},
+      {"transport", required_argument, nullptr, 'X'},
+      {"tls-cert", required_argument, nullptr, 'C'},
+      {"tls-key", required_argument, nullptr, 'K'},
+      {"tls-ca", required_argument, nullptr, 'A'},
       {nullptr, 0, nullptr, 0},
   };
   int c;
@@ -135,6 +143,18 @@ Options parse(int argc, char** argv) {
         break;
       case 'B':
         o.base = atoi(optarg);
+        break;
+      case 'X':
+        o.transport = optarg;
+        break;
+      case 'C':
+        o.tlsCert = optarg;
+        break;
+      case 'K':
+        o.tlsKey = optarg;
+        break;
+      case 'A':
+        o.tlsCa = optarg;
         break;
       case 'V':
         o.verify = false;
@@ -544,8 +564,18 @@ int main(int argc, char** argv) {
     store = std::make_shared<HashStore>();
   }
 
-  tcp::TcpAttr attr;
-  auto device = tcp::createTcpDevice(attr);
+  std::shared_ptr<transport::Device> device;
+  if (o.transport == "tls") {
+    tcp::tls::TlsAttr tattr;
+    tattr.certFile = o.tlsCert;
+    tattr.pkeyFile = o.tlsKey;
+    tattr.caFile = o.tlsCa;
+    device = tcp::tls::createTlsDevice(tattr);
+  } else {
+    GA_ENFORCE(o.transport == "tcp", "unknown --transport ", o.transport);
+    tcp::TcpAttr attr;
+    device = tcp::createTcpDevice(attr);
+  }
   auto ctx = std::make_shared<Context>(o.rank, o.size, o.base);
   ctx->connectFullMesh(*store, device);
 
