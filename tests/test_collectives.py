@@ -410,3 +410,24 @@ def test_alltoall_8(spawn_threads):
         return True
 
     spawn_threads(size, fn)
+
+
+def test_allreduce_per_op_timeout(spawn_threads):
+    """Per-op timeout override (reference allreduce_test.cc:386 parity):
+    a rank alone in the collective times out after ~timeout_ms, not the
+    context default."""
+    import time
+
+    def fn(ctx, rank, _):
+        if rank == 0:
+            x = fixture(0, 1000)
+            t0 = time.monotonic()
+            with pytest.raises(ga.TimeoutError):
+                ga.allreduce(ctx, [x.ctypes.data], 1000, ga.DType.f32,
+                             ga.ReduceOp.sum, timeout_ms=200)
+            assert time.monotonic() - t0 < 5.0  # not the 30s default
+        else:
+            time.sleep(1.0)  # never joins in time; ctx gets poisoned
+        return True
+
+    spawn_threads(2, fn)
