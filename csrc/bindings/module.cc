@@ -164,6 +164,34 @@ PYBIND11_MODULE(_C, m) {
           [](IStore& s, const std::string& key, int64_t delta) {
             py::gil_scoped_release rel;
             return s.add(key, delta);
+          })
+      .def(
+          "multi_set",
+          [](IStore& s,
+             const std::vector<std::string>& keys,
+             const std::vector<py::bytes>& values) {
+            std::vector<std::vector<char>> vals;
+            vals.reserve(values.size());
+            for (const auto& v : values) {
+              std::string d = v;
+              vals.emplace_back(d.begin(), d.end());
+            }
+            py::gil_scoped_release rel;
+            s.multiSet(keys, vals);
+          })
+      .def(
+          "multi_get",
+          [](IStore& s, const std::vector<std::string>& keys) {
+            std::vector<std::vector<char>> vals;
+            {
+              py::gil_scoped_release rel;
+              vals = s.multiGet(keys);
+            }
+            py::list out;
+            for (const auto& v : vals) {
+              out.append(py::bytes(v.data(), v.size()));
+            }
+            return out;
           });
 
   py::class_<HashStore, IStore, std::shared_ptr<HashStore>>(m, "HashStore")

@@ -112,3 +112,9 @@ def test_store_v2_add_concurrent(tmp_path):
     [t.start() for t in ths]
     [t.join() for t in ths]
     assert s.add("ctr", 0) == 4 * per
+
+
+def test_store_multi_ops():
+    s = ga.HashStore()
+    s.multi_set(["a", "b"], [b"1", b"22"])
+    assert s.multi_get(["a", "b"]) == [b"1", b"22"]
