@@ -298,7 +298,9 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
                     inp.numel() // self.size(), inp.element_size(),
                     stream=_cur_stream(out))
             return _ret_work(output)
-        h_out = out.cpu() if out.is_cuda else out
+        out_staged = out.is_cuda or not out.is_contiguous()
+        h_out = (torch.empty(out.numel(), dtype=out.dtype)
+                 if out_staged else out)
         h_in = inp.cpu() if inp.is_cuda else inp
         with self._lock:
             tag = self._tag()
@@ -312,8 +314,8 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
                 out_counts = [int(s) * row for s in output_split_sizes]
                 ga.alltoallv(self._ctx, h_out.data_ptr(), h_in.data_ptr(),
                              in_counts, out_counts, _gdtype(inp), tag=tag)
-        if out.is_cuda:
-            out.copy_(h_out)
+        if out_staged:
+            out.copy_(h_out.view_as(out))
         return _ret_work(output)
 
     def alltoall(self, output_tensors, input_tensors, opts=None):
@@ -401,13 +403,13 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
     def recv(self, tensors, src_rank, tag=0):
         for t in tensors:
             t_ = t.detach()
-            h = torch.empty(
-                t_.numel(), dtype=t_.dtype) if t_.is_cuda else t_
+            staged = t_.is_cuda or not t_.is_contiguous()
+            h = torch.empty(t_.numel(), dtype=t_.dtype) if staged else t_
             ub = self._ctx.create_unbound_buffer(
                 h.data_ptr(), h.numel() * h.element_size())
             ub.recv(src_rank, self._p2p_slot(tag))
             ub.wait_recv()
-            if t_.is_cuda:
+            if staged:
                 t_.copy_(h.view_as(t_))
         return _ret_work(tensors)
 

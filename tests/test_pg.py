@@ -197,3 +197,24 @@ def test_init_process_group_ddp():
         outs.append(out.decode())
     assert all(p.returncode == 0 for p in procs), "\n".join(outs)
     assert all("DDP-OK" in o for o in outs), "\n".join(outs)
+
+
+def test_pg_noncontiguous_tensors():
+    """Collectives and p2p on views/transposes (staged automatically)."""
+
+    def fn(pg, rank, size):
+        m = torch.arange(16, dtype=torch.float32).reshape(4, 4) + rank
+        col = m[:, 1]  # stride-4 view
+        pg.allreduce([col]).wait()
+        expect = (torch.arange(16, dtype=torch.float32).reshape(4, 4)[:, 1]
+                  * size + sum(range(size)))
+        assert torch.allclose(col, expect), col
+        tr = (torch.arange(9, dtype=torch.float32).reshape(3, 3) * 0 + rank).T
+        if rank == 0:
+            pg.send([torch.full((3, 3), 7.0).T.contiguous()], 1, tag=3).wait()
+        elif rank == 1:
+            pg.recv([tr], 0, tag=3).wait()
+            assert torch.allclose(tr, torch.full((3, 3), 7.0)), tr
+        return True
+
+    spawn_pg(2, fn)
