@@ -150,8 +150,9 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
         return self._hip_a2a[device]
 
     def _staged(self, t, fn):
-        """Run a CPU collective on a host copy of a CUDA tensor."""
-        host = t.detach().cpu()
+        """Run a CPU collective on a contiguous host copy of t (CUDA or
+        non-contiguous CPU), writing the result back."""
+        host = t.detach().contiguous().cpu()
         fn(host)
         t.detach().copy_(host)
 
@@ -246,7 +247,7 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
                           tag=self._tag())
 
             with self._lock:
-                if t_.is_cuda:
+                if t_.is_cuda or not t_.is_contiguous():
                     self._staged(t_, run)
                 else:
                     run(t_)
