@@ -431,3 +431,46 @@ def test_allreduce_per_op_timeout(spawn_threads):
         return True
 
     spawn_threads(2, fn)
+
+
+def test_all_collectives_torture_8rank(spawn_threads):
+    """Integration: every v2 collective interleaved for many rounds on
+    one 8-rank context (slot/tag hygiene across op types)."""
+    size, n, iters = 8, 512, 8
+
+    def fn(ctx, rank, _):
+        for it in range(iters):
+            x = fixture(rank + it, n)
+            ga.allreduce(ctx, [x.ctypes.data], n, ga.DType.f32,
+                         ga.ReduceOp.sum, tag=1)
+            assert np.allclose(
+                x, sum(fixture(r + it, n) for r in range(size)))
+
+            inp = fixture(rank, n)
+            out = np.zeros(n * size, dtype=np.float32)
+            ga.allgather(ctx, out.ctypes.data, inp.ctypes.data, n, tag=2)
+            assert np.allclose(out[rank * n:(rank + 1) * n], inp)
+
+            b = fixture(it, n) if rank == it % size else np.zeros(
+                n, dtype=np.float32)
+            ga.broadcast(ctx, b.ctypes.data, 0, n, ga.DType.f32,
+                         root=it % size, tag=3)
+            assert np.allclose(b, fixture(it, n))
+
+            a2a_in = np.concatenate(
+                [fixture(rank * size + d, 16) for d in range(size)])
+            a2a_out = np.zeros(16 * size, dtype=np.float32)
+            ga.alltoall(ctx, a2a_out.ctypes.data, a2a_in.ctypes.data, 16,
+                        tag=4)
+            assert np.allclose(a2a_out[:16], fixture(rank, 16))
+
+            r = fixture(rank, n)
+            ga.reduce(ctx, r.ctypes.data, r.ctypes.data, n, ga.DType.f32,
+                      ga.ReduceOp.max, root=0, tag=5)
+            if rank == 0:
+                assert np.allclose(r, fixture(size - 1, n))
+
+            ga.barrier(ctx, tag=6)
+        return True
+
+    spawn_threads(size, fn)
