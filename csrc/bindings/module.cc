@@ -96,6 +96,26 @@ PYBIND11_MODULE(_C, m) {
   m.def("interface_speed", &getInterfaceSpeedByName);
   m.def("interface_to_bus_id", &interfaceToBusID);
   m.def("pci_distance", &pciDistance);
+  m.def("gpu_pci_bus_id", &hip::gpuPCIBusID);
+  // NIC nearest a GPU: reference findCudaDevicePointerClosestToDevice
+  // inverted for the one-NIC-per-GPU-process launch pattern.
+  m.def("closest_interface_to_gpu", [](int device) {
+    std::string gpuBus = hip::gpuPCIBusID(device);
+    std::string best;
+    int bestDist = 1 << 30;
+    for (const auto& ifname : listInterfaces()) {
+      std::string bus = interfaceToBusID(ifname);
+      if (bus.empty()) {
+        continue;
+      }
+      int d = pciDistance(bus, gpuBus);
+      if (d < bestDist) {
+        bestDist = d;
+        best = ifname;
+      }
+    }
+    return best;
+  });
 
   // --- stores ---------------------------------------------------------------
   // Trampoline so Python classes (e.g. a torch.distributed Store adapter)

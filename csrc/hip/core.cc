@@ -23,6 +23,12 @@ int deviceCount() {
   return n;
 }
 
+std::string gpuPCIBusID(int device) {
+  char buf[64] = {0};
+  GA_HIP_CHECK(hipDeviceGetPCIBusId(buf, sizeof(buf), device));
+  return std::string(buf);
+}
+
 HipStream::HipStream(int device, bool highPriority) : device_(device) {
   if (device_ >= 0) {
     GA_HIP_CHECK(hipSetDevice(device_));

@@ -39,6 +39,9 @@ std::mutex& allocMutex();
 
 bool available(); // any HIP device present (never throws)
 int deviceCount();
+// PCI bus id of a GPU ("0000:03:00.0"), for NIC-locality decisions
+// (reference cuda_private.cu:96-114 getCudaPCIBusID parity).
+std::string gpuPCIBusID(int device);
 
 class HipStream {
  public:

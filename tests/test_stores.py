@@ -118,3 +118,12 @@ def test_store_multi_ops():
     s = ga.HashStore()
     s.multi_set(["a", "b"], [b"1", b"22"])
     assert s.multi_get(["a", "b"]) == [b"1", b"22"]
+
+
+def test_pci_distance_helpers():
+    """Topology helpers (reference common/linux.cc + cuda_private.cu
+    PCI locality)."""
+    near = ga._C.pci_distance("0000:03:00.0", "0000:03:00.1")
+    far = ga._C.pci_distance("0000:03:00.0", "0001:44:00.0")
+    assert near < far
+    assert isinstance(ga._C.list_interfaces(), list)
