@@ -675,6 +675,11 @@ void HipAllreduceDirect::run(
   const int C =
       std::max<int>(1, static_cast<int>((perRank * es + chunkCap - 1) /
                                         chunkCap));
+  // The (src, parity) slot grid needs 2*P*chunkCap bytes. The mesh lays
+  // its two inboxes out contiguously (mesh.h: work | inbox0 | inbox1),
+  // so asking for P*chunkCap PER inbox and indexing the grid from
+  // inbox(0) uses both regions as one arena. This mesh is private to
+  // this algorithm instance, so inbox(1) has no other user.
   mesh_->ensureCapacity(elements * es, P * chunkCap);
   char* work = mesh_->work();
   const size_t slotStride = chunkCap; // one chunk per (src, parity) slot
