@@ -9,6 +9,7 @@
 // lock"/"data race" reports. setTimeout(-1) below routes all transport
 // waits through plain pthread_cond_wait (intercepted). Verified clean
 // (0 warnings, 3 runs) as of the eager-send + busy-poll protocol.
+#include <cstdlib>
 #include <thread>
 #include <vector>
 #include <cstdio>
@@ -22,7 +23,11 @@ using namespace glooamd;
 
 int main() {
   auto store = std::make_shared<HashStore>();
-  auto dev = tcp::createTcpDevice(tcp::TcpAttr{});
+  tcp::TcpAttr attr;
+  if (getenv("GLOO_AMD_STRESS_UV")) {
+    attr.useLibuv = true; // exercise the libuv loop under the sanitizer
+  }
+  auto dev = tcp::createTcpDevice(attr);
   const int P = 2;
   std::vector<std::thread> ths;
   for (int r = 0; r < P; r++) {

@@ -196,10 +196,10 @@ class UvLoop : public Loop {
     bool doneFlag = false;
     defer([&] {
       fn();
-      {
-        std::lock_guard<std::mutex> lock(m);
-        doneFlag = true;
-      }
+      // Notify while holding the mutex: the waiter re-acquires it before
+      // returning, so the stack cv cannot be destroyed mid-broadcast.
+      std::lock_guard<std::mutex> lock(m);
+      doneFlag = true;
       cv.notify_all();
     });
     std::unique_lock<std::mutex> lock(m);
