@@ -398,3 +398,44 @@ def test_eager_zero_length(spawn_threads):
         return True
 
     spawn_threads(2, fn)
+
+
+def test_connect_rendezvous_timeout():
+    """Bootstrap with an absent peer fails within the context timeout
+    (reference tcp_test.cc connect-timeout unit)."""
+    import time
+
+    store = ga.HashStore()
+    dev = ga.create_tcp_device()
+    ctx = ga.Context(0, 2)
+    ctx.set_timeout(500)
+    t0 = time.monotonic()
+    with pytest.raises(ga.GlooAmdError):
+        ctx.connect_full_mesh(store, dev)
+    assert time.monotonic() - t0 < 10.0
+
+
+def test_unbound_buffer_lifetime(spawn_threads):
+    """Destroying an unbound buffer with an unmatched posted op must
+    drop it cleanly (reference memory_test.cc semantics)."""
+
+    def fn(ctx, rank, size):
+        n = 256
+        # recv that will never match: destroyed before any send exists
+        orphan = np.zeros(n, dtype=np.float32)
+        ub = ctx.create_unbound_buffer(orphan.ctypes.data, orphan.nbytes)
+        ub.recv((rank + 1) % size, slot=77)
+        del ub  # queued op dropped; no crash, no spurious delivery later
+        # the slot remains usable for fresh matched traffic
+        x = np.full(n, float(rank), dtype=np.float32)
+        y = np.zeros(n, dtype=np.float32)
+        us = ctx.create_unbound_buffer(x.ctypes.data, x.nbytes)
+        ur = ctx.create_unbound_buffer(y.ctypes.data, y.nbytes)
+        ur.recv((rank + 1) % size, slot=78)
+        us.send((rank + 1) % size, slot=78)
+        ur.wait_recv()
+        us.wait_send()
+        assert np.all(y == float((rank + 1) % size))
+        return True
+
+    spawn_threads(2, fn)
