@@ -33,7 +33,7 @@ def main():
     ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--elements", type=int, default=5_000_000)
     ap.add_argument("--algorithm", default="ring_chunked",
-                    choices=["ring", "ring_chunked", "halving_doubling"])
+                    choices=["ring", "ring_chunked", "halving_doubling", "direct"])
     ap.add_argument("--dtype", default="f32", choices=["f32", "bf16", "f16"])
     ap.add_argument("--sweep", action="store_true",
                     help="also print the reference element sweep to stderr")
@@ -90,7 +90,9 @@ def main():
                      buf2.data_ptr()],
                     n, gdt, ga.ReduceOp.sum, local_rank)
         else:
-            if (args.algorithm == "halving_doubling"
+            if args.algorithm == "direct" and world <= 8:
+                algo = ga._C.HipAllreduceDirect(ctx, local_rank)
+            elif (args.algorithm == "halving_doubling"
                     and world & (world - 1) == 0):
                 algo = ga._C.HipAllreduceHalvingDoubling(ctx, local_rank)
             else:

@@ -91,9 +91,12 @@ HipAllreduceRing::HipAllreduceRing(
   // Ring strides coprime to size: each ring's neighbor hop uses a
   // different xGMI link, so the rings' wire traffic runs concurrently.
   const int P = ctx_->size;
+  if (numRings == 0) {
+    // tunable without a rebuild for link-level experiments on hardware
+    numRings = static_cast<int>(getEnvInt("GLOO_AMD_NUM_RINGS", 4));
+  }
   const int maxRings =
-      std::max(1, std::min(numRings == 0 ? 4 : numRings,
-                           kStreamPoolSize / 2));
+      std::max(1, std::min(numRings, kStreamPoolSize / 2));
   for (int st = 1; st < std::max(P, 2) &&
        static_cast<int>(strides_.size()) < maxRings;
        st++) {
