@@ -12,6 +12,7 @@
 
 #include "algorithms/factory.h"
 #include "collectives/collectives.h"
+#include "collectives/schedule.h"
 #include "collectives/reduce_fns.h"
 #include "common/linux.h"
 #include "common/store.h"
@@ -90,6 +91,21 @@ PYBIND11_MODULE(_C, m) {
       .value("max", ReduceOp::MAX);
 
   m.def("dtype_size", &dtypeSize);
+
+  // schedule math (shared by CPU collectives and device engines); bound
+  // for property tests — both sides of a wire must agree on boundaries.
+  m.def("_block_of", [](size_t n, int p, int b) {
+    auto s = sched::blockOf(n, p, b);
+    return std::make_pair(s.off, s.len);
+  });
+  m.def("_segment_of", [](size_t n, int p, int b, int q, int S) {
+    auto s = sched::segmentOf(n, p, b, q, S);
+    return std::make_pair(s.off, s.len);
+  });
+  m.def("_subspan_of", [](size_t off, size_t len, int j, int base) {
+    auto s = sched::subspanOf({off, len}, j, base);
+    return std::make_pair(s.off, s.len);
+  });
 
   // topology helpers (common/linux)
   m.def("list_interfaces", &listInterfaces);
