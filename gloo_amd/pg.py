@@ -13,6 +13,7 @@ This is the MI355X counterpart of the reference's role as the `gloo`
 backend of torch.distributed.
 """
 import threading
+from concurrent.futures import ThreadPoolExecutor
 from datetime import timedelta
 
 import torch
@@ -96,6 +97,10 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
         self._ctx.set_timeout(int(timeout.total_seconds() * 1000))
         self._ctx.connect_full_mesh(self._store, ga.create_tcp_device())
         self._lock = threading.Lock()
+        # One helper thread: p2p waits complete in post order off the
+        # caller's thread (isend/irecv return pending Works).
+        self._p2p_pool = ThreadPoolExecutor(
+            max_workers=1, thread_name_prefix="glooamd-p2p")
         self._hip_ring = {}  # device -> HipAllreduceRing
         self._hip_bcast = {}  # (device, root) -> HipBroadcastOneToAll
         self._hip_ag = {}  # device -> HipAllgatherRing
@@ -391,17 +396,44 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
             ga.barrier(self._ctx, tag=self._tag())
         return _ret_work(True)
 
+    def _p2p_async(self, finish):
+        """Run finish() (a blocking wait + optional copy-back) on the p2p
+        helper thread and return a torch Work tied to it. Keeps send and
+        recv truly asynchronous so batch_isend_irecv-style patterns
+        (post everything, then wait) cannot deadlock on the rendezvous."""
+        fut = Future()
+
+        def runner():
+            try:
+                finish()
+                fut.set_result(None)
+            except Exception as e:  # noqa: BLE001
+                try:
+                    fut.set_exception(e)
+                except Exception:  # pragma: no cover
+                    pass
+
+        self._p2p_pool.submit(runner)
+        return _create_work_from_future(fut)
+
     def send(self, tensors, dst_rank, tag=0):
+        ubs = []
         for t in tensors:
             t_ = t.detach().contiguous()
             h = t_.cpu() if t_.is_cuda else t_
             ub = self._ctx.create_unbound_buffer(
                 h.data_ptr(), h.numel() * h.element_size())
             ub.send(dst_rank, self._p2p_slot(tag))
-            ub.wait_send()
-        return _ret_work(tensors)
+            ubs.append((ub, h))  # keep the host staging alive until done
+
+        def finish():
+            for ub, _h in ubs:
+                ub.wait_send()
+
+        return self._p2p_async(finish)
 
     def recv(self, tensors, src_rank, tag=0):
+        posted = []
         for t in tensors:
             t_ = t.detach()
             staged = t_.is_cuda or not t_.is_contiguous()
@@ -409,10 +441,15 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
             ub = self._ctx.create_unbound_buffer(
                 h.data_ptr(), h.numel() * h.element_size())
             ub.recv(src_rank, self._p2p_slot(tag))
-            ub.wait_recv()
-            if staged:
-                t_.copy_(h.view_as(t_))
-        return _ret_work(tensors)
+            posted.append((ub, h, t_, staged))
+
+        def finish():
+            for ub, h, t_, staged in posted:
+                ub.wait_recv()
+                if staged:
+                    t_.copy_(h.view_as(t_))
+
+        return self._p2p_async(finish)
 
     @staticmethod
     def _p2p_slot(tag):

@@ -54,6 +54,18 @@ def main():
         req.wait()
         assert torch.allclose(g2, torch.full((4,), 7.0))
 
+    # batch_isend_irecv (pipeline-parallel shape): both ranks post their
+    # sends before any recv; large payloads force the rendezvous, so this
+    # deadlocks unless isend/irecv return genuinely pending Works.
+    big = 200_000
+    send_t = torch.full((big,), float(rank))
+    recv_t = torch.zeros(big)
+    ops = [dist.P2POp(dist.isend, send_t, (rank + 1) % size),
+           dist.P2POp(dist.irecv, recv_t, (rank - 1) % size)]
+    for req in dist.batch_isend_irecv(ops):
+        req.wait()
+    assert torch.allclose(recv_t, torch.full((big,), float((rank - 1) % size)))
+
     # fused-tensor collectives the dispatcher lowers to *_base
     out = torch.zeros(size * 8)
     dist.all_gather_into_tensor(out, torch.full((8,), float(rank)))
