@@ -474,3 +474,17 @@ def test_all_collectives_torture_8rank(spawn_threads):
         return True
 
     spawn_threads(size, fn)
+
+
+def test_allreduce_16rank(spawn_threads):
+    """Rank-count headroom beyond one node (reference sweeps to 16/32):
+    16 thread-ranks, full mesh = 120 pairs in one process."""
+
+    def fn(ctx, rank, _):
+        x = fixture(rank, 5000)
+        ga.allreduce(ctx, [x.ctypes.data], 5000, ga.DType.f32,
+                     ga.ReduceOp.sum)
+        assert np.allclose(x, sum(fixture(r, 5000) for r in range(16)))
+        return True
+
+    spawn_threads(16, fn)
