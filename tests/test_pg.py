@@ -218,3 +218,17 @@ def test_pg_noncontiguous_tensors():
         return True
 
     spawn_pg(2, fn)
+
+
+def test_pg_lifecycle_reuse():
+    """Create, use, destroy and re-create process groups in one process
+    (device/loop/context teardown hygiene)."""
+
+    def fn(pg, rank, size):
+        t = torch.full((256,), float(rank + 1))
+        pg.allreduce([t]).wait()
+        assert torch.allclose(t, torch.full((256,), float(sum(range(1, size + 1)))))
+        return True
+
+    for _ in range(3):
+        spawn_pg(2, fn)
