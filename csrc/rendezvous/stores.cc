@@ -394,6 +394,13 @@ class TcpStore::Server {
           cv_.notify_all();
         }
         writeAll(fd, &result, 8);
+      } else if (op == 'C') {
+        uint8_t exists;
+        {
+          std::lock_guard<std::mutex> lock(mu_);
+          exists = map_.count(key) > 0 ? 1 : 0;
+        }
+        writeAll(fd, &exists, 1);
       } else if (op == 'G' || op == 'W') {
         std::unique_lock<std::mutex> lock(mu_);
         cv_.wait(lock, [&] { return map_.count(key) > 0 || done_.load(); });
@@ -528,14 +535,30 @@ std::vector<char> TcpStore::get(const std::string& key) {
 void TcpStore::wait(
     const std::vector<std::string>& keys,
     const std::chrono::milliseconds& timeout) {
-  (void)timeout; // server-side blocking; client socket has no deadline yet
-  std::lock_guard<std::mutex> lock(clientMu_);
+  // Poll the non-blocking check op so the client honors its deadline
+  // (a blocked server-side 'W' would leave a stray reply in the stream
+  // if the client gave up). Negative timeout = wait forever.
+  auto deadline = std::chrono::steady_clock::now() + timeout;
   for (const auto& key : keys) {
-    uint8_t op = 'W';
-    writeAll(clientFd_, &op, 1);
-    writeString(clientFd_, key);
-    uint8_t ok;
-    GA_ENFORCE(readAll(clientFd_, &ok, 1), "TcpStore: server closed");
+    for (;;) {
+      uint8_t exists;
+      {
+        std::lock_guard<std::mutex> lock(clientMu_);
+        uint8_t op = 'C';
+        writeAll(clientFd_, &op, 1);
+        writeString(clientFd_, key);
+        GA_ENFORCE(
+            readAll(clientFd_, &exists, 1), "TcpStore: server closed");
+      }
+      if (exists) {
+        break;
+      }
+      if (timeout.count() >= 0 &&
+          std::chrono::steady_clock::now() > deadline) {
+        throw TimeoutException("TcpStore::wait timeout for key " + key);
+      }
+      usleep(10 * 1000);
+    }
   }
 }
 

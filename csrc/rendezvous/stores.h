@@ -80,9 +80,9 @@ class PrefixStore : public IStore {
 
 // TCP key/value store. One process runs the server (isServer=true, usually
 // rank 0); every process (including the server's own) connects as a client.
-// Wire format: u8 op ('S','G','W','A','I') + u32 keylen + key
+// Wire format: u8 op ('S','G','W','C','A','I') + u32 keylen + key
 // [+ u32 vallen + val | i64 delta]; GET blocks server-side until the
-// key exists; 'A' appends, 'I' atomically adds and returns the value.
+// key exists; 'C' checks without blocking; 'A' appends, 'I' atomically adds and returns the value.
 class TcpStore : public IStore {
  public:
   TcpStore(

@@ -134,3 +134,18 @@ def test_pci_distance_helpers():
         assert ga._C.pci_distance(devs[0], devs[0]) == 0
     assert isinstance(ga._C.list_interfaces(), list)
     assert hasattr(ga._C, "closest_interface_to_gpu")
+
+
+def test_tcp_store_wait_timeout():
+    """TcpStore.wait honors its deadline for absent keys."""
+    import random
+    import time
+
+    port = random.randint(20000, 40000)
+    s = ga.TcpStore("127.0.0.1", port, is_server=True)
+    t0 = time.monotonic()
+    with pytest.raises(ga.TimeoutError):
+        s.wait(["never-set"], timeout_ms=300)
+    assert 0.2 < time.monotonic() - t0 < 5.0
+    s.set("later", b"x")
+    s.wait(["later"], timeout_ms=1000)  # present key returns fast
