@@ -101,6 +101,7 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
         # caller's thread (isend/irecv return pending Works).
         self._p2p_pool = ThreadPoolExecutor(
             max_workers=1, thread_name_prefix="glooamd-p2p")
+        self._mb_seq = 0
         self._hip_ring = {}  # device -> HipAllreduceRing
         self._hip_bcast = {}  # (device, root) -> HipBroadcastOneToAll
         self._hip_ag = {}  # device -> HipAllgatherRing
@@ -395,6 +396,47 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
         with self._lock:
             ga.barrier(self._ctx, tag=self._tag())
         return _ret_work(True)
+
+    def monitored_barrier(self, timeout=None, wait_all_ranks=False):
+        """Rank 0 collects an ack from every rank and reports WHICH ranks
+        failed to arrive (the reference backend's role for this API in
+        torch.distributed)."""
+        tmo_ms = int(timeout.total_seconds() * 1000) if timeout else 30000
+        rank, size = self.rank(), self.size()
+        if size == 1:
+            return
+        with self._lock:
+            slot = self._p2p_slot(0xB0000 + (self._mb_seq % 4096))
+            self._mb_seq += 1
+            byte = torch.zeros(1, dtype=torch.uint8)
+            if rank == 0:
+                missing = []
+                for src in range(1, size):
+                    try:
+                        ub = self._ctx.create_unbound_buffer(
+                            byte.data_ptr(), 1)
+                        ub.recv(src, slot, 0, 0)
+                        ub.wait_recv(tmo_ms)
+                    except ga.GlooAmdError:
+                        # timed out, or the peer already died (closed pair)
+                        missing.append(src)
+                        if not wait_all_ranks:
+                            break
+                if missing:
+                    raise RuntimeError(
+                        "monitored_barrier: rank(s) "
+                        f"{missing} failed to arrive within {tmo_ms}ms")
+                for dst in range(1, size):
+                    ub = self._ctx.create_unbound_buffer(byte.data_ptr(), 1)
+                    ub.send(dst, slot + 1, 0, 0)
+                    ub.wait_send(tmo_ms)
+            else:
+                ub = self._ctx.create_unbound_buffer(byte.data_ptr(), 1)
+                ub.send(0, slot, 0, 0)
+                ub.wait_send(tmo_ms)
+                ub2 = self._ctx.create_unbound_buffer(byte.data_ptr(), 1)
+                ub2.recv(0, slot + 1, 0, 0)
+                ub2.wait_recv(tmo_ms)
 
     def _p2p_async(self, finish):
         """Run finish() (a blocking wait + optional copy-back) on the p2p

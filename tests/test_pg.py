@@ -232,3 +232,38 @@ def test_pg_lifecycle_reuse():
 
     for _ in range(3):
         spawn_pg(2, fn)
+
+
+def test_pg_monitored_barrier():
+    def fn(pg, rank, size):
+        import time
+
+        time.sleep(0.02 * rank)  # staggered arrival
+        pg.monitored_barrier(timeout=timedelta(seconds=10))
+        return True
+
+    from datetime import timedelta  # noqa: F811
+    spawn_pg(3, fn)
+
+
+def test_pg_monitored_barrier_reports_missing():
+    """Rank 0 names the rank that never arrived."""
+    from datetime import timedelta
+
+    def fn(pg, rank, size):
+        if rank == 2:
+            return True  # never joins
+        if rank == 0:
+            try:
+                pg.monitored_barrier(timeout=timedelta(milliseconds=400))
+                raise AssertionError("expected monitored_barrier to fail")
+            except RuntimeError as e:
+                assert "2" in str(e), e
+        else:
+            try:
+                pg.monitored_barrier(timeout=timedelta(seconds=5))
+            except Exception:
+                pass  # context poisoned by rank 0's timeout is fine
+        return True
+
+    spawn_pg(3, fn)
