@@ -54,6 +54,19 @@ def main():
         req.wait()
         assert torch.allclose(g2, torch.full((4,), 7.0))
 
+    # object gather/scatter through gather()/scatter()
+    got = [None] * size if rank == 0 else None
+    dist.gather_object({"r": rank}, got, dst=0)
+    if rank == 0:
+        assert [g["r"] for g in got] == list(range(size))
+    out_obj = [None]
+    scatter_src = [("obj", i) for i in range(size)] if rank == 0 else None
+    dist.scatter_object_list(out_obj, scatter_src, src=0)
+    assert out_obj[0] == ("obj", rank)
+
+    # monitored barrier (success path)
+    dist.monitored_barrier()
+
     # batch_isend_irecv (pipeline-parallel shape): both ranks post their
     # sends before any recv; large payloads force the rendezvous, so this
     # deadlocks unless isend/irecv return genuinely pending Works.
