@@ -64,8 +64,10 @@ def main():
     dist.scatter_object_list(out_obj, scatter_src, src=0)
     assert out_obj[0] == ("obj", rank)
 
-    # monitored barrier (success path)
-    dist.monitored_barrier()
+    # monitored barrier (success path). The module-level wrapper
+    # hard-codes the "gloo" backend name, so call the ProcessGroup
+    # method our backend implements.
+    dist.distributed_c10d._get_default_group().monitored_barrier()
 
     # batch_isend_irecv (pipeline-parallel shape): both ranks post their
     # sends before any recv; large payloads force the rendezvous, so this
