@@ -11,6 +11,10 @@ versions are per-op upgrades, not API changes).
 
 This is the MI355X counterpart of the reference's role as the `gloo`
 backend of torch.distributed.
+
+Note: `torch.distributed.monitored_barrier()` hard-codes the "gloo"
+backend name; call the ProcessGroup method instead:
+`dist.distributed_c10d._get_default_group().monitored_barrier()`.
 """
 import threading
 from concurrent.futures import ThreadPoolExecutor
