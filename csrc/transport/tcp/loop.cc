@@ -135,6 +135,16 @@ void EpollLoop::run() {
       fn();
     }
   }
+  // Drain cleanup work queued during shutdown (e.g. a Device dropping
+  // pending unrouted connections) — skipping it would leak their fds.
+  std::vector<std::function<void()>> fns;
+  {
+    std::lock_guard<std::mutex> lock(mu_);
+    fns.swap(deferred_);
+  }
+  for (auto& fn : fns) {
+    fn();
+  }
   // final tick so waiters don't hang at shutdown
   {
     std::lock_guard<std::mutex> lock(mu_);
