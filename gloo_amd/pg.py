@@ -161,10 +161,16 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
 
     def _staged(self, t, fn):
         """Run a CPU collective on a contiguous host copy of t (CUDA or
-        non-contiguous CPU), writing the result back."""
-        host = t.detach().contiguous().cpu()
+        non-contiguous CPU), writing the result back. CUDA tensors stage
+        through pinned memory (sync DMA instead of pageable copies)."""
+        t_ = t.detach()
+        if t_.is_cuda:
+            host = torch.empty(t_.numel(), dtype=t_.dtype, pin_memory=True)
+            host.copy_(t_.reshape(-1))
+        else:
+            host = t_.contiguous()
         fn(host)
-        t.detach().copy_(host)
+        t_.copy_(host.view_as(t_))
 
     # -- collectives ---------------------------------------------------------
 
@@ -483,7 +489,8 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
         for t in tensors:
             t_ = t.detach()
             staged = t_.is_cuda or not t_.is_contiguous()
-            h = torch.empty(t_.numel(), dtype=t_.dtype) if staged else t_
+            h = (torch.empty(t_.numel(), dtype=t_.dtype,
+                             pin_memory=t_.is_cuda) if staged else t_)
             ub = self._ctx.create_unbound_buffer(
                 h.data_ptr(), h.numel() * h.element_size())
             ub.recv(src_rank, self._p2p_slot(tag))
