@@ -374,6 +374,44 @@ def test_pg_cuda_collectives():
             pg.broadcast([b]).wait()
             assert torch.allclose(
                 _host(b), torch.arange(50_000, dtype=torch.float32))
+
+            # host-staged CUDA paths: reduce, gather, scatter, p2p
+            red = torch.full((30_000,), float(rank + 1), device="cuda")
+            import torch.distributed as dist
+
+            class _RO:
+                rootRank = 0
+                reduceOp = dist.ReduceOp.SUM
+            pg.reduce([red], _RO()).wait()
+            if rank == 0:
+                assert torch.all(red == 3.0)
+
+            g_in = torch.full((10_000,), float(rank + 5), device="cuda")
+            if rank == 0:
+                gouts = [[torch.zeros(10_000, device="cuda")
+                          for _ in range(2)]]
+                pg.gather(gouts, [g_in]).wait()
+                assert torch.all(gouts[0][0] == 5) and torch.all(
+                    gouts[0][1] == 6)
+            else:
+                pg.gather([], [g_in]).wait()
+
+            s_out = torch.zeros(10_000, device="cuda")
+            if rank == 0:
+                s_ins = [[torch.full((10_000,), float(10 + d),
+                                     device="cuda") for d in range(2)]]
+                pg.scatter([s_out], s_ins).wait()
+            else:
+                pg.scatter([s_out], []).wait()
+            assert torch.all(s_out == 10 + rank)
+
+            if rank == 0:
+                pg.send([torch.full((20_000,), 9.0, device="cuda")],
+                        1, tag=4).wait()
+            else:
+                pr = torch.zeros(20_000, device="cuda")
+                pg.recv([pr], 0, tag=4).wait()
+                assert torch.all(pr == 9.0)
         except Exception:  # noqa: BLE001
             import traceback
 
