@@ -255,7 +255,10 @@ def test_pg_monitored_barrier_reports_missing():
             return True  # never joins
         if rank == 0:
             try:
-                pg.monitored_barrier(timeout=timedelta(milliseconds=400))
+                # wait_all_ranks so a slow-but-alive rank 1 cannot mask
+                # the genuinely missing rank 2 in the report
+                pg.monitored_barrier(timeout=timedelta(milliseconds=800),
+                                     wait_all_ranks=True)
                 raise AssertionError("expected monitored_barrier to fail")
             except RuntimeError as e:
                 assert "2" in str(e), e
