@@ -3,6 +3,7 @@
 #include <cstring>
 
 #include "common/logging.h"
+#include "common/utils.h"
 #include "transport/tcp/device.h"
 
 namespace glooamd {
@@ -117,8 +118,62 @@ void TcpContext::connectFullMesh(IStore& store) {
   }
 }
 
+static std::string detail2(const std::unordered_map<uint64_t, uint64_t>& m) {
+  std::string d;
+  for (auto& kv : m) {
+    if (kv.second != 0) {
+      d += " " + std::to_string(kv.first) + ":" + std::to_string(kv.second);
+    }
+  }
+  return d.empty() ? std::string(" -") : d;
+}
+
+void TcpContext::dumpStateLocked(const char* why) {
+  // GLOO_AMD_DUMP_ON_TIMEOUT aid: one line per pair with queue depths.
+  std::string out = std::string("ctx rank ") + std::to_string(rank) +
+      " dump (" + why + ")\n";
+  for (auto& p : pairs_) {
+    if (!p) {
+      continue;
+    }
+    out += "  " + p->str() + " tx=" + std::to_string(p->tx_.size());
+    out += p->error_ ? " FAILED" : "";
+    auto sum = [](auto& m) {
+      size_t t = 0;
+      for (auto& kv : m) {
+        t += kv.second.size();
+      }
+      return t;
+    };
+    auto detail = [](auto& m) {
+      std::string d;
+      for (auto& kv : m) {
+        if (!kv.second.empty()) {
+          d += " " + std::to_string(kv.first) + ":" +
+              std::to_string(kv.second.size());
+        }
+      }
+      return d.empty() ? std::string(" -") : d;
+    };
+    out += " pendSend[" + detail(p->localPendingSend_) + " ]";
+    out += " pendRecv[" + detail(p->localPendingRecv_) + " ]";
+    out += " stash[" + detail(p->eagerStash_) + " ]";
+    out += " credits[" + detail2(p->remoteRecvCredits_) + " ]";
+    out += " Sn[" + detail2(p->sendReadySeen_) + " ]";
+    out += " R[" + detail2(p->recvsPosted_) + " ]";
+    out += " epollOut=" + std::to_string(p->epollOutArmed_ ? 1 : 0);
+    out += " rxActive=" + std::to_string(p->rxActive_ ? 1 : 0);
+    out += "\n";
+  }
+  out += "  anyRecvs=" + std::to_string(anyRecvs_.size());
+  GA_ERROR << out;
+}
+
 void TcpContext::signalException(const std::string& msg) {
   std::lock_guard<std::mutex> lock(mu_);
+  if (getEnvFlag("GLOO_AMD_DUMP_ON_TIMEOUT")) {
+    dumpStateLocked(msg.c_str());
+  }
   auto e = std::make_exception_ptr(IoException(msg));
   for (auto& p : pairs_) {
     if (p) {

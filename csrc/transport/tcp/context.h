@@ -68,6 +68,7 @@ class TcpContext : public transport::Context,
 
   // Close all pairs and fail every pending operation with IoException.
   void signalException(const std::string& msg);
+  void dumpStateLocked(const char* why); // GLOO_AMD_DUMP_ON_TIMEOUT aid
 
   TcpDevice* device() {
     return device_.get();

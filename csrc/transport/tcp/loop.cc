@@ -8,6 +8,7 @@
 #include <chrono>
 #include <cstdlib>
 #include <cstring>
+#include <thread>
 
 #include "common/logging.h"
 
@@ -110,6 +111,11 @@ void EpollLoop::run() {
       break;
     }
     if (n == 0 && spinning) {
+      // Cooperative spin: on an oversubscribed host (ranks >= cores,
+      // e.g. threaded tests) a hard spin here starves the user threads
+      // that produce the very work we are polling for. yield() costs
+      // ~100ns when nothing else is runnable.
+      std::this_thread::yield();
       continue; // nothing ready, no tick work owed to anyone
     }
     for (int i = 0; i < n; i++) {

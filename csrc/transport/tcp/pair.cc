@@ -282,6 +282,28 @@ void TcpPair::ubufSend(
   }
 }
 
+void TcpPair::checkInvariantLocked(const char* where, uint64_t slot) {
+  static const bool on = getenv("GLOO_AMD_CHECK_INVARIANTS") != nullptr;
+  if (!on) {
+    return;
+  }
+  auto qi = localPendingRecv_.find(slot);
+  auto si = eagerStash_.find(slot);
+  bool qne = qi != localPendingRecv_.end() && !qi->second.empty();
+  bool sne = si != eagerStash_.end() && !si->second.empty();
+  if (qne && sne) {
+    GA_ERROR << "INVARIANT VIOLATION at " << where << " " << str()
+             << " slot=" << slot
+             << " pendRecv=" << qi->second.size()
+             << " stash=" << si->second.size()
+             << " rxActive=" << rxActive_
+             << " rxIsEagerSpill=" << rxIsEagerSpill_
+             << " rxSlot=" << rxPre_.slot
+             << " rxOpcode=" << rxPre_.opcode;
+    abort();
+  }
+}
+
 size_t TcpPair::eagerMaxBytes() {
   static size_t v = [] {
     if (const char* e = getenv("GLOO_AMD_EAGER_MAX")) {
@@ -339,6 +361,7 @@ void TcpPair::postRecv(
     ctx_->consumeUnclaimedLocked(slot, peerRank_);
   }
   localPendingRecv_[slot].push_back({buf, off, nb});
+  checkInvariantLocked("postRecv", slot);
   TxOp op{};
   op.pre = {RECV_READY, static_cast<uint32_t>(ctx_->rank), slot, nb, 0};
   enqueueTxLocked(std::move(op));
@@ -629,6 +652,7 @@ void TcpPair::dispatchPreamble() {
         }
         rxActive_ = false;
       }
+      checkInvariantLocked("dispatchData", pre.slot);
       return;
     }
     case BOUND_DATA: {
@@ -696,6 +720,7 @@ void TcpPair::finishRx() {
     } else {
       eagerStash_[rxPre_.slot].push_back(std::move(rxSpill_));
     }
+    checkInvariantLocked("finishRxEager", rxPre_.slot);
     rxSpill_ = std::string();
     rxIsEagerSpill_ = false;
   } else if (rxBbuf_ != nullptr) {

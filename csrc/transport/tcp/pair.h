@@ -155,6 +155,7 @@ class TcpPair : public transport::Pair, public Handler {
   };
 
   static size_t eagerMaxBytes(); // GLOO_AMD_EAGER_MAX, default 8 KiB
+  void checkInvariantLocked(const char* where, uint64_t slot);
 
   void enqueueTxLocked(TxOp op);
   void flushTxLocked();

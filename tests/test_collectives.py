@@ -488,3 +488,42 @@ def test_allreduce_16rank(spawn_threads):
         return True
 
     spawn_threads(16, fn)
+
+
+def test_sustained_concurrent_streams(spawn_threads):
+    """Sustained load: 4 ranks x 3 concurrent collective streams x many
+    iterations (regression net for protocol liveness under contention;
+    all ranks run identical iteration counts — collectives are
+    collective)."""
+    size = 4
+    plan = [(1, 64, 120), (2, 3000, 80), (3, 100_000, 25)]
+
+    def fn(ctx, rank, _):
+        import threading as th
+
+        errs = []
+
+        def stream(tag, n, iters):
+            try:
+                for it in range(iters):
+                    x = (np.arange(n, dtype=np.float64) + rank + it).astype(
+                        np.float32)
+                    ga.allreduce(ctx, [x.ctypes.data], n, ga.DType.f32,
+                                 ga.ReduceOp.sum, tag=tag)
+                    if it % 17 == 0:
+                        ref = sum(
+                            (np.arange(n, dtype=np.float64) + r + it).astype(
+                                np.float32) for r in range(size))
+                        assert np.allclose(x, ref), (tag, it)
+            except Exception:  # noqa: BLE001
+                import traceback
+
+                errs.append(traceback.format_exc())
+
+        ths = [th.Thread(target=stream, args=p) for p in plan]
+        [t.start() for t in ths]
+        [t.join() for t in ths]
+        assert not errs, errs[0]
+        return True
+
+    spawn_threads(size, fn)
