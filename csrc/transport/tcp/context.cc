@@ -174,6 +174,15 @@ void TcpContext::signalException(const std::string& msg) {
   if (getEnvFlag("GLOO_AMD_DUMP_ON_TIMEOUT")) {
     dumpStateLocked(msg.c_str());
   }
+  for (auto& p : pairs_) {
+    if (p && !p->eagerStash_.empty()) {
+      GA_WARN << "rank " << rank << ": timeout with undelivered payloads "
+              << "from " << p->str() << " — a peer likely advanced to a "
+              << "different collective (ranks must invoke collectives "
+              << "equally; GLOO_AMD_DUMP_ON_TIMEOUT=1 shows queue state)";
+      break;
+    }
+  }
   auto e = std::make_exception_ptr(IoException(msg));
   for (auto& p : pairs_) {
     if (p) {
