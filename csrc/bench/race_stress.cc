@@ -1,5 +1,8 @@
-// Race-detection stress harness: two ranks in one process, fixed
-// contexts, concurrent v2 collectives (large-ring + tiny-eager paths).
+// Race-detection stress harness: GLOO_AMD_STRESS_RANKS thread-ranks
+// (default 2) in one process, fixed contexts, GLOO_AMD_STRESS_STREAMS
+// (default 2) concurrent collective streams per rank mixing large-ring
+// and tiny-eager paths. Every rank runs identical iteration counts —
+// collectives are collective.
 //
 // Build with `make SANITIZE=thread race_stress` and run under TSan.
 // NOTE: waits must be untimed for a faithful TSan model: libstdc++-11's
@@ -28,34 +31,54 @@ int main() {
     attr.useLibuv = true; // exercise the libuv loop under the sanitizer
   }
   auto dev = tcp::createTcpDevice(attr);
-  const int P = 2;
+  auto envInt = [](const char* k, int d) {
+    const char* v = getenv(k);
+    return v ? atoi(v) : d;
+  };
+  const int P = envInt("GLOO_AMD_STRESS_RANKS", 2);
+  const int streams = envInt("GLOO_AMD_STRESS_STREAMS", 2);
+  const int iters = envInt("GLOO_AMD_STRESS_ITERS", 300);
   std::vector<std::thread> ths;
   for (int r = 0; r < P; r++) {
     ths.emplace_back([&, r] {
       auto ctx = std::make_shared<Context>(r, P);
       ctx->setTimeout(std::chrono::milliseconds(-1));
       ctx->connectFullMesh(*store, dev);
-      std::vector<float> x(10000);
-      for (int it = 0; it < 300; it++) {
-        for (size_t i = 0; i < x.size(); i++) x[i] = float(i % 7 + r);
-        AllreduceOptions o(ctx);
-        o.setOutput(x.data(), x.size());
-        o.reduce = cpuReduceFn(DType::F32, ReduceOp::SUM);
-        o.tag = 1;
-        allreduce(o);
+      std::vector<std::thread> sts;
+      for (int t = 0; t < streams; t++) {
+        sts.emplace_back([&, r, t] {
+          // odd streams tiny (eager + recursive doubling), even streams
+          // large (segmented ring)
+          const size_t n = (t & 1) ? 64 : 10000;
+          std::vector<float> x(n);
+          for (int it = 0; it < iters; it++) {
+            for (size_t i = 0; i < n; i++) {
+              x[i] = float(i % 7 + r + it);
+            }
+            AllreduceOptions o(ctx);
+            o.setOutput(x.data(), n);
+            o.reduce = cpuReduceFn(DType::F32, ReduceOp::SUM);
+            o.tag = static_cast<uint32_t>(t + 1);
+            allreduce(o);
+            float expect = 0;
+            for (int rr = 0; rr < P; rr++) {
+              expect += float(0 % 7 + rr + it);
+            }
+            if (x[0] != expect) {
+              std::printf("MISMATCH rank %d stream %d it %d\n", r, t, it);
+              abort();
+            }
+          }
+        });
       }
-      // tiny path (eager + recursive doubling) on a second slot
-      for (int it = 0; it < 300; it++) {
-        std::vector<float> y(64, float(r + 1));
-        AllreduceOptions o(ctx);
-        o.setOutput(y.data(), y.size());
-        o.reduce = cpuReduceFn(DType::F32, ReduceOp::SUM);
-        o.tag = 2;
-        allreduce(o);
+      for (auto& t2 : sts) {
+        t2.join();
       }
       std::printf("rank %d done\n", r);
     });
   }
-  for (auto& t : ths) t.join();
+  for (auto& t : ths) {
+    t.join();
+  }
   return 0;
 }
