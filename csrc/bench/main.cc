@@ -571,6 +571,10 @@ int main(int argc, char** argv) {
     tattr.pkeyFile = o.tlsKey;
     tattr.caFile = o.tlsCa;
     device = tcp::tls::createTlsDevice(tattr);
+  } else if (o.transport == "uds") {
+    tcp::TcpAttr attr;
+    attr.useUds = true;
+    device = tcp::createTcpDevice(attr);
   } else {
     GA_ENFORCE(o.transport == "tcp", "unknown --transport ", o.transport);
     tcp::TcpAttr attr;

@@ -257,15 +257,17 @@ PYBIND11_MODULE(_C, m) {
 
   m.def(
       "create_tcp_device",
-      [](const std::string& hostname, bool useLibuv) {
+      [](const std::string& hostname, bool useLibuv, bool useUds) {
         tcp::TcpAttr attr;
         attr.hostname = hostname;
         attr.useLibuv = useLibuv;
+        attr.useUds = useUds;
         return std::static_pointer_cast<transport::Device>(
             tcp::createTcpDevice(attr));
       },
       py::arg("hostname") = std::string(),
-      py::arg("use_libuv") = false);
+      py::arg("use_libuv") = false,
+      py::arg("use_uds") = false);
 
   m.def(
       "create_tls_device",

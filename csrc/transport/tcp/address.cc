@@ -1,6 +1,7 @@
 #include "transport/tcp/address.h"
 
 #include <arpa/inet.h>
+#include <sys/un.h>
 
 #include "common/logging.h"
 
@@ -31,6 +32,8 @@ socklen_t TcpAddress::sockaddrLen() const {
       return sizeof(struct sockaddr_in);
     case AF_INET6:
       return sizeof(struct sockaddr_in6);
+    case AF_UNIX:
+      return sizeof(struct sockaddr_un); // abstract name, zero-padded
     default:
       return sizeof(ss_);
   }
@@ -39,6 +42,9 @@ socklen_t TcpAddress::sockaddrLen() const {
 std::string TcpAddress::str() const {
   char host[INET6_ADDRSTRLEN] = {0};
   int port = 0;
+  if (ss_.ss_family == AF_UNIX) {
+    return std::string("uds#") + std::to_string(seq_);
+  }
   if (ss_.ss_family == AF_INET) {
     auto* in = reinterpret_cast<const struct sockaddr_in*>(&ss_);
     inet_ntop(AF_INET, &in->sin_addr, host, sizeof(host));

@@ -2,11 +2,13 @@
 
 #include <fcntl.h>
 #include <netdb.h>
+#include <sys/un.h>
 #include <netinet/tcp.h>
 #include <sys/epoll.h>
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include <atomic>
 #include <cstring>
 
 #include "common/logging.h"
@@ -42,6 +44,33 @@ TcpDevice::TcpDevice(const TcpAttr& attr) {
     loop_ = makeUvLoop();
   } else {
     loop_ = std::make_unique<EpollLoop>();
+  }
+  if (attr.useUds) {
+    // Abstract-namespace unix socket: unique zero-padded name, no
+    // filesystem residue. bind/connect both use sizeof(sockaddr_un) so
+    // the (padded) name length is consistent end to end.
+    listenFd_ = socket(AF_UNIX, SOCK_STREAM | SOCK_CLOEXEC, 0);
+    GA_ENFORCE_GE(listenFd_, 0, "socket(AF_UNIX): ", strerror(errno));
+    struct sockaddr_un sun;
+    std::memset(&sun, 0, sizeof(sun));
+    sun.sun_family = AF_UNIX;
+    static std::atomic<uint64_t> ctr{0};
+    snprintf(
+        sun.sun_path + 1,
+        sizeof(sun.sun_path) - 1,
+        "gloo_amd_%d_%llu",
+        static_cast<int>(getpid()),
+        static_cast<unsigned long long>(ctr.fetch_add(1)));
+    int rv = bind(
+        listenFd_, reinterpret_cast<struct sockaddr*>(&sun), sizeof(sun));
+    GA_ENFORCE_EQ(rv, 0, "bind(uds): ", strerror(errno));
+    rv = listen(listenFd_, 1024);
+    GA_ENFORCE_EQ(rv, 0, "listen(uds): ", strerror(errno));
+    std::memset(&advertised_, 0, sizeof(advertised_));
+    std::memcpy(&advertised_, &sun, sizeof(sun));
+    setNonBlocking(listenFd_);
+    loop_->registerDescriptor(listenFd_, EPOLLIN, this);
+    return;
   }
   std::string host = attr.hostname.empty() ? "127.0.0.1" : attr.hostname;
 

@@ -29,6 +29,10 @@ struct TcpAttr {
   // Drive the transport with a libuv loop instead of raw epoll
   // (reference uv-transport parity; same wire protocol).
   bool useLibuv = false;
+  // Unix-domain stream sockets (abstract namespace) instead of TCP:
+  // same wire protocol, lower latency and higher throughput for the
+  // single-node case. Peers must share a kernel.
+  bool useUds = false;
 };
 
 class TcpDevice;

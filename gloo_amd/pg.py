@@ -99,7 +99,14 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
             self._store = _TorchStoreAdapter(store)
         self._ctx = ga.Context(rank, size)
         self._ctx.set_timeout(int(timeout.total_seconds() * 1000))
-        self._ctx.connect_full_mesh(self._store, ga.create_tcp_device())
+        import os
+
+        # Single-node deployments can route the CPU control plane over
+        # unix-domain sockets (lower latency/higher throughput than TCP
+        # loopback); multi-node needs TCP.
+        uds = os.environ.get("GLOO_AMD_UDS", "0") == "1"
+        self._ctx.connect_full_mesh(
+            self._store, ga.create_tcp_device(use_uds=uds))
         self._lock = threading.Lock()
         # One helper thread: p2p waits complete in post order off the
         # caller's thread (isend/irecv return pending Works).

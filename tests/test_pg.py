@@ -270,3 +270,16 @@ def test_pg_monitored_barrier_reports_missing():
         return True
 
     spawn_pg(3, fn)
+
+
+def test_pg_over_uds(monkeypatch):
+    """GLOO_AMD_UDS=1 routes the PG control plane over unix sockets."""
+    monkeypatch.setenv("GLOO_AMD_UDS", "1")
+
+    def fn(pg, rank, size):
+        t = torch.full((2048,), float(rank + 1))
+        pg.allreduce([t]).wait()
+        assert torch.allclose(t, torch.full((2048,), 3.0))
+        return True
+
+    spawn_pg(2, fn)

@@ -34,6 +34,8 @@ def _spawn(size, fn, transport, tls_certs):
         try:
             if transport == "tcp":
                 dev = ga.create_tcp_device()
+            elif transport == "uds":
+                dev = ga.create_tcp_device(use_uds=True)
             elif transport == "uv":
                 dev = ga.create_tcp_device(use_libuv=True)
             else:
@@ -59,7 +61,7 @@ def _spawn(size, fn, transport, tls_certs):
     assert not errors, errors[0]
 
 
-TRANSPORTS = ["tcp", "uv"] + (["tls"] if openssl else [])
+TRANSPORTS = ["tcp", "uds", "uv"] + (["tls"] if openssl else [])
 
 
 @pytest.mark.parametrize("transport", TRANSPORTS)
