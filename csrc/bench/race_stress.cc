@@ -30,6 +30,9 @@ int main() {
   if (getenv("GLOO_AMD_STRESS_UV")) {
     attr.useLibuv = true; // exercise the libuv loop under the sanitizer
   }
+  if (getenv("GLOO_AMD_STRESS_UDS")) {
+    attr.useUds = true; // exercise the unix-socket path
+  }
   auto dev = tcp::createTcpDevice(attr);
   auto envInt = [](const char* k, int d) {
     const char* v = getenv(k);
