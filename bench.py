@@ -47,15 +47,6 @@ def main():
     master_addr = os.environ.get("MASTER_ADDR", "127.0.0.1")
     master_port = int(os.environ.get("MASTER_PORT", "29500")) + 1
 
-    if world > 1:
-        store = ga.TcpStore(master_addr, master_port, is_server=(rank == 0))
-    else:
-        store = ga.HashStore()
-    dev = ga.create_tcp_device(master_addr if world > 1 else "")
-    ctx = ga.Context(rank, world)
-    ctx.connect_full_mesh(store, dev)
-    ctx.set_timeout(120000)
-
     have_gpu = False
     torch = None
     try:
@@ -64,6 +55,19 @@ def main():
         have_gpu = torch.cuda.is_available()
     except ImportError:
         pass
+
+    if world > 1:
+        store = ga.TcpStore(master_addr, master_port, is_server=(rank == 0))
+    else:
+        store = ga.HashStore()
+    # CPU-only fallback is single-node by construction: use the
+    # unix-socket transport (GPU runs keep TCP for the control plane).
+    use_uds = (not have_gpu) and master_addr in ("127.0.0.1", "localhost")
+    dev = ga.create_tcp_device(
+        master_addr if world > 1 else "", use_uds=use_uds)
+    ctx = ga.Context(rank, world)
+    ctx.connect_full_mesh(store, dev)
+    ctx.set_timeout(120000)
 
     tmap = {"f32": "float32", "bf16": "bfloat16", "f16": "float16"}
     gmap = {"f32": ga.DType.f32, "bf16": ga.DType.bf16, "f16": ga.DType.f16}
@@ -198,7 +202,8 @@ def main():
                 "collective": args.algorithm,
                 "elements": args.elements,
                 "payload_mb": round(nbytes / 1e6, 1),
-                "transport": "xgmi-ipc" if have_gpu else "tcp-localhost",
+                "transport": ("xgmi-ipc" if have_gpu else
+                              ("uds-local" if use_uds else "tcp-localhost")),
                 "p50_us": round(p50_us, 1),
                 "p99_us": round(p99_us, 1),
                 "bus_GBps": round(bus_gbps, 2),
