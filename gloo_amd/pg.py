@@ -75,7 +75,10 @@ class _TorchStoreAdapter(ga.Store):
         return bytes(self._s.get(key))
 
     def wait(self, keys, timeout_ms):
-        self._s.wait(list(keys))
+        if timeout_ms and timeout_ms > 0:
+            self._s.wait(list(keys), timedelta(milliseconds=timeout_ms))
+        else:
+            self._s.wait(list(keys))
 
 
 def _gdtype(t):
