@@ -343,6 +343,20 @@ PYBIND11_MODULE(_C, m) {
             return b.waitSend(ms(timeoutMs));
           },
           py::arg("timeout_ms") = -1)
+      .def(
+          "try_wait_recv",
+          [](transport::UnboundBuffer& b, long timeoutMs) {
+            int src = -1;
+            bool ok;
+            {
+              py::gil_scoped_release rel;
+              ok = b.tryWaitRecv(&src, ms(timeoutMs));
+            }
+            return py::make_tuple(ok, src);
+          },
+          py::arg("timeout_ms") = -1,
+          "Probing wait: returns (False, -1) on timeout without "
+          "poisoning the context (monitored_barrier liveness probes).")
       .def("abort_wait_recv", &transport::UnboundBuffer::abortWaitRecv)
       .def("abort_wait_send", &transport::UnboundBuffer::abortWaitSend);
 

@@ -176,8 +176,12 @@ void smallAllreduce(AllreduceOptions& opts) {
       tmpBuf->recv(peer, slot + 1 + t, 0, bytes);
       outBuf->send(peer, slot + 1 + t, 0, bytes);
       tmpBuf->waitRecv(timeout);
-      opts.reduce(out, out, tmp.get(), opts.elements);
+      // The transport serializes send payloads lazily (the TxOp holds a
+      // pointer into `out`); the in-place reduce below would corrupt a
+      // partially-flushed send. Wait for the send first — deadlock-free
+      // because both peers posted their recvs before sending.
       outBuf->waitSend(timeout);
+      opts.reduce(out, out, tmp.get(), opts.elements);
     }
   }
   if (r < extras) {

@@ -519,6 +519,10 @@ int64_t TcpStore::add(const std::string& key, int64_t delta) {
 }
 
 std::vector<char> TcpStore::get(const std::string& key) {
+  // The 'G' opcode blocks server-side on a condvar with no deadline;
+  // bound it with the store timeout by first polling for existence
+  // (wait honors deadlines client-side), then fetching.
+  wait({key}, timeout_);
   std::lock_guard<std::mutex> lock(clientMu_);
   uint8_t op = 'G';
   writeAll(clientFd_, &op, 1);

@@ -519,6 +519,35 @@ bool TcpUnboundBuffer::waitSend(std::chrono::milliseconds timeout) {
   std::rethrow_exception(error_);
 }
 
+bool TcpUnboundBuffer::tryWaitRecv(
+    int* srcRank,
+    std::chrono::milliseconds timeout) {
+  std::unique_lock<std::mutex> lock(ctx_->mu_);
+  if (timeout.count() < 0) {
+    timeout = ctx_->getTimeout();
+  }
+  auto pred = [&] {
+    return !recvCompletions_.empty() || error_ != nullptr || abortRecv_;
+  };
+  if (timeout.count() < 0) {
+    ctx_->cv_.wait(lock, pred);
+  } else if (!ctx_->cv_.wait_for(lock, timeout, pred)) {
+    return false; // timed out; context left intact (probe semantics)
+  }
+  if (!recvCompletions_.empty()) {
+    if (srcRank != nullptr) {
+      *srcRank = recvCompletions_.front();
+    }
+    recvCompletions_.pop_front();
+    return true;
+  }
+  if (abortRecv_) {
+    abortRecv_ = false;
+    return false;
+  }
+  std::rethrow_exception(error_);
+}
+
 void TcpUnboundBuffer::abortWaitRecv() {
   std::lock_guard<std::mutex> lock(ctx_->mu_);
   abortRecv_ = true;

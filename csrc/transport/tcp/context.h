@@ -33,6 +33,7 @@ class TcpUnboundBuffer : public transport::UnboundBuffer {
       size_t offset,
       size_t nbytes) override;
   bool waitRecv(int* srcRank, std::chrono::milliseconds timeout) override;
+  bool tryWaitRecv(int* srcRank, std::chrono::milliseconds timeout) override;
   bool waitSend(std::chrono::milliseconds timeout) override;
   void abortWaitRecv() override;
   void abortWaitSend() override;

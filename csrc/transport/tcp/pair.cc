@@ -100,8 +100,12 @@ void TcpPair::connect(const std::vector<char>& peerAddressBytes) {
 
   if (initiator) {
     // Initiator: dial the peer's listener and write the peer pair's seq.
+    // Retry until the context timeout deadline (mirroring the listener
+    // side's waitForConnection) — a fixed attempt cap spuriously fails
+    // slow or large-scale rendezvous.
+    const auto deadline = std::chrono::steady_clock::now() +
+        (timeout.count() > 0 ? timeout : std::chrono::milliseconds(30000));
     int fd = -1;
-    int attempts = 0;
     for (;;) {
       fd = socket(peer_.sockaddr().ss_family, SOCK_STREAM | SOCK_CLOEXEC, 0);
       GA_ENFORCE_GE(fd, 0, "socket: ", strerror(errno));
@@ -114,12 +118,12 @@ void TcpPair::connect(const std::vector<char>& peerAddressBytes) {
       }
       ::close(fd);
       fd = -1;
-      if (++attempts >= 30) {
+      if (std::chrono::steady_clock::now() >= deadline) {
         GA_THROW_IO(
             "connect to ", peer_.str(), " failed: ", strerror(errno));
       }
       usleep(100 * 1000); // peer's listener may not be up in cross-process
-                          // rendezvous races; retry briefly
+                          // rendezvous races; retry until deadline
     }
     uint64_t seq = peer_.seq();
     size_t written = 0;

@@ -146,6 +146,12 @@ class UnboundBuffer {
 
   virtual void abortWaitRecv() = 0;
   virtual void abortWaitSend() = 0;
+
+  // Probing wait: like waitRecv(timeout) but a timeout returns false
+  // WITHOUT poisoning the context (no signalException, no throw). Used
+  // by liveness probes (monitored_barrier) that must keep the context
+  // usable after a peer fails to arrive. A pair error still throws.
+  virtual bool tryWaitRecv(int* srcRank, std::chrono::milliseconds timeout) = 0;
 };
 
 class Pair {

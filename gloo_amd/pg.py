@@ -17,6 +17,7 @@ backend name; call the ProcessGroup method instead:
 `dist.distributed_c10d._get_default_group().monitored_barrier()`.
 """
 import threading
+import time
 from concurrent.futures import ThreadPoolExecutor
 from datetime import timedelta
 
@@ -430,22 +431,34 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
             self._mb_seq += 1
             byte = torch.zeros(1, dtype=torch.uint8)
             if rank == 0:
-                missing = []
+                # Post every probe first, then wait with try_wait_recv:
+                # a timeout returns False without poisoning the context,
+                # so later ranks are still probed truthfully and the
+                # context stays usable after the report.
+                deadline = time.monotonic() + tmo_ms / 1000.0
+                missing, probes = [], []
                 for src in range(1, size):
                     try:
                         ub = self._ctx.create_unbound_buffer(
                             byte.data_ptr(), 1)
                         ub.recv(src, slot, 0, 0)
-                        ub.wait_recv(tmo_ms)
+                        probes.append((src, ub))
                     except ga.GlooAmdError:
-                        # timed out, or the peer already died (closed pair)
+                        missing.append(src)  # pair already dead
+                for src, ub in probes:
+                    rem = max(0, int((deadline - time.monotonic()) * 1000))
+                    try:
+                        ok, _ = ub.try_wait_recv(rem)
+                    except ga.GlooAmdError:
+                        ok = False  # peer died while probing
+                    if not ok:
                         missing.append(src)
-                        if not wait_all_ranks:
-                            break
+                del probes  # detaches any unmatched pending recvs
                 if missing:
                     raise RuntimeError(
                         "monitored_barrier: rank(s) "
-                        f"{missing} failed to arrive within {tmo_ms}ms")
+                        f"{sorted(missing)} failed to arrive within "
+                        f"{tmo_ms}ms")
                 for dst in range(1, size):
                     ub = self._ctx.create_unbound_buffer(byte.data_ptr(), 1)
                     ub.send(dst, slot + 1, 0, 0)

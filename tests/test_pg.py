@@ -272,6 +272,30 @@ def test_pg_monitored_barrier_reports_missing():
     spawn_pg(3, fn)
 
 
+def test_pg_monitored_barrier_keeps_context_usable():
+    """A monitored_barrier timeout must not poison the context: the probe
+    uses try_wait_recv (non-poisoning), so after the missing-rank report
+    the same process group still completes collectives."""
+    from datetime import timedelta
+
+    def fn(pg, rank, size):
+        if rank == 0:
+            try:
+                pg.monitored_barrier(timeout=timedelta(milliseconds=500),
+                                     wait_all_ranks=True)
+                raise AssertionError("expected monitored_barrier to fail")
+            except RuntimeError as e:
+                assert "1" in str(e), e
+        # rank 1 never joined the monitored_barrier; both ranks now run a
+        # normal collective, which only works if rank 0's context survived
+        t = torch.full((512,), float(rank + 1))
+        pg.allreduce([t]).wait()
+        assert torch.allclose(t, torch.full((512,), 3.0))
+        return True
+
+    spawn_pg(2, fn)
+
+
 def test_pg_over_uds(monkeypatch):
     """GLOO_AMD_UDS=1 routes the PG control plane over unix sockets."""
     monkeypatch.setenv("GLOO_AMD_UDS", "1")
