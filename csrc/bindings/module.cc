@@ -419,6 +419,8 @@ PYBIND11_MODULE(_C, m) {
       .def("close", &Context::closeConnections,
            py::call_guard<py::gil_scoped_release>())
       .def("next_slot", &Context::nextSlot, py::arg("num_to_skip") = 1)
+      .def("slot_counter", &Context::slotCounter)
+      .def("reset_slot_counter", &Context::resetSlotCounter, py::arg("value"))
       .def(
           "get_pair",
           [](Context& ctx, int rank) { return ctx.getPair(rank); },
@@ -837,6 +839,23 @@ PYBIND11_MODULE(_C, m) {
           py::arg("elements"),
           py::arg("dtype") = DType::F32,
           py::arg("op") = ReduceOp::SUM,
+          py::arg("stream") = 0)
+      .def(
+          "run_multi",
+          [](hip::HipAllreduceRing& a, const std::vector<uintptr_t>& ptrs,
+             size_t n, DType dt, ReduceOp op, uintptr_t stream) {
+            std::vector<void*> vp;
+            vp.reserve(ptrs.size());
+            for (auto p : ptrs) {
+              vp.push_back(reinterpret_cast<void*>(p));
+            }
+            py::gil_scoped_release rel;
+            a.run(vp, n, dt, op, reinterpret_cast<hipStream_t>(stream));
+          },
+          py::arg("ptrs"),
+          py::arg("elements"),
+          py::arg("dtype") = DType::F32,
+          py::arg("op") = ReduceOp::SUM,
           py::arg("stream") = 0);
 
   py::class_<hip::HipAllreduceHalvingDoubling>(m, "HipAllreduceHalvingDoubling")
@@ -908,6 +927,23 @@ PYBIND11_MODULE(_C, m) {
                   reinterpret_cast<hipStream_t>(stream));
           },
           py::arg("ptr"),
+          py::arg("elements"),
+          py::arg("dtype") = DType::F32,
+          py::arg("op") = ReduceOp::SUM,
+          py::arg("stream") = 0)
+      .def(
+          "run_multi",
+          [](hip::HipAllreduceDirect& a, const std::vector<uintptr_t>& ptrs,
+             size_t n, DType dt, ReduceOp op, uintptr_t stream) {
+            std::vector<void*> vp;
+            vp.reserve(ptrs.size());
+            for (auto p : ptrs) {
+              vp.push_back(reinterpret_cast<void*>(p));
+            }
+            py::gil_scoped_release rel;
+            a.run(vp, n, dt, op, reinterpret_cast<hipStream_t>(stream));
+          },
+          py::arg("ptrs"),
           py::arg("elements"),
           py::arg("dtype") = DType::F32,
           py::arg("op") = ReduceOp::SUM,

@@ -49,6 +49,15 @@ class Context {
   uint32_t nextSlot(uint32_t numToSkip = 1) {
     return slotCounter_.fetch_add(numToSkip);
   }
+  // Resync support: after a rank-divergent failure (e.g. a collective
+  // constructor threw mid-setup), ranks agree on max(slotCounter) out of
+  // band and jump every counter there so tag allocation realigns.
+  uint32_t slotCounter() const {
+    return slotCounter_.load();
+  }
+  void resetSlotCounter(uint32_t v) {
+    slotCounter_.store(v);
+  }
 
   void setTimeout(std::chrono::milliseconds timeout);
   std::chrono::milliseconds getTimeout() const {

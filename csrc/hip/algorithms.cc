@@ -134,11 +134,22 @@ void HipAllreduceRing::run(
     DType dtype,
     ReduceOp op,
     hipStream_t callerStream) {
+  std::vector<void*> ptrs{devPtr};
+  run(ptrs, elements, dtype, op, callerStream);
+}
+
+void HipAllreduceRing::run(
+    const std::vector<void*>& ptrs,
+    size_t elements,
+    DType dtype,
+    ReduceOp op,
+    hipStream_t callerStream) {
   TraceRange tr("gloo_amd::hip_allreduce_ring");
   GA_HIP_CHECK(hipSetDevice(device_));
+  GA_ENFORCE(!ptrs.empty());
   const size_t es = dtypeSize(dtype);
   const size_t bytes = elements * es;
-  if (ctx_->size == 1 || elements == 0) {
+  if (elements == 0 || (ctx_->size == 1 && ptrs.size() == 1)) {
     return;
   }
   {
@@ -155,10 +166,41 @@ void HipAllreduceRing::run(
     }
     GA_HIP_CHECK(hipEventDestroy(ev));
   }
-  if (bytes < onDeviceThreshold()) {
-    runHostStaged(static_cast<char*>(devPtr), bytes, elements, dtype, op);
-  } else {
-    runDevice(static_cast<char*>(devPtr), bytes, elements, dtype, op);
+  char* buf = static_cast<char*>(ptrs[0]);
+  if (ptrs.size() > 1) {
+    // Fused local reduction of the caller's pointers into ptrs[0] as the
+    // ring's copy-in stage (reference cuda_allreduce_ring.cc:72-120
+    // reduces N local pointers before the wire phase; here one fused
+    // k-way kernel replaces its k-1 sweeps).
+    const int k = static_cast<int>(std::min<size_t>(ptrs.size(), 8));
+    launchReduceN(
+        buf, const_cast<const void* const*>(ptrs.data()), k, elements,
+        dtype, op, cs_[0]->stream());
+    for (size_t i = 8; i < ptrs.size(); i++) {
+      launchReduce2(buf, buf, ptrs[i], elements, dtype, op,
+                    cs_[0]->stream());
+    }
+    HipEvent red(device_);
+    red.record(cs_[0]->stream());
+    for (size_t j = 0; j < cs_.size(); j++) {
+      red.streamWait(cs_[j]->stream());
+      red.streamWait(ks_[j]->stream());
+    }
+  }
+  if (ctx_->size > 1) {
+    if (bytes < onDeviceThreshold()) {
+      runHostStaged(buf, bytes, elements, dtype, op);
+    } else {
+      runDevice(buf, bytes, elements, dtype, op);
+    }
+  }
+  if (ptrs.size() > 1) {
+    // Broadcast the reduced result back to every caller pointer.
+    for (size_t i = 1; i < ptrs.size(); i++) {
+      GA_HIP_CHECK(hipMemcpyAsync(
+          ptrs[i], buf, bytes, hipMemcpyDeviceToDevice, cs_[0]->stream()));
+    }
+    cs_[0]->synchronize();
   }
 }
 
@@ -315,18 +357,15 @@ void HipAllreduceRing::runDevice(
     DType dt,
     ReduceOp op) {
   const size_t es = bytes / n;
-  mesh_->ensureCapacity(bytes, inboxCap_);
-  char* work = mesh_->work();
+  // In-place on the user buffer: the ring schedule only ever writes peer
+  // INBOXES (never a peer's work region), so no staging copy is needed —
+  // r01 staged through mesh work both ways, 2*bytes of extra HBM traffic
+  // per allreduce.
+  mesh_->ensureCapacity(0, inboxCap_);
   const int R = static_cast<int>(strides_.size());
-
-  // Stage in once; every ring is ordered after it.
-  GA_HIP_CHECK(hipMemcpyAsync(
-      work, buf, bytes, hipMemcpyDeviceToDevice, cs_[0]->stream()));
-  HipEvent staged(device_);
-  staged.record(cs_[0]->stream());
   for (int j = 0; j < R; j++) {
-    staged.streamWait(cs_[j]->stream());
-    staged.streamWait(ks_[j]->stream());
+    // Carries the caller-stream dependency (gated in run()) into the
+    // k<S steps of each ring's schedule.
     initEvent_[j]->record(cs_[j]->stream());
   }
 
@@ -335,17 +374,14 @@ void HipAllreduceRing::runDevice(
   for (int j = 0; j < R; j++) {
     Seg part = subspanOf({0, n}, j, R);
     if (part.len > 0) {
-      enqueueRing(j, work, part.off, part.len, es, dt, op);
+      enqueueRing(j, buf, part.off, part.len, es, dt, op);
     }
   }
 
-  // Copy-out after every ring's compute stream finished its part.
   for (int j = 1; j < R; j++) {
     doneEvent_[j]->record(cs_[j]->stream());
     doneEvent_[j]->streamWait(cs_[0]->stream());
   }
-  GA_HIP_CHECK(hipMemcpyAsync(
-      buf, work, bytes, hipMemcpyDeviceToDevice, cs_[0]->stream()));
   doneEvent_[0]->record(cs_[0]->stream());
 
   auto timeout = ctx_->getTimeout();
@@ -372,13 +408,15 @@ HipAllreduceHalvingDoubling::HipAllreduceHalvingDoubling(
       device_(device),
       inboxCap_(inboxCap == 0 ? kDefaultInboxCap : inboxCap) {
   const int P = ctx_->size;
-  GA_ENFORCE(
-      P > 0 && (P & (P - 1)) == 0,
-      "hip_allreduce_halving_doubling requires power-of-2 size, got ",
-      P,
-      " (use hip_allreduce_ring)");
+  GA_ENFORCE_GT(P, 0);
+  // Non-power-of-2 sizes fold the extra ranks into partners before and
+  // after a power-of-2 exchange (the CPU legacy HD's scheme,
+  // csrc/algorithms/allreduce_halving_doubling.h; reference instead uses
+  // binary blocks, gloo/allreduce_halving_doubling.h:38-64).
+  pow2_ = 1;
   log2P_ = 0;
-  while ((1 << log2P_) < P) {
+  while (pow2_ * 2 <= P) {
+    pow2_ *= 2;
     log2P_++;
   }
   GA_HIP_CHECK(hipSetDevice(device_));
@@ -392,6 +430,12 @@ HipAllreduceHalvingDoubling::HipAllreduceHalvingDoubling(
   fDATA_ = mesh_->allocFlags(T * 2);
   fACK_ = mesh_->allocFlags(T * 2);
   fAGD_ = mesh_->allocFlags(T);
+  if (P != pow2_) {
+    fFOLD_ = mesh_->allocFlags(2); // pre-fold chunk data, x2 parity
+    fFACK_ = mesh_->allocFlags(2); // pre-fold chunk consumed
+    fPOST_ = mesh_->allocFlags(1); // post-fold result landed
+    fPACK_ = mesh_->allocFlags(1); // extra's copy-out done
+  }
 }
 
 void HipAllreduceHalvingDoubling::run(
@@ -411,11 +455,15 @@ void HipAllreduceHalvingDoubling::run(
   }
   gateStreams(callerStream, {cs_->stream(), ks_->stream()});
   char* buf = static_cast<char*>(devPtr);
-  const int T = log2P_;
+  const int T = std::max(1, log2P_);
+  const int extras = P - pow2_;
+  const bool isExtra = r >= pow2_;
   mesh_->ensureCapacity(bytes, inboxCap_);
-  // Sub-inbox layout: the 2*inboxCap inbox region split into T steps x 2
-  // parities.
-  const size_t subBytes = (2 * mesh_->inboxCap()) / (2 * T);
+  // Sub-inbox layout: the 2*inboxCap inbox region split into rows x 2
+  // parities; rows = T steps plus (when non-pow2) one fold row.
+  const int rows = T + (extras > 0 ? 1 : 0);
+  const size_t subBytes = (2 * mesh_->inboxCap()) / (2 * rows);
+  GA_ENFORCE_GE(subBytes, es, "inbox too small for the step count");
   auto subInbox = [&](int t, int par) {
     return mesh_->inbox(0) + (static_cast<size_t>(t) * 2 + par) * subBytes;
   };
@@ -428,8 +476,72 @@ void HipAllreduceHalvingDoubling::run(
   auto csm = cs_->stream();
   auto ksm = ks_->stream();
 
+  // Fold-chunk geometry (element-aligned), identical on both sides.
+  const size_t foldChunkElems = subBytes / es;
+  const int foldNc = static_cast<int>(
+      (elements + foldChunkElems - 1) / foldChunkElems);
+  auto foldChunk = [&](int c) {
+    const size_t off = static_cast<size_t>(c) * foldChunkElems;
+    return Seg{off, std::min(foldChunkElems, elements - off)};
+  };
+
+  if (isExtra) {
+    // ---- extra rank: fold into the partner, then receive the result.
+    const int partner = r - pow2_;
+    for (int c = 0; c < foldNc; c++) {
+      const int par = c & 1;
+      foldSeq_++;
+      if (lastFold_[par] > 0) {
+        launchWaitFlagGte(mesh_->flag(fFACK_ + par), lastFold_[par], ksm);
+      }
+      Seg ch = foldChunk(c);
+      GA_HIP_CHECK(hipMemcpyAsync(
+          peerSubInbox(partner, T, par),
+          buf + ch.off * es,
+          ch.len * es,
+          hipMemcpyDeviceToDevice,
+          ksm));
+      launchWriteFlag(mesh_->peerFlag(partner, fFOLD_ + par), foldSeq_, ksm);
+      lastFold_[par] = foldSeq_;
+    }
+    // Post-fold: partner writes the final result into our work region.
+    postSeq_++;
+    launchWaitFlagGte(mesh_->flag(fPOST_), postSeq_, csm);
+    GA_HIP_CHECK(
+        hipMemcpyAsync(buf, work, bytes, hipMemcpyDeviceToDevice, csm));
+    launchWriteFlag(mesh_->peerFlag(partner, fPACK_), postSeq_, csm);
+    doneEvent_->record(csm);
+    auto timeout = ctx_->getTimeout();
+    watchdogWait(*doneEvent_, *mesh_, timeout, "hip_allreduce_hd (fold cs)");
+    initEvent_->record(ksm);
+    watchdogWait(*initEvent_, *mesh_, timeout, "hip_allreduce_hd (fold ks)");
+    cs_->synchronize();
+    ks_->synchronize();
+    return;
+  }
+
   GA_HIP_CHECK(
       hipMemcpyAsync(work, buf, bytes, hipMemcpyDeviceToDevice, csm));
+
+  // ---- pre-fold: reduce the extra rank's chunks into work (cs) ----
+  if (r < extras) {
+    for (int c = 0; c < foldNc; c++) {
+      const int par = c & 1;
+      foldSeq_++;
+      launchWaitFlagGte(mesh_->flag(fFOLD_ + par), foldSeq_, csm);
+      Seg ch = foldChunk(c);
+      launchReduce2(
+          work + ch.off * es,
+          work + ch.off * es,
+          subInbox(T, par),
+          ch.len,
+          dtype,
+          op,
+          csm);
+      launchWriteFlag(
+          mesh_->peerFlag(r + pow2_, fFACK_ + par), foldSeq_, csm);
+    }
+  }
   initEvent_->record(csm);
 
   // Host-tracked per-(step,parity) last-issued ack/data seqs; identical
@@ -531,6 +643,19 @@ void HipAllreduceHalvingDoubling::run(
   // Final: wait all incoming allgather halves, stage out.
   for (int t = 0; t < T; t++) {
     launchWaitFlagGte(mesh_->flag(fAGD_ + t), lastAgd_[t], csm);
+  }
+  if (r < extras) {
+    // Post-fold: push the final result into the extra rank's work
+    // region (single writer). Run n+1's write is gated on the extra's
+    // copy-out ack from run n so we never clobber an in-flight read.
+    postSeq_++;
+    const int extra = r + pow2_;
+    if (postSeq_ > 1) {
+      launchWaitFlagGte(mesh_->flag(fPACK_), postSeq_ - 1, csm);
+    }
+    GA_HIP_CHECK(hipMemcpyAsync(
+        mesh_->peerWork(extra), work, bytes, hipMemcpyDeviceToDevice, csm));
+    launchWriteFlag(mesh_->peerFlag(extra, fPOST_), postSeq_, csm);
   }
   GA_HIP_CHECK(
       hipMemcpyAsync(buf, work, bytes, hipMemcpyDeviceToDevice, csm));
@@ -661,15 +786,51 @@ void HipAllreduceDirect::run(
     DType dtype,
     ReduceOp op,
     hipStream_t callerStream) {
+  std::vector<void*> ptrs{devPtr};
+  run(ptrs, elements, dtype, op, callerStream);
+}
+
+void HipAllreduceDirect::run(
+    const std::vector<void*>& ptrs,
+    size_t elements,
+    DType dtype,
+    ReduceOp op,
+    hipStream_t callerStream) {
   TraceRange tr("gloo_amd::hip_allreduce_direct");
   GA_HIP_CHECK(hipSetDevice(device_));
+  GA_ENFORCE(!ptrs.empty());
   const int P = ctx_->size;
   const int r = ctx_->rank;
   const size_t es = dtypeSize(dtype);
-  if (P == 1 || elements == 0) {
+  if (elements == 0) {
     return;
   }
-  char* user = static_cast<char*>(devPtr);
+  if (P == 1 || ptrs.size() > 1) {
+    // Fused local reduction of the caller's pointers into ptrs[0] before
+    // the wire phase (reference cuda_allreduce_ring.cc:72-120 role).
+    if (ptrs.size() > 1) {
+      gateStreams(callerStream, {cs_->stream()});
+      const int k = static_cast<int>(std::min<size_t>(ptrs.size(), 8));
+      launchReduceN(
+          ptrs[0], const_cast<const void* const*>(ptrs.data()), k, elements,
+          dtype, op, cs_->stream());
+      for (size_t i = 8; i < ptrs.size(); i++) {
+        launchReduce2(ptrs[0], ptrs[0], ptrs[i], elements, dtype, op,
+                      cs_->stream());
+      }
+      callerStream = cs_->stream(); // wire phase ordered after the fuse
+    }
+    if (P == 1) {
+      for (size_t i = 1; i < ptrs.size(); i++) {
+        GA_HIP_CHECK(hipMemcpyAsync(
+            ptrs[i], ptrs[0], elements * es, hipMemcpyDeviceToDevice,
+            cs_->stream()));
+      }
+      cs_->synchronize();
+      return;
+    }
+  }
+  char* user = static_cast<char*>(ptrs[0]);
   const size_t perRank = (elements + P - 1) / P;
   // Chunk the per-rank block so scatter / reduce / broadcast pipeline
   // for large payloads. Inbox layout: P sources x 2 parities x chunkCap.
@@ -830,6 +991,14 @@ void HipAllreduceDirect::run(
     }
   }
   chunkSeqBase_ += C;
+  if (ptrs.size() > 1) {
+    for (size_t i = 1; i < ptrs.size(); i++) {
+      GA_HIP_CHECK(hipMemcpyAsync(
+          ptrs[i], user, elements * es, hipMemcpyDeviceToDevice,
+          cs_->stream()));
+    }
+    cs_->synchronize();
+  }
 }
 
 // ===========================================================================

@@ -63,6 +63,15 @@ class HipAllreduceRing {
       DType dtype,
       ReduceOp op,
       hipStream_t callerStream = nullptr);
+  // Multi-input form (reference cuda benchmark --inputs sweep parity):
+  // fused k-way local reduction into ptrs[0] as the copy-in stage, then
+  // the wire phase in place on ptrs[0], result broadcast to every ptr.
+  void run(
+      const std::vector<void*>& ptrs,
+      size_t elements,
+      DType dtype,
+      ReduceOp op,
+      hipStream_t callerStream = nullptr);
 
   XgmiMesh* mesh() {
     return mesh_.get();
@@ -123,6 +132,7 @@ class HipAllreduceHalvingDoubling {
   int device_;
   size_t inboxCap_;
   int log2P_;
+  int pow2_; // largest power of 2 <= size; ranks >= pow2_ fold
   std::unique_ptr<XgmiMesh> mesh_;
   HipStream* cs_;
   HipStream* ks_;
@@ -131,7 +141,12 @@ class HipAllreduceHalvingDoubling {
   std::unique_ptr<HipEvent> doneEvent_;
   std::vector<std::unique_ptr<HipEvent>> stepEvents_;
   int fDATA_, fACK_, fAGD_; // per step t: [t*2 + parity]
+  // Non-pow2 folding flags (allocated only when size != pow2_).
+  int fFOLD_{-1}, fFACK_{-1}, fPOST_{-1}, fPACK_{-1};
   uint64_t seq_{0};
+  uint64_t foldSeq_{0}; // private partner<->extra chunk counter
+  uint64_t postSeq_{0}; // per-run post-fold handshake counter
+  uint64_t lastFold_[2] = {0, 0};
   // Last-issued doorbell seq per (step, parity) / per step; identical on
   // all ranks (symmetric schedule) and persistent across runs so inbox
   // reuse is safe across run boundaries.
@@ -182,6 +197,14 @@ class HipAllreduceDirect {
 
   void run(
       void* devPtr,
+      size_t elements,
+      DType dtype,
+      ReduceOp op,
+      hipStream_t callerStream = nullptr);
+  // Multi-input form: fused local reduceN into ptrs[0], then the one-shot
+  // wire phase, result broadcast to every ptr.
+  void run(
+      const std::vector<void*>& ptrs,
       size_t elements,
       DType dtype,
       ReduceOp op,

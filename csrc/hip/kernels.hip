@@ -584,21 +584,29 @@ namespace {
 // when the device supports them. GLOO_AMD_STREAM_OPS=0 forces kernels.
 bool useStreamOps() {
   static int cached = [] {
+    int v = 1;
+    const char* why = "hipDeviceAttributeCanUseStreamWaitValue=1";
     const char* env = std::getenv("GLOO_AMD_STREAM_OPS");
     if (env != nullptr && env[0] == '0') {
-      return 0;
+      v = 0;
+      why = "GLOO_AMD_STREAM_OPS=0";
+    } else {
+      int dev = 0;
+      int canWait = 0;
+      if (hipGetDevice(&dev) != hipSuccess ||
+          hipDeviceGetAttribute(
+              &canWait, hipDeviceAttributeCanUseStreamWaitValue, dev) !=
+              hipSuccess ||
+          canWait == 0) {
+        v = 0;
+        why = "device cannot use stream wait-value packets";
+      }
     }
-    int dev = 0;
-    if (hipGetDevice(&dev) != hipSuccess) {
-      return 0;
-    }
-    int canWait = 0;
-    if (hipDeviceGetAttribute(
-            &canWait, hipDeviceAttributeCanUseStreamWaitValue, dev) !=
-        hipSuccess) {
-      return 0;
-    }
-    return canWait ? 1 : 0;
+    GA_INFO << "doorbell path: "
+            << (v ? "hipStreamWriteValue64/WaitValue64 (command processor)"
+                  : "system-scope atomic kernels")
+            << " (" << why << ")";
+    return v;
   }();
   return cached == 1;
 }
