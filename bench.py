@@ -32,6 +32,13 @@ import statistics
 import sys
 import time
 
+# Must land before the HIP runtime initializes (first torch.cuda call):
+# the engines run up to 8 streams per process (4 rings x cs/ks, or
+# 1+7 direct fanout); ROCm's default of 4 HW queues aliases them, and a
+# stream-op wait packet halts its whole HW queue (measured 37x on the
+# direct engine at 4 ranks, profiles/r02 tuning notes).
+os.environ.setdefault("GPU_MAX_HW_QUEUES", "8")
+
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 REF_P50_US_5M = 10608.0  # allreduce_ring_chunked 5M fp32 p50
@@ -105,6 +112,13 @@ def main():
 
     if have_gpu:
         # one GPU per rank; wrap on boxes with fewer GPUs than ranks
+        if world > torch.cuda.device_count():
+            # Timeshared preflight (N ranks on one GPU): every cross-rank
+            # dependency hop pays a process-scheduling quantum, so deep
+            # multi-ring schedules convoy (measured 2.7s -> 105ms at 8
+            # timeshared ranks with 1 ring). Real multi-GPU runs keep the
+            # multi-ring default.
+            os.environ.setdefault("GLOO_AMD_NUM_RINGS", "1")
         local_rank = local_rank % torch.cuda.device_count()
         torch.cuda.set_device(local_rank)
         tdt = getattr(torch, tmap[args.dtype])
