@@ -78,6 +78,12 @@ void EpollLoop::unregisterDescriptor(int fd) {
   }
 }
 
+void EpollLoop::barrier() {
+  if (!inLoopThread()) {
+    waitForTick();
+  }
+}
+
 void EpollLoop::waitForTick() {
   std::unique_lock<std::mutex> lock(mu_);
   uint64_t current = tick_;

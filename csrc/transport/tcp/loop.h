@@ -42,6 +42,12 @@ class Loop {
   // Run fn on the loop thread soon.
   virtual void defer(std::function<void()> fn) = 0;
 
+  // Returns once the loop has moved past its current dispatch batch (and,
+  // for deferred-removal loops, past previously deferred teardown), so a
+  // handler unregistered via unregisterNoWait can be safely deleted.
+  // No-op from the loop thread itself.
+  virtual void barrier() = 0;
+
   virtual bool inLoopThread() const = 0;
 };
 
@@ -55,6 +61,7 @@ class EpollLoop : public Loop {
   void unregisterDescriptor(int fd) override;
   void unregisterNoWait(int fd) override;
   void defer(std::function<void()> fn) override;
+  void barrier() override;
 
   bool inLoopThread() const override {
     return std::this_thread::get_id() == threadId_;

@@ -78,6 +78,13 @@ TcpPair::TcpPair(TcpContext* ctx, TcpDevice* dev, int peerRank)
 
 TcpPair::~TcpPair() {
   close();
+  // If the pair was torn down earlier via failLocked (unregisterNoWait +
+  // deferred ::close), close() above returned immediately — but a poll
+  // callback dispatched before the deferred removal may still hold this
+  // handler. Wait one loop barrier before the memory goes away.
+  if (everRegistered_) {
+    dev_->loop().barrier();
+  }
 }
 
 std::string TcpPair::str() const {
@@ -150,17 +157,19 @@ void TcpPair::connect(const std::vector<char>& peerAddressBytes) {
     std::lock_guard<std::mutex> lock(ctx_->mu_);
     state_ = CONNECTED;
   }
+  everRegistered_ = true;
   dev_->loop().registerDescriptor(fd_, EPOLLIN, this);
 }
 
 void TcpPair::close() {
   int fd;
   {
-    std::lock_guard<std::mutex> lock(ctx_->mu_);
+    std::unique_lock<std::mutex> lock(ctx_->mu_);
     if (state_ == CLOSED) {
       return;
     }
-    state_ = CLOSED;
+    state_ = CLOSED; // a mid-drain flusher aborts at its next relock
+    ctx_->cv_.wait(lock, [&] { return !txBusy_; });
     fd = fd_;
     fd_ = -1;
   }
@@ -414,28 +423,47 @@ void TcpPair::completeTxLocked(TxOp& op) {
 }
 
 void TcpPair::flushTxLocked() {
-  if (state_ != CONNECTED || fd_ < 0) {
+  if (state_ != CONNECTED || fd_ < 0 || txBusy_) {
+    // txBusy_: another thread is mid-drain (it re-checks tx_ after its
+    // unlocked writev returns, so ops enqueued meanwhile are not lost).
     return;
   }
-  // Cap the bytes pushed per lock acquisition: with a 16 MiB SO_SNDBUF a
-  // single writev of a large payload memcpys it all into the kernel
-  // while holding the context mutex, stalling the loop thread's RX for
-  // every pair of this context. Past the cap, EPOLLOUT continues the
-  // drain on the loop thread in bounded slices.
+  // Cap the bytes pushed per call so a loop-thread EPOLLOUT drain cannot
+  // monopolize the loop for an unbounded large payload; user threads get
+  // their RX handled concurrently regardless because the writev below
+  // runs with the context mutex RELEASED (reference: per-pair mutex in
+  // gloo/transport/tcp/pair.cc:658-688 serves the same goal).
   static const size_t kMaxFlushBytes = [] {
     if (const char* e = getenv("GLOO_AMD_MAX_FLUSH")) {
       long v = atol(e);
       return v <= 0 ? std::numeric_limits<size_t>::max()
                     : static_cast<size_t>(v);
     }
-    return static_cast<size_t>(1 << 20);
+    return static_cast<size_t>(4 << 20);
   }();
+  txBusy_ = true;
   size_t flushed = 0;
-  while (!tx_.empty()) {
-    if (flushed >= kMaxFlushBytes) {
-      armEpollOutLocked();
-      return;
+  bool arm = false;
+  while (true) {
+    if (error_) {
+      // failLocked ran (possibly while we were unlocked): it fans the
+      // error out but defers queue/fd teardown to us (teardownDeferred_).
+      tx_.clear();
+      if (teardownDeferred_) {
+        teardownDeferred_ = false;
+        teardownFdLocked();
+      }
+      break;
     }
+    if (state_ != CONNECTED || fd_ < 0 || tx_.empty()) {
+      break;
+    }
+    if (flushed >= kMaxFlushBytes) {
+      arm = true;
+      break;
+    }
+    // tx_.front() is stable while txBusy_: only this thread pops, and
+    // deque push_back never invalidates references to existing elements.
     TxOp& op = tx_.front();
     struct iovec iov[2];
     int iovcnt = 0;
@@ -448,8 +476,6 @@ void TcpPair::flushTxLocked() {
     if (op.payloadWritten < op.payloadLen) {
       iov[iovcnt].iov_base =
           const_cast<char*>(op.payload) + op.payloadWritten;
-      // A single writev into a roomy SO_SNDBUF copies everything at
-      // once; clamp it to the per-flush quota too.
       iov[iovcnt].iov_len = std::min(
           op.payloadLen - op.payloadWritten, kMaxFlushBytes - flushed);
       iovcnt++;
@@ -459,18 +485,27 @@ void TcpPair::flushTxLocked() {
       tx_.pop_front();
       continue;
     }
+    // The syscall (a memcpy of up to the quota into the kernel sndbuf)
+    // runs unlocked so it never serializes other pairs of this context.
+    // fd lifetime: close()/failLocked defer the ::close while txBusy_.
+    ctx_->mu_.unlock();
     ssize_t n = ioWritev(iov, iovcnt);
+    int savedErrno = errno;
+    ctx_->mu_.lock();
+    if (error_ || state_ != CONNECTED || fd_ < 0) {
+      continue; // top of loop handles teardown/exit
+    }
     if (n < 0) {
-      if (errno == EAGAIN || errno == EWOULDBLOCK) {
-        armEpollOutLocked();
-        return;
+      if (savedErrno == EAGAIN || savedErrno == EWOULDBLOCK) {
+        arm = true;
+        break;
       }
-      if (errno == EINTR) {
+      if (savedErrno == EINTR) {
         continue;
       }
-      failLocked(std::make_exception_ptr(
-          IoException("writev to " + str() + ": " + strerror(errno))));
-      return;
+      failLocked(std::make_exception_ptr(IoException(
+          "writev to " + str() + ": " + strerror(savedErrno))));
+      continue; // top of loop performs the deferred teardown
     }
     // Account written bytes across preamble then payload.
     flushed += static_cast<size_t>(n);
@@ -487,6 +522,13 @@ void TcpPair::flushTxLocked() {
     }
     // else: partial; loop retries writev (kernel buffer may have space)
   }
+  txBusy_ = false;
+  if (arm) {
+    armEpollOutLocked();
+  }
+  // Wake close()/detachBuffer/~TcpUnboundBuffer waiting on !txBusy_ or
+  // on tx_ references draining.
+  ctx_->cv_.notify_all();
 }
 
 void TcpPair::armEpollOutLocked() {
@@ -810,11 +852,25 @@ void TcpPair::failLocked(std::exception_ptr e) {
   // eagerStash_ is deliberately kept: it holds fully-received payloads,
   // which stay deliverable after the peer closes.
 
-  // Tear the socket down without waiting for a loop tick: failLocked runs
-  // with the context mutex held, and the loop thread may be blocked on
-  // that same mutex (tick-wait here would deadlock). epoll_ctl(DEL) stops
-  // new dispatches; the close itself is deferred to the loop thread so it
-  // runs strictly after any handler that may still be inside read().
+  // Tear the socket down. If a flusher thread is mid-writev on this fd
+  // with the mutex released, defer the teardown to it (closing now could
+  // let the kernel recycle the fd under the in-flight syscall).
+  if (fd_ >= 0) {
+    if (txBusy_) {
+      teardownDeferred_ = true;
+    } else {
+      teardownFdLocked();
+    }
+  }
+  ctx_->cv_.notify_all();
+}
+
+void TcpPair::teardownFdLocked() {
+  // Without waiting for a loop tick: this can run with the context mutex
+  // held while the loop thread is blocked on that same mutex (tick-wait
+  // would deadlock). epoll_ctl(DEL) stops new dispatches; the close
+  // itself is deferred to the loop thread so it runs strictly after any
+  // handler that may still be inside read().
   if (fd_ >= 0) {
     int fd = fd_;
     fd_ = -1;
@@ -822,7 +878,6 @@ void TcpPair::failLocked(std::exception_ptr e) {
     dev_->loop().unregisterNoWait(fd);
     dev_->loop().defer([fd] { ::close(fd); });
   }
-  ctx_->cv_.notify_all();
 }
 
 } // namespace tcp

@@ -159,6 +159,7 @@ class TcpPair : public transport::Pair, public Handler {
 
   void enqueueTxLocked(TxOp op);
   void flushTxLocked();
+  void teardownFdLocked();
   void armEpollOutLocked();
   void readLoop(); // loop thread; takes/releases ctx lock internally
   bool readPreamble(); // no lock
@@ -207,6 +208,12 @@ class TcpPair : public transport::Pair, public Handler {
 
   std::deque<TxOp> tx_;
   bool epollOutArmed_{false};
+  // True while one thread drains tx_ with the context mutex dropped
+  // around the writev syscall; guards tx_.front() stability and defers
+  // fd teardown (see flushTxLocked / failLocked / close).
+  bool txBusy_{false};
+  bool teardownDeferred_{false};
+  bool everRegistered_{false}; // gated loop barrier in the destructor
 
   // --- rx state machine (loop thread only, except targets set under lock) ---
   Preamble rxPre_;

@@ -112,6 +112,15 @@ class UvLoop : public Loop {
     defer([this, fd] { removePoll(fd); });
   }
 
+  void barrier() override {
+    if (!inLoopThread()) {
+      // Deferred fns run FIFO on the loop thread, serialized with poll
+      // callbacks: when this no-op has run, every earlier removePoll has
+      // too and no handler dispatch is in flight.
+      runOnLoopAndWait([] {});
+    }
+  }
+
   void defer(std::function<void()> fn) override {
     {
       std::lock_guard<std::mutex> lock(mu_);
