@@ -1028,6 +1028,50 @@ PYBIND11_MODULE(_C, m) {
           py::arg("element_size") = 4,
           py::arg("stream") = 0);
 
+  py::class_<hip::HipP2P>(m, "HipP2P")
+      .def(
+          py::init([](std::shared_ptr<Context> ctx, int device,
+                      size_t chunkCap) {
+            py::gil_scoped_release rel;
+            return std::make_unique<hip::HipP2P>(ctx, device, chunkCap);
+          }),
+          py::arg("context"),
+          py::arg("device"),
+          py::arg("chunk_cap") = 0)
+      .def(
+          "post_send",
+          [](hip::HipP2P& p, int dst, uintptr_t ptr, size_t bytes,
+             uintptr_t stream) {
+            py::gil_scoped_release rel;
+            p.postSend(dst, reinterpret_cast<const void*>(ptr), bytes,
+                       reinterpret_cast<hipStream_t>(stream));
+          },
+          py::arg("dst"),
+          py::arg("ptr"),
+          py::arg("bytes"),
+          py::arg("stream") = 0)
+      .def(
+          "post_recv",
+          [](hip::HipP2P& p, int src, uintptr_t ptr, size_t bytes,
+             uintptr_t stream) {
+            py::gil_scoped_release rel;
+            p.postRecv(src, reinterpret_cast<void*>(ptr), bytes,
+                       reinterpret_cast<hipStream_t>(stream));
+          },
+          py::arg("src"),
+          py::arg("ptr"),
+          py::arg("bytes"),
+          py::arg("stream") = 0)
+      .def("flush_sends",
+           [](hip::HipP2P& p) {
+             py::gil_scoped_release rel;
+             p.flushSends();
+           })
+      .def("flush_recvs", [](hip::HipP2P& p) {
+        py::gil_scoped_release rel;
+        p.flushRecvs();
+      });
+
   py::class_<hip::HipBroadcastOneToAll>(m, "HipBroadcastOneToAll")
       .def(
           py::init([](std::shared_ptr<Context> ctx, int device, int root,

@@ -1391,6 +1391,108 @@ void HipAlltoall::run(
 }
 
 // ===========================================================================
+// HipP2P
+// ===========================================================================
+
+HipP2P::HipP2P(std::shared_ptr<Context> ctx, int device, size_t chunkCap)
+    : ctx_(std::move(ctx)),
+      device_(device),
+      chunkCap_(chunkCap == 0 ? (4 << 20) : chunkCap) {
+  GA_HIP_CHECK(hipSetDevice(device_));
+  // Arena: P lanes x 2 parity slots x chunkCap, laid out over the
+  // mesh's two contiguous inboxes (work | inbox0 | inbox1 in mesh.h).
+  mesh_ = std::make_unique<XgmiMesh>(
+      ctx_, device_, 0, static_cast<size_t>(ctx_->size) * chunkCap_);
+  ss_ = std::make_unique<HipStream>(device_);
+  rs_ = std::make_unique<HipStream>(device_);
+  fDATA_ = mesh_->allocFlags(ctx_->size);
+  fACK_ = mesh_->allocFlags(ctx_->size);
+  sendChunkSeq_.assign(ctx_->size, 0);
+  recvChunkSeq_.assign(ctx_->size, 0);
+}
+
+void HipP2P::postSend(
+    int dst,
+    const void* devPtr,
+    size_t bytes,
+    hipStream_t callerStream) {
+  TraceRange tr("gloo_amd::hip_p2p_send");
+  std::lock_guard<std::mutex> lock(mu_);
+  GA_HIP_CHECK(hipSetDevice(device_));
+  const int r = ctx_->rank;
+  GA_ENFORCE(dst != r, "p2p send to self");
+  gateStreams(callerStream, {ss_->stream()});
+  const char* src = static_cast<const char*>(devPtr);
+  const size_t nc = (bytes + chunkCap_ - 1) / chunkCap_;
+  for (size_t c = 0; c < std::max<size_t>(nc, 1); c++) {
+    const uint64_t seq = ++sendChunkSeq_[dst];
+    const int par = static_cast<int>(seq & 1);
+    if (seq > 2) {
+      // Same-parity slot reuse: dst consumed chunk seq-2.
+      launchWaitFlagGte(mesh_->flag(fACK_ + dst), seq - 2, ss_->stream());
+    }
+    const size_t off = c * chunkCap_;
+    const size_t len = std::min(chunkCap_, bytes - std::min(bytes, off));
+    if (len > 0) {
+      GA_HIP_CHECK(hipMemcpyAsync(
+          slotOf(mesh_->peerInbox(dst, 0), r, par),
+          src + off,
+          len,
+          hipMemcpyDeviceToDevice,
+          ss_->stream()));
+    }
+    launchWriteFlag(mesh_->peerFlag(dst, fDATA_ + r), seq, ss_->stream());
+  }
+}
+
+void HipP2P::postRecv(
+    int src,
+    void* devPtr,
+    size_t bytes,
+    hipStream_t callerStream) {
+  TraceRange tr("gloo_amd::hip_p2p_recv");
+  std::lock_guard<std::mutex> lock(mu_);
+  GA_HIP_CHECK(hipSetDevice(device_));
+  GA_ENFORCE(src != ctx_->rank, "p2p recv from self");
+  gateStreams(callerStream, {rs_->stream()});
+  char* dstPtr = static_cast<char*>(devPtr);
+  const size_t nc = (bytes + chunkCap_ - 1) / chunkCap_;
+  for (size_t c = 0; c < std::max<size_t>(nc, 1); c++) {
+    const uint64_t seq = ++recvChunkSeq_[src];
+    const int par = static_cast<int>(seq & 1);
+    launchWaitFlagGte(mesh_->flag(fDATA_ + src), seq, rs_->stream());
+    const size_t off = c * chunkCap_;
+    const size_t len = std::min(chunkCap_, bytes - std::min(bytes, off));
+    if (len > 0) {
+      GA_HIP_CHECK(hipMemcpyAsync(
+          dstPtr + off,
+          slotOf(mesh_->inbox(0), src, par),
+          len,
+          hipMemcpyDeviceToDevice,
+          rs_->stream()));
+    }
+    launchWriteFlag(
+        mesh_->peerFlag(src, fACK_ + ctx_->rank), seq, rs_->stream());
+  }
+}
+
+void HipP2P::flushSends() {
+  GA_HIP_CHECK(hipSetDevice(device_));
+  HipEvent done(device_);
+  done.record(ss_->stream());
+  watchdogWait(done, *mesh_, ctx_->getTimeout(), "hip_p2p (sends)");
+  ss_->synchronize();
+}
+
+void HipP2P::flushRecvs() {
+  GA_HIP_CHECK(hipSetDevice(device_));
+  HipEvent done(device_);
+  done.record(rs_->stream());
+  watchdogWait(done, *mesh_, ctx_->getTimeout(), "hip_p2p (recvs)");
+  rs_->synchronize();
+}
+
+// ===========================================================================
 // hipAllreduceLocal
 // ===========================================================================
 

@@ -342,6 +342,59 @@ class HipAlltoall {
   uint64_t seq_{0};
 };
 
+// Device-native tagged-order point-to-point engine (pipeline-parallel
+// p2p + gather/scatter staging). Each (src->dst) lane is a
+// double-buffered chunk slot in the receiver's inbox arena; posts
+// enqueue the full chunked schedule (copies + doorbell waits) on
+// dedicated streams and return immediately, so a host thread can post a
+// send AND a recv before waiting either (batch_isend_irecv safety).
+// Ordering per lane is post order on both sides (c10d's matching rule).
+// Replaces the r01 PG host-staged path for CUDA send/recv/gather/
+// scatter (reference role: gloo tagged unbound send/recv over pairs).
+class HipP2P {
+ public:
+  HipP2P(
+      std::shared_ptr<Context> ctx,
+      int device,
+      size_t chunkCap = 0 /*0 -> 4 MiB per lane slot*/);
+
+  // Enqueue a chunked device-to-device send/recv (async; returns once
+  // enqueued). The caller's tensor must stay alive until the matching
+  // flush returns.
+  void postSend(
+      int dst,
+      const void* devPtr,
+      size_t bytes,
+      hipStream_t callerStream = nullptr);
+  void postRecv(
+      int src,
+      void* devPtr,
+      size_t bytes,
+      hipStream_t callerStream = nullptr);
+  // Block until every posted send / recv completed (watchdog-guarded).
+  void flushSends();
+  void flushRecvs();
+
+ private:
+  char* slotOf(char* inboxBase, int src, int par) {
+    return inboxBase + (static_cast<size_t>(src) * 2 + par) * chunkCap_;
+  }
+
+  std::shared_ptr<Context> ctx_;
+  int device_;
+  size_t chunkCap_;
+  std::unique_ptr<XgmiMesh> mesh_;
+  // Dedicated streams (not pooled): p2p waits must never interleave
+  // with a collective engine's schedule on a shared stream.
+  std::unique_ptr<HipStream> ss_;
+  std::unique_ptr<HipStream> rs_;
+  int fDATA_; // [src]: chunks sent to me by src (monotonic)
+  int fACK_; // [dst]: chunks dst consumed of my sends (monotonic)
+  std::vector<uint64_t> sendChunkSeq_; // per dst lane
+  std::vector<uint64_t> recvChunkSeq_; // per src lane
+  std::mutex mu_; // posts may come from several host threads
+};
+
 // Single-process multi-pointer allreduce: fused k-way reduction into
 // ptrs[0] then broadcast copies (all on one device/stream; blocking).
 void hipAllreduceLocal(

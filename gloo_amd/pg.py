@@ -122,6 +122,14 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
         self._hip_ag = {}  # device -> HipAllgatherRing
         self._hip_rs = {}  # device -> HipReduceScatterRing
         self._hip_a2a = {}  # device -> HipAlltoall
+        # Device-native p2p engine (send/recv/gather/scatter without a
+        # pinned round trip). Constructed EAGERLY: its mesh exchange is
+        # collective, and p2p calls are not a safe collective point.
+        self._hip_p2p = None
+        self._hip_p2p_dev = -1
+        if size > 1 and torch.cuda.is_available():
+            self._hip_p2p_dev = torch.cuda.current_device()
+            self._hip_p2p = ga._C.HipP2P(self._ctx, self._hip_p2p_dev)
 
     # -- helpers -------------------------------------------------------------
 
@@ -267,6 +275,18 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
         op = _OPS[opts.reduceOp] if opts is not None else ga.ReduceOp.sum
         for t in tensors:
             t_ = t.detach()
+            if t_.is_cuda and t_.is_contiguous():
+                # Device-native: run the device allreduce; the root holds
+                # the reduced result, non-root contents are unspecified
+                # by the c10d contract (here: also the full result).
+                with self._lock:
+                    self._ring(t_.get_device()).run(
+                        t_.data_ptr(), t_.numel(), _gdtype(t_), op,
+                        stream=_cur_stream(t_))
+                if opts is not None \
+                        and opts.reduceOp == dist.ReduceOp.AVG:
+                    t_.div_(self.size())
+                continue
 
             def run(host):
                 ga.reduce(self._ctx, host.data_ptr(), host.data_ptr(),
@@ -365,6 +385,29 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
     def gather(self, output_tensors, input_tensors, opts=None):
         root = opts.rootRank if opts is not None else 0
         inp = input_tensors[0].detach().contiguous()
+        if self._p2p_device_ok([input_tensors[0]]) and (
+                self.rank() != root
+                or all(o.is_cuda and o.is_contiguous()
+                       for o in output_tensors[0])):
+            # Device-native: non-roots stream their block to the root's
+            # inbox lanes; the root collects each lane straight into the
+            # output tensor (no pinned staging).
+            nbytes = inp.numel() * inp.element_size()
+            if self.rank() == root:
+                outs = output_tensors[0]
+                for src in range(self.size()):
+                    if src == root:
+                        outs[src].detach().copy_(inp.view_as(outs[src]))
+                    else:
+                        self._hip_p2p.post_recv(
+                            src, outs[src].data_ptr(), nbytes,
+                            _cur_stream(outs[src]))
+                self._hip_p2p.flush_recvs()
+            else:
+                self._hip_p2p.post_send(
+                    root, inp.data_ptr(), nbytes, _cur_stream(inp))
+                self._hip_p2p.flush_sends()
+            return _ret_work(output_tensors)
         h_in = inp.cpu() if inp.is_cuda else inp
         if self.rank() == root:
             outs = output_tensors[0]
@@ -385,6 +428,26 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
     def scatter(self, output_tensors, input_tensors, opts=None):
         root = opts.rootRank if opts is not None else 0
         out = output_tensors[0].detach()
+        if self._p2p_device_ok([output_tensors[0]]) \
+                and out.is_contiguous() and (
+                self.rank() != root
+                or all(t.is_cuda for t in input_tensors[0])):
+            nbytes = out.numel() * out.element_size()
+            if self.rank() == root:
+                ins = [t.detach().contiguous() for t in input_tensors[0]]
+                for dst in range(self.size()):
+                    if dst == root:
+                        out.copy_(ins[dst].view_as(out))
+                    else:
+                        self._hip_p2p.post_send(
+                            dst, ins[dst].data_ptr(), nbytes,
+                            _cur_stream(ins[dst]))
+                self._hip_p2p.flush_sends()
+            else:
+                self._hip_p2p.post_recv(
+                    root, out.data_ptr(), nbytes, _cur_stream(out))
+                self._hip_p2p.flush_recvs()
+            return _ret_work(output_tensors)
         h_out = torch.empty(out.numel(), dtype=out.dtype)
         if self.rank() == root:
             flat = torch.cat(
@@ -491,7 +554,28 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
         self._p2p_pool.submit(runner)
         return _create_work_from_future(fut)
 
+    def _p2p_device_ok(self, tensors):
+        return (self._hip_p2p is not None
+                and all(t.is_cuda and t.get_device() == self._hip_p2p_dev
+                        for t in tensors))
+
     def send(self, tensors, dst_rank, tag=0):
+        if self._p2p_device_ok(tensors):
+            # Device-native: chunks move straight over xGMI inbox lanes;
+            # the post enqueues the whole schedule, the helper thread
+            # only syncs (no pinned round trip).
+            keep = []
+            for t in tensors:
+                t_ = t.detach().contiguous()
+                keep.append(t_)  # storage alive until the flush
+                self._hip_p2p.post_send(
+                    dst_rank, t_.data_ptr(),
+                    t_.numel() * t_.element_size(), _cur_stream(t_))
+
+            def finish(keep=keep):
+                self._hip_p2p.flush_sends()
+
+            return self._p2p_async(finish)
         ubs = []
         for t in tensors:
             t_ = t.detach().contiguous()
@@ -508,6 +592,19 @@ class ProcessGroupGlooAmd(dist.ProcessGroup):
         return self._p2p_async(finish)
 
     def recv(self, tensors, src_rank, tag=0):
+        if self._p2p_device_ok(tensors) and all(
+                t.is_contiguous() for t in tensors):
+            for t in tensors:
+                t_ = t.detach()
+                self._hip_p2p.post_recv(
+                    src_rank, t_.data_ptr(),
+                    t_.numel() * t_.element_size(), _cur_stream(t_))
+            tens = list(tensors)
+
+            def finish(tens=tens):
+                self._hip_p2p.flush_recvs()
+
+            return self._p2p_async(finish)
         posted = []
         for t in tensors:
             t_ = t.detach()

@@ -442,6 +442,27 @@ def test_hip_allreduce_direct_multi_input():
     _two_rank_generic(fn)
 
 
+def test_hip_p2p_bidirectional_chunked():
+    """HipP2P: both ranks post a send AND a recv before flushing either
+    (batch_isend_irecv pattern); 20MB payloads exercise the multi-chunk
+    double-buffered lane protocol."""
+    def fn(ctx, rank):
+        n = 5_000_000  # 20 MB > 2 x 4MB lane slots -> needs acks
+        peer = 1 - rank
+        src = _dev(torch.full((n,), float(rank + 1)))
+        dst = torch.zeros(n, device="cuda")
+        eng = ga._C.HipP2P(ctx, 0)
+        for it in range(2):  # cross-run lane seq reuse
+            eng.post_send(peer, src.data_ptr(), n * 4)
+            eng.post_recv(peer, dst.data_ptr(), n * 4)
+            eng.flush_sends()
+            eng.flush_recvs()
+            assert torch.all(_host(dst) == peer + 1), (rank, it)
+            ga.barrier(ctx, tag=971 + it)
+
+    _two_rank_generic(fn)
+
+
 def test_pg_cuda_collectives():
     """ProcessGroup with CUDA tensors: device-native paths."""
     from gloo_amd.pg import ProcessGroupGlooAmd
