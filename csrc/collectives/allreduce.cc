@@ -193,6 +193,22 @@ void smallAllreduce(AllreduceOptions& opts) {
   }
 }
 
+// Factorize P into group sizes per step: repeated `base` factors, then
+// the remainder as one final (possibly larger) factor — reference
+// computeGroupSizePerStep semantics (gloo/allreduce.cc:397-408): a prime
+// P degenerates to one step = direct reduce-scatter + allgather.
+std::vector<int> bcubeFactors(int P, int base) {
+  std::vector<int> f;
+  while (P % base == 0) {
+    f.push_back(base);
+    P /= base;
+  }
+  if (P > 1) {
+    f.push_back(P);
+  }
+  return f;
+}
+
 void bcubeAllreduce(AllreduceOptions& opts) {
   auto& ctx = opts.context;
   const int P = ctx->size;
@@ -202,26 +218,24 @@ void bcubeAllreduce(AllreduceOptions& opts) {
   const int base = std::max(2, ctx->base);
   char* out = static_cast<char*>(opts.outputs[0]);
 
-  // Number of steps: P must be base^k (caller dispatch guarantees).
-  int k = 0;
-  int p = 1;
-  while (p < P) {
-    p *= base;
-    k++;
+  const std::vector<int> factors = bcubeFactors(P, base);
+  const int k = static_cast<int>(factors.size());
+  int maxFactor = 2;
+  for (int f : factors) {
+    maxFactor = std::max(maxFactor, f);
   }
-  GA_ENFORCE_EQ(p, P, "bcube requires size == base^k");
 
   const uint64_t slot = Slot::build(SlotPrefix::kAllreduce, opts.tag);
   const auto timeout = opts.timeout.count() > 0 ? opts.timeout
                                                 : ctx->getTimeout();
 
-  // tmp: (base-1) slots sized for the largest piece we ever receive
-  // (step 0 receives a 1/base sub-span of the full buffer).
-  const size_t pieceCap = ((N + base - 1) / base + 1) * es;
-  auto tmp = makeAligned((base - 1) * pieceCap + 64);
+  // tmp: (maxFactor-1) slots sized for the largest piece we ever receive
+  // (step 0 receives a 1/factors[0] sub-span of the full buffer).
+  const size_t pieceCap = ((N + factors[0] - 1) / factors[0] + 1) * es;
+  auto tmp = makeAligned((maxFactor - 1) * pieceCap + 64);
   auto outBuf = ctx->createUnboundBuffer(out, N * es);
   auto tmpBuf =
-      ctx->createUnboundBuffer(tmp.get(), (base - 1) * pieceCap + 64);
+      ctx->createUnboundBuffer(tmp.get(), (maxFactor - 1) * pieceCap + 64);
 
   Seg span{0, N};
   std::vector<Seg> spanAtStep(k);
@@ -229,13 +243,14 @@ void bcubeAllreduce(AllreduceOptions& opts) {
   // Reduce-scatter stage.
   int stride = 1;
   for (int t = 0; t < k; t++) {
+    const int b = factors[t];
     spanAtStep[t] = span;
-    const int myIdx = (r / stride) % base;
+    const int myIdx = (r / stride) % b;
     // Post recvs of my sub-span from each peer.
-    Seg mine = subspanOf(span, myIdx, base);
+    Seg mine = subspanOf(span, myIdx, b);
     GA_ENFORCE_LE(mine.len * es, pieceCap);
     int nrecv = 0;
-    for (int j = 0; j < base; j++) {
+    for (int j = 0; j < b; j++) {
       if (j == myIdx) {
         continue;
       }
@@ -244,12 +259,12 @@ void bcubeAllreduce(AllreduceOptions& opts) {
       nrecv++;
     }
     // Send each peer its sub-span.
-    for (int j = 0; j < base; j++) {
+    for (int j = 0; j < b; j++) {
       if (j == myIdx) {
         continue;
       }
       const int peer = r + (j - myIdx) * stride;
-      Seg theirs = subspanOf(span, j, base);
+      Seg theirs = subspanOf(span, j, b);
       outBuf->send(peer, slot + t, theirs.off * es, theirs.len * es);
     }
     // Reduce received pieces (arrival order unknown across peers, but each
@@ -266,40 +281,41 @@ void bcubeAllreduce(AllreduceOptions& opts) {
             mine.len);
       }
     }
-    for (int j = 0; j < base - 1; j++) {
+    for (int j = 0; j < b - 1; j++) {
       outBuf->waitSend(timeout);
     }
     span = mine;
-    stride *= base;
+    stride *= b;
   }
 
   // Allgather stage (mirror).
   for (int t = k - 1; t >= 0; t--) {
-    stride /= base;
-    const int myIdx = (r / stride) % base;
+    const int b = factors[t];
+    stride /= b;
+    const int myIdx = (r / stride) % b;
     const Seg stepSpan = spanAtStep[t];
     // Post recvs for every other sub-span directly into out.
-    for (int j = 0; j < base; j++) {
+    for (int j = 0; j < b; j++) {
       if (j == myIdx) {
         continue;
       }
       const int peer = r + (j - myIdx) * stride;
-      Seg theirs = subspanOf(stepSpan, j, base);
+      Seg theirs = subspanOf(stepSpan, j, b);
       outBuf->recv(peer, slot + k + t, theirs.off * es, theirs.len * es);
     }
     // Send my sub-span to every peer.
-    Seg mine = subspanOf(stepSpan, myIdx, base);
-    for (int j = 0; j < base; j++) {
+    Seg mine = subspanOf(stepSpan, myIdx, b);
+    for (int j = 0; j < b; j++) {
       if (j == myIdx) {
         continue;
       }
       const int peer = r + (j - myIdx) * stride;
       outBuf->send(peer, slot + k + t, mine.off * es, mine.len * es);
     }
-    for (int j = 0; j < base - 1; j++) {
+    for (int j = 0; j < b - 1; j++) {
       outBuf->waitRecv(timeout);
     }
-    for (int j = 0; j < base - 1; j++) {
+    for (int j = 0; j < b - 1; j++) {
       outBuf->waitSend(timeout);
     }
     span = stepSpan;
@@ -319,17 +335,10 @@ void allreduce(AllreduceOptions& opts) {
 
   const int P = opts.context->size;
   if (P > 1 && opts.elements > 0) {
-    bool bcubeOk = [&] {
-      if (opts.algorithm != AllreduceOptions::Algorithm::BCUBE) {
-        return false;
-      }
-      int base = std::max(2, opts.context->base);
-      int p = 1;
-      while (p < P) {
-        p *= base;
-      }
-      return p == P;
-    }();
+    // bcube handles any size: P factorizes into repeated `base` factors
+    // plus a remainder step (prime P = one direct exchange step).
+    const bool bcubeOk =
+        opts.algorithm == AllreduceOptions::Algorithm::BCUBE;
     static const size_t smallThreshold = static_cast<size_t>(
         getEnvInt("GLOO_AMD_SMALL_ALLREDUCE", 16384));
     if (opts.elements * opts.elementSize <= smallThreshold &&

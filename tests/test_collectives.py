@@ -48,6 +48,30 @@ def test_allreduce_bcube_power(spawn_threads, size):
     spawn_threads(size, fn, base=2)
 
 
+@pytest.mark.parametrize("size,base", [
+    (6, 2),   # 2*3: repeated base factors + remainder group
+    (5, 2),   # prime: one direct-exchange step
+    (12, 2),  # 2*2*3
+    (9, 3),   # base 3 perfect power
+    (6, 3),   # 3*2
+])
+def test_allreduce_bcube_factorized(spawn_threads, size, base):
+    """True base-B grouped exchange at non-perfect-power sizes
+    (reference computeGroupSizePerStep, gloo/allreduce.cc:397-408)."""
+
+    def fn(ctx, rank, _):
+        x = fixture(rank, 10_001)
+        ga.allreduce(
+            ctx, [x.ctypes.data], x.size, ga.DType.f32, ga.ReduceOp.sum,
+            algorithm="bcube",
+        )
+        expected = sum(fixture(r, 10_001) for r in range(size))
+        assert np.allclose(x, expected)
+        return True
+
+    spawn_threads(size, fn, base=base)
+
+
 @pytest.mark.parametrize("op,npop", [
     (ga.ReduceOp.sum, np.add),
     (ga.ReduceOp.product, np.multiply),
