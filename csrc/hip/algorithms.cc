@@ -75,6 +75,17 @@ void gateStreams(
 // HipAllreduceRing
 // ===========================================================================
 
+HipAllreduceRing::~HipAllreduceRing() {
+  for (auto& kv : graphs_) {
+    if (kv.second.exec != nullptr) {
+      (void)hipGraphExecDestroy(kv.second.exec);
+    }
+  }
+  if (seqBaseDev_ != nullptr) {
+    (void)hipFree(seqBaseDev_);
+  }
+}
+
 HipAllreduceRing::HipAllreduceRing(
     std::shared_ptr<Context> ctx,
     int device,
@@ -232,14 +243,16 @@ void HipAllreduceRing::runHostStaged(
   cs_[0]->synchronize();
 }
 
-void HipAllreduceRing::enqueueRing(
+int HipAllreduceRing::enqueueRing(
     int j,
     char* work,
     size_t elemOff,
     size_t n,
     size_t es,
     DType dt,
-    ReduceOp op) {
+    ReduceOp op,
+    bool rel,
+    int64_t crossRunOff) {
   const int P = ctx_->size;
   const int r = ctx_->rank;
   const int stride = strides_[j];
@@ -302,10 +315,23 @@ void HipAllreduceRing::enqueueRing(
   for (int k = 0; k < K; k++) {
     const Step& st = steps[k];
     const int par = k & 1;
-    const uint64_t prevAck =
-        (k >= 2) ? seqOf(k - 2) : lastAckPerRing_[j][par];
-    if (prevAck > 0) {
-      launchWaitFlagGte(mesh_->flag(fACK_[j] + par), prevAck, ksm);
+    if (rel) {
+      // Graph mode: doorbell targets are *seqBaseDev_ + offset at
+      // EXECUTION time. Within-run gates are exact; the k<2 cross-run
+      // inbox-reuse gate needs the previous run's same-key shape
+      // (runDeviceGraph only replays when the previous run had this
+      // key), corrected per ring for K_j < max K (crossRunOff).
+      const int64_t off = (k >= 2)
+          ? static_cast<int64_t>(k) - 1
+          : crossRunOff + static_cast<int64_t>(k) - 1;
+      launchWaitFlagGteRel(mesh_->flag(fACK_[j] + par), seqBaseDev_, off,
+                           ksm);
+    } else {
+      const uint64_t prevAck =
+          (k >= 2) ? seqOf(k - 2) : lastAckPerRing_[j][par];
+      if (prevAck > 0) {
+        launchWaitFlagGte(mesh_->flag(fACK_[j] + par), prevAck, ksm);
+      }
     }
     if (k >= S) {
       evs[(k - S) % pool]->streamWait(ksm);
@@ -320,9 +346,16 @@ void HipAllreduceRing::enqueueRing(
           hipMemcpyDeviceToDevice,
           ksm));
     }
-    launchWriteFlag(mesh_->peerFlag(right, fDATA_[j] + par), seqOf(k), ksm);
-
-    launchWaitFlagGte(mesh_->flag(fDATA_[j] + par), seqOf(k), csm);
+    if (rel) {
+      launchWriteFlagRel(
+          mesh_->peerFlag(right, fDATA_[j] + par), seqBaseDev_, k + 1, ksm);
+      launchWaitFlagGteRel(
+          mesh_->flag(fDATA_[j] + par), seqBaseDev_, k + 1, csm);
+    } else {
+      launchWriteFlag(mesh_->peerFlag(right, fDATA_[j] + par), seqOf(k),
+                      ksm);
+      launchWaitFlagGte(mesh_->flag(fDATA_[j] + par), seqOf(k), csm);
+    }
     if (st.recv.len > 0) {
       if (st.reduceStep) {
         launchReduce2(
@@ -342,13 +375,31 @@ void HipAllreduceRing::enqueueRing(
             csm));
       }
     }
-    launchWriteFlag(mesh_->peerFlag(left, fACK_[j] + par), seqOf(k), csm);
+    if (rel) {
+      launchWriteFlagRel(
+          mesh_->peerFlag(left, fACK_[j] + par), seqBaseDev_, k + 1, csm);
+    } else {
+      launchWriteFlag(mesh_->peerFlag(left, fACK_[j] + par), seqOf(k), csm);
+    }
     evs[k % pool]->record(csm);
   }
-  for (int k = std::max(0, K - 2); k < K; k++) {
-    lastAckPerRing_[j][k & 1] = seqOf(k);
+  if (!rel) {
+    for (int k = std::max(0, K - 2); k < K; k++) {
+      lastAckPerRing_[j][k & 1] = seqOf(k);
+    }
   }
+  return K;
 }
+
+namespace {
+bool graphsEnabled() {
+  static int v = [] {
+    const char* e = std::getenv("GLOO_AMD_GRAPH");
+    return (e != nullptr && e[0] == '0') ? 0 : 1;
+  }();
+  return v == 1;
+}
+} // namespace
 
 void HipAllreduceRing::runDevice(
     char* buf,
@@ -363,6 +414,21 @@ void HipAllreduceRing::runDevice(
   // per allreduce.
   mesh_->ensureCapacity(0, inboxCap_);
   const int R = static_cast<int>(strides_.size());
+
+  // hipGraph replay: the many-segment multi-ring schedule is hundreds of
+  // host enqueues per run (the host becomes the bottleneck at 8 ranks x
+  // 4 rings); a captured graph replays it as one launch. Valid only in
+  // steady state: the previous run must have the same shape, so the
+  // first run of a shape goes eager (exact cross-run gates) and capture
+  // happens from the second run on.
+  GraphKey key{buf, n, static_cast<int>(dt), static_cast<int>(op)};
+  if (graphsEnabled() && !graphBroken_ && lastKey_ == key) {
+    if (runDeviceGraph(buf, n, es, dt, op)) {
+      return;
+    }
+  }
+  lastKey_ = key;
+
   for (int j = 0; j < R; j++) {
     // Carries the caller-stream dependency (gated in run()) into the
     // k<S steps of each ring's schedule.
@@ -374,7 +440,9 @@ void HipAllreduceRing::runDevice(
   for (int j = 0; j < R; j++) {
     Seg part = subspanOf({0, n}, j, R);
     if (part.len > 0) {
-      enqueueRing(j, buf, part.off, part.len, es, dt, op);
+      ringSteps_ = std::max<uint64_t>(
+          ringSteps_,
+          enqueueRing(j, buf, part.off, part.len, es, dt, op));
     }
   }
 
@@ -394,6 +462,145 @@ void HipAllreduceRing::runDevice(
     cs_[j]->synchronize();
   }
   seqBase_ += ringSteps_;
+}
+
+bool HipAllreduceRing::runDeviceGraph(
+    char* buf,
+    size_t n,
+    size_t es,
+    DType dt,
+    ReduceOp op) {
+  const int R = static_cast<int>(strides_.size());
+  if (seqBaseDev_ == nullptr) {
+    if (hipMalloc(reinterpret_cast<void**>(&seqBaseDev_), 8) !=
+        hipSuccess) {
+      graphBroken_ = true;
+      return false;
+    }
+  }
+  GraphKey key{buf, n, static_cast<int>(dt), static_cast<int>(op)};
+  auto it = graphs_.find(key);
+  if (it == graphs_.end()) {
+    if (graphs_.size() >= 8) {
+      // Shape churn: graphs stop paying for themselves.
+      graphBroken_ = true;
+      return false;
+    }
+    // Per-ring step counts (needed for the k<2 cross-run gate when a
+    // ring has fewer steps than the longest one).
+    GraphEntry entry;
+    entry.ringK.assign(R, 0);
+    int maxK = 0;
+    // Capture the whole multi-ring schedule rooted at cs_[0].
+    if (hipStreamBeginCapture(
+            cs_[0]->stream(), hipStreamCaptureModeThreadLocal) !=
+        hipSuccess) {
+      (void)hipGetLastError();
+      graphBroken_ = true;
+      return false;
+    }
+    bool ok = true;
+    try {
+      HipEvent fork(device_);
+      fork.record(cs_[0]->stream());
+      for (int j = 0; j < R; j++) {
+        if (j > 0) {
+          fork.streamWait(cs_[j]->stream());
+        }
+        fork.streamWait(ks_[j]->stream());
+        initEvent_[j]->record(cs_[j]->stream());
+      }
+      // Dry pass for per-ring K (pure shape math, no enqueue): K_j =
+      // 2 * (P-1) * S_j with S_j from the ring's partition length.
+      std::vector<Seg> parts(R);
+      for (int j = 0; j < R; j++) {
+        parts[j] = subspanOf({0, n}, j, R);
+      }
+      const size_t subCap = inboxCap_ / R;
+      for (int j = 0; j < R; j++) {
+        if (parts[j].len == 0) {
+          continue;
+        }
+        const size_t perRank =
+            (parts[j].len + ctx_->size - 1) / ctx_->size;
+        const int S = chunked_
+            ? std::max<int>(
+                  2,
+                  static_cast<int>((perRank * es + subCap - 1) / subCap))
+            : std::max<int>(
+                  1,
+                  static_cast<int>((perRank * es + subCap - 1) / subCap));
+        entry.ringK[j] = 2 * (ctx_->size - 1) * S;
+        maxK = std::max(maxK, entry.ringK[j]);
+      }
+      for (int j = 0; j < R; j++) {
+        if (parts[j].len > 0) {
+          const int K = enqueueRing(
+              j, buf, parts[j].off, parts[j].len, es, dt, op,
+              /*rel=*/true,
+              /*crossRunOff=*/static_cast<int64_t>(entry.ringK[j]) - maxK);
+          GA_ENFORCE_EQ(K, entry.ringK[j], "graph shape math diverged");
+        }
+      }
+      // Join every stream back into cs_[0].
+      for (int j = 0; j < R; j++) {
+        if (j > 0) {
+          doneEvent_[j]->record(cs_[j]->stream());
+          doneEvent_[j]->streamWait(cs_[0]->stream());
+        }
+        HipEvent join(device_);
+        join.record(ks_[j]->stream());
+        join.streamWait(cs_[0]->stream());
+      }
+    } catch (...) {
+      ok = false;
+    }
+    entry.steps = maxK;
+    hipGraph_t g = nullptr;
+    hipError_t ec = hipStreamEndCapture(cs_[0]->stream(), &g);
+    if (!ok || ec != hipSuccess || g == nullptr) {
+      (void)hipGetLastError();
+      if (g != nullptr) {
+        (void)hipGraphDestroy(g);
+      }
+      GA_WARN << "hipGraph capture of the ring schedule failed; "
+                 "falling back to eager enqueue";
+      graphBroken_ = true;
+      return false;
+    }
+    hipGraphExec_t exec = nullptr;
+    hipError_t ei = hipGraphInstantiate(&exec, g, nullptr, nullptr, 0);
+    (void)hipGraphDestroy(g);
+    if (ei != hipSuccess || exec == nullptr) {
+      (void)hipGetLastError();
+      GA_WARN << "hipGraph instantiate failed; eager fallback";
+      graphBroken_ = true;
+      return false;
+    }
+    entry.exec = exec;
+    it = graphs_.emplace(key, entry).first;
+    GA_INFO << "captured ring schedule graph: " << R << " rings, "
+            << maxK << " steps";
+  }
+
+  // Replay: bump the device seq base, launch, watchdog.
+  const GraphEntry& entry = it->second;
+  GA_HIP_CHECK(hipMemcpyAsync(
+      seqBaseDev_, &seqBase_, 8, hipMemcpyHostToDevice, cs_[0]->stream()));
+  GA_HIP_CHECK(hipGraphLaunch(entry.exec, cs_[0]->stream()));
+  doneEvent_[0]->record(cs_[0]->stream());
+  watchdogWait(
+      *doneEvent_[0], *mesh_, ctx_->getTimeout(),
+      "hip_allreduce_ring (graph)");
+  cs_[0]->synchronize();
+  // Keep the eager bookkeeping exact so eager and graph runs interleave.
+  for (int j = 0; j < R; j++) {
+    for (int k = std::max(0, entry.ringK[j] - 2); k < entry.ringK[j]; k++) {
+      lastAckPerRing_[j][k & 1] = seqBase_ + k + 1;
+    }
+  }
+  seqBase_ += entry.steps;
+  return true;
 }
 
 // ===========================================================================

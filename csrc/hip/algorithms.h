@@ -29,7 +29,10 @@
 #pragma once
 
 #include <array>
+#include <map>
 #include <memory>
+#include <mutex>
+#include <tuple>
 #include <vector>
 
 #include "collectives/reduce_fns.h"
@@ -77,20 +80,29 @@ class HipAllreduceRing {
     return mesh_.get();
   }
 
+  ~HipAllreduceRing();
+
  private:
   void runDevice(char* buf, size_t bytes, size_t n, DType dt, ReduceOp op);
   void runHostStaged(char* buf, size_t bytes, size_t n, DType dt, ReduceOp op);
+  // hipGraph steady-state replay of the whole multi-ring schedule (one
+  // launch instead of hundreds of enqueues). Returns false (and may set
+  // graphBroken_) when the eager path must run instead.
+  bool runDeviceGraph(char* buf, size_t n, size_t es, DType dt, ReduceOp op);
   // Enqueue one ring's full schedule (reduce-scatter + allgather over
   // the element range [elemOff, elemOff+elems) with neighbor stride
-  // strides_[j]) on that ring's stream pair.
-  void enqueueRing(
+  // strides_[j]) on that ring's stream pair; returns the step count.
+  // rel=true uses *seqBaseDev_-relative doorbells (graph capture).
+  int enqueueRing(
       int j,
       char* work,
       size_t elemOff,
       size_t elems,
       size_t es,
       DType dt,
-      ReduceOp op);
+      ReduceOp op,
+      bool rel = false,
+      int64_t crossRunOff = 0);
 
   std::shared_ptr<Context> ctx_;
   int device_;
@@ -111,6 +123,28 @@ class HipAllreduceRing {
   // pinned staging for the host path
   void* hostStage_{nullptr};
   size_t hostStageCap_{0};
+  // hipGraph replay state (see runDeviceGraph)
+  struct GraphKey {
+    const void* ptr;
+    size_t n;
+    int dt;
+    int op;
+    bool operator<(const GraphKey& o) const {
+      return std::tie(ptr, n, dt, op) < std::tie(o.ptr, o.n, o.dt, o.op);
+    }
+    bool operator==(const GraphKey& o) const {
+      return ptr == o.ptr && n == o.n && dt == o.dt && op == o.op;
+    }
+  };
+  struct GraphEntry {
+    hipGraphExec_t exec{nullptr};
+    int steps{0};
+    std::vector<int> ringK;
+  };
+  std::map<GraphKey, GraphEntry> graphs_;
+  GraphKey lastKey_{nullptr, 0, -1, -1};
+  uint64_t* seqBaseDev_{nullptr};
+  bool graphBroken_{false};
 };
 
 class HipAllreduceHalvingDoubling {

@@ -57,6 +57,21 @@ void launchWriteFlag(uint64_t* addr, uint64_t val, hipStream_t stream);
 // makes peer SDMA-written data visible to later kernels on the stream).
 void launchWaitFlagGte(const uint64_t* addr, uint64_t val, hipStream_t stream);
 
+// Relative doorbells for hipGraph replay: target = *basePtr + offset,
+// evaluated at execution time (offset may be negative; non-positive
+// targets are no-ops). Always kernel-based (stream-op packets cannot be
+// captured into graphs).
+void launchWriteFlagRel(
+    uint64_t* addr,
+    const uint64_t* basePtr,
+    int64_t offset,
+    hipStream_t stream);
+void launchWaitFlagGteRel(
+    const uint64_t* addr,
+    const uint64_t* basePtr,
+    int64_t offset,
+    hipStream_t stream);
+
 // Test-fixture fill: ptr[i] = (i % 47) * stride + val (reference parity:
 // gloo/cuda_private.cu:38-61 initializeMemory).
 void launchFillPattern(

@@ -287,6 +287,42 @@ __global__ __launch_bounds__(64) void waitFlagGteKernel(
   }
 }
 
+// Relative doorbells for hipGraph replay: the flag target is
+// *basePtr + offset, read at EXECUTION time, so one captured graph can
+// replay with monotonically advancing sequence numbers (the host bumps
+// the device-resident base before each launch). basePtr is local
+// device memory (plain load); the flag itself stays system-scope.
+__global__ __launch_bounds__(64) void writeFlagRelKernel(
+    uint64_t* addr,
+    const unsigned long long* basePtr,
+    long long offset) {
+  if (threadIdx.x == 0) {
+    const long long t = static_cast<long long>(*basePtr) + offset;
+    __hip_atomic_store(
+        addr,
+        static_cast<uint64_t>(t < 0 ? 0 : t),
+        __ATOMIC_RELEASE,
+        __HIP_MEMORY_SCOPE_SYSTEM);
+  }
+}
+
+__global__ __launch_bounds__(64) void waitFlagGteRelKernel(
+    const uint64_t* addr,
+    const unsigned long long* basePtr,
+    long long offset) {
+  if (threadIdx.x == 0) {
+    const long long t = static_cast<long long>(*basePtr) + offset;
+    if (t <= 0) {
+      return; // first-ever run: nothing to wait for
+    }
+    const uint64_t target = static_cast<uint64_t>(t);
+    while (__hip_atomic_load(addr, __ATOMIC_ACQUIRE,
+                             __HIP_MEMORY_SCOPE_SYSTEM) < target) {
+      __builtin_amdgcn_s_sleep(16);
+    }
+  }
+}
+
 template <typename T>
 __global__ __launch_bounds__(kBlock) void fillPatternKernel(
     T* __restrict__ ptr,
@@ -639,6 +675,30 @@ void launchWaitFlagGte(
   }
   hipLaunchKernelGGL(
       waitFlagGteKernel, dim3(1), dim3(64), 0, stream, addr, val);
+  GA_HIP_CHECK(hipGetLastError());
+}
+
+void launchWriteFlagRel(
+    uint64_t* addr,
+    const uint64_t* basePtr,
+    int64_t offset,
+    hipStream_t stream) {
+  hipLaunchKernelGGL(
+      writeFlagRelKernel, dim3(1), dim3(64), 0, stream, addr,
+      reinterpret_cast<const unsigned long long*>(basePtr),
+      static_cast<long long>(offset));
+  GA_HIP_CHECK(hipGetLastError());
+}
+
+void launchWaitFlagGteRel(
+    const uint64_t* addr,
+    const uint64_t* basePtr,
+    int64_t offset,
+    hipStream_t stream) {
+  hipLaunchKernelGGL(
+      waitFlagGteRelKernel, dim3(1), dim3(64), 0, stream, addr,
+      reinterpret_cast<const unsigned long long*>(basePtr),
+      static_cast<long long>(offset));
   GA_HIP_CHECK(hipGetLastError());
 }
 
