@@ -481,9 +481,8 @@ bool HipAllreduceRing::runDeviceGraph(
   GraphKey key{buf, n, static_cast<int>(dt), static_cast<int>(op)};
   auto it = graphs_.find(key);
   if (it == graphs_.end()) {
-    if (graphs_.size() >= 8) {
-      // Shape churn: graphs stop paying for themselves.
-      graphBroken_ = true;
+    if (graphs_.size() >= 16) {
+      // Shape churn: run new shapes eagerly, keep replaying cached ones.
       return false;
     }
     // Per-ring step counts (needed for the k<2 cross-run gate when a
