@@ -106,6 +106,19 @@ PYBIND11_MODULE(_C, m) {
     auto s = sched::subspanOf({off, len}, j, base);
     return std::make_pair(s.off, s.len);
   });
+  m.def("block_of_a", [](size_t n, int p, int b, size_t a) {
+    auto s = sched::blockOfA(n, p, b, a);
+    return std::make_pair(s.off, s.len);
+  });
+  m.def("segment_of_a", [](size_t n, int p, int b, int q, int S, size_t a) {
+    auto s = sched::segmentOfA(n, p, b, q, S, a);
+    return std::make_pair(s.off, s.len);
+  });
+  m.def("subspan_of_a", [](size_t off, size_t len, int j, int parts,
+                           size_t a) {
+    auto s = sched::subspanOfA({off, len}, j, parts, a);
+    return std::make_pair(s.off, s.len);
+  });
 
   // topology helpers (common/linux)
   m.def("list_interfaces", &listInterfaces);

@@ -54,5 +54,49 @@ inline Seg chunkOf(Seg span, int c, int nc) {
   return {start, end - start};
 }
 
+inline size_t alignUp(size_t v, size_t a) {
+  return a <= 1 ? v : ((v + a - 1) / a) * a;
+}
+inline size_t alignDown(size_t v, size_t a) {
+  return a <= 1 ? v : (v / a) * a;
+}
+
+// Aligned variants for the DEVICE engines: every split boundary is a
+// multiple of alignElems (elements worth of 16 bytes), so reduction
+// kernels stay on the vectorized 16-B/lane path and D2D copies stay
+// dwordx4-aligned; only the clamped tail piece may be shorter. All
+// ranks compute these from collective arguments, so wire peers agree.
+inline Seg blockOfA(size_t N, int P, int b, size_t A) {
+  size_t perRank = alignUp((N + P - 1) / P, A);
+  size_t start = std::min<size_t>(static_cast<size_t>(b) * perRank, N);
+  size_t end = std::min<size_t>(start + perRank, N);
+  return {start, end - start};
+}
+
+inline Seg segmentOfA(size_t N, int P, int b, int s, int S, size_t A) {
+  Seg blk = blockOfA(N, P, b, A);
+  size_t perSeg = alignUp((blk.len + S - 1) / S, A);
+  size_t start =
+      std::min(blk.off + static_cast<size_t>(s) * perSeg, blk.off + blk.len);
+  size_t end = std::min(start + perSeg, blk.off + blk.len);
+  return {start, end - start};
+}
+
+inline Seg subspanOfA(Seg span, int j, int parts, size_t A) {
+  size_t per = alignUp((span.len + parts - 1) / parts, A);
+  size_t start = std::min(span.off + static_cast<size_t>(j) * per,
+                          span.off + span.len);
+  size_t end = std::min(start + per, span.off + span.len);
+  return {start, end - start};
+}
+
+inline Seg chunkOfA(Seg span, int c, int nc, size_t A) {
+  size_t per = alignUp((span.len + nc - 1) / nc, A);
+  size_t start = std::min(span.off + static_cast<size_t>(c) * per,
+                          span.off + span.len);
+  size_t end = std::min(start + per, span.off + span.len);
+  return {start, end - start};
+}
+
 } // namespace sched
 } // namespace glooamd

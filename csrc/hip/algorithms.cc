@@ -243,6 +243,26 @@ void HipAllreduceRing::runHostStaged(
   cs_[0]->synchronize();
 }
 
+// Segment count for one ring's per-rank block: sized to the per-ring
+// inbox slot, with split boundaries rounded to 16 B (so a segment can
+// grow by up to 16 B past the naive ceil — hence the fit loop).
+int HipAllreduceRing::ringSegments(size_t partLen, size_t es) const {
+  const int P = ctx_->size;
+  const int R = static_cast<int>(strides_.size());
+  const size_t A = std::max<size_t>(1, 16 / es);
+  const size_t subCap = inboxCap_ / R;
+  const size_t perRank = sched::alignUp((partLen + P - 1) / P, A);
+  int S = chunked_
+      ? std::max<int>(
+            2, static_cast<int>((perRank * es + subCap - 1) / subCap))
+      : std::max<int>(
+            1, static_cast<int>((perRank * es + subCap - 1) / subCap));
+  while (sched::alignUp((perRank + S - 1) / S, A) * es > subCap) {
+    S++;
+  }
+  return S;
+}
+
 int HipAllreduceRing::enqueueRing(
     int j,
     char* work,
@@ -269,15 +289,12 @@ int HipAllreduceRing::enqueueRing(
 
   const int R = static_cast<int>(strides_.size());
   const size_t subCap = inboxCap_ / R; // per-(ring,parity) inbox bytes
-  const size_t perRank = (n + P - 1) / P;
-  const int S = chunked_
-      ? std::max<int>(
-            2, static_cast<int>((perRank * es + subCap - 1) / subCap))
-      : std::max<int>(
-            1, static_cast<int>((perRank * es + subCap - 1) / subCap));
+  const size_t A = std::max<size_t>(1, 16 / es); // 16-B split alignment
+  const int S = ringSegments(n, es);
   auto ringInbox = [&](char* base, int par) {
     return base + (static_cast<size_t>(j) * 2 + par) * subCap;
   };
+  (void)subCap;
 
   struct Step {
     Seg send;
@@ -289,14 +306,14 @@ int HipAllreduceRing::enqueueRing(
   steps.reserve(2 * K1);
   for (int k = 0; k < K1; k++) {
     int i = k / S, q = k % S;
-    steps.push_back({segmentOf(n, P, (vr - i + P) % P, q, S),
-                     segmentOf(n, P, (vr - i - 1 + 2 * P) % P, q, S),
+    steps.push_back({sched::segmentOfA(n, P, (vr - i + P) % P, q, S, A),
+                     sched::segmentOfA(n, P, (vr - i - 1 + 2 * P) % P, q, S, A),
                      true});
   }
   for (int k = 0; k < K1; k++) {
     int i = k / S, q = k % S;
-    steps.push_back({segmentOf(n, P, (vr + 1 - i + P) % P, q, S),
-                     segmentOf(n, P, (vr - i + P) % P, q, S),
+    steps.push_back({sched::segmentOfA(n, P, (vr + 1 - i + P) % P, q, S, A),
+                     sched::segmentOfA(n, P, (vr - i + P) % P, q, S, A),
                      false});
   }
   const int K = static_cast<int>(steps.size());
@@ -435,10 +452,12 @@ void HipAllreduceRing::runDevice(
     initEvent_[j]->record(cs_[j]->stream());
   }
 
-  // Partition the buffer across the rings and enqueue each schedule.
+  // Partition the buffer across the rings (16-B-aligned splits so the
+  // reductions stay on the vectorized path) and enqueue each schedule.
+  const size_t A = std::max<size_t>(1, 16 / es);
   ringSteps_ = 0;
   for (int j = 0; j < R; j++) {
-    Seg part = subspanOf({0, n}, j, R);
+    Seg part = sched::subspanOfA({0, n}, j, R, A);
     if (part.len > 0) {
       ringSteps_ = std::max<uint64_t>(
           ringSteps_,
@@ -511,25 +530,17 @@ bool HipAllreduceRing::runDeviceGraph(
       }
       // Dry pass for per-ring K (pure shape math, no enqueue): K_j =
       // 2 * (P-1) * S_j with S_j from the ring's partition length.
+      const size_t A = std::max<size_t>(1, 16 / es);
       std::vector<Seg> parts(R);
       for (int j = 0; j < R; j++) {
-        parts[j] = subspanOf({0, n}, j, R);
+        parts[j] = sched::subspanOfA({0, n}, j, R, A);
       }
-      const size_t subCap = inboxCap_ / R;
       for (int j = 0; j < R; j++) {
         if (parts[j].len == 0) {
           continue;
         }
-        const size_t perRank =
-            (parts[j].len + ctx_->size - 1) / ctx_->size;
-        const int S = chunked_
-            ? std::max<int>(
-                  2,
-                  static_cast<int>((perRank * es + subCap - 1) / subCap))
-            : std::max<int>(
-                  1,
-                  static_cast<int>((perRank * es + subCap - 1) / subCap));
-        entry.ringK[j] = 2 * (ctx_->size - 1) * S;
+        entry.ringK[j] =
+            2 * (ctx_->size - 1) * ringSegments(parts[j].len, es);
         maxK = std::max(maxK, entry.ringK[j]);
       }
       for (int j = 0; j < R; j++) {
@@ -668,8 +679,10 @@ void HipAllreduceHalvingDoubling::run(
   // Sub-inbox layout: the 2*inboxCap inbox region split into rows x 2
   // parities; rows = T steps plus (when non-pow2) one fold row.
   const int rows = T + (extras > 0 ? 1 : 0);
-  const size_t subBytes = (2 * mesh_->inboxCap()) / (2 * rows);
-  GA_ENFORCE_GE(subBytes, es, "inbox too small for the step count");
+  const size_t A = std::max<size_t>(1, 16 / es); // 16-B split alignment
+  const size_t subBytes =
+      sched::alignDown((2 * mesh_->inboxCap()) / (2 * rows), 16);
+  GA_ENFORCE_GE(subBytes, size_t(16), "inbox too small for the step count");
   auto subInbox = [&](int t, int par) {
     return mesh_->inbox(0) + (static_cast<size_t>(t) * 2 + par) * subBytes;
   };
@@ -769,11 +782,14 @@ void HipAllreduceHalvingDoubling::run(
     spanAt[t] = span;
     const int peer = r ^ (1 << t);
     const int bit = (r >> t) & 1;
-    Seg kp = subspanOf(span, bit, 2);
-    Seg gv = subspanOf(span, 1 - bit, 2);
-    const size_t maxHalf = std::max(kp.len, gv.len) * es;
-    const int nc =
-        std::max<int>(1, static_cast<int>((maxHalf + subBytes - 1) / subBytes));
+    Seg kp = sched::subspanOfA(span, bit, 2, A);
+    Seg gv = sched::subspanOfA(span, 1 - bit, 2, A);
+    const size_t maxHalf = std::max(kp.len, gv.len);
+    int nc = std::max<int>(
+        1, static_cast<int>((maxHalf * es + subBytes - 1) / subBytes));
+    while (sched::alignUp((maxHalf + nc - 1) / nc, A) * es > subBytes) {
+      nc++;
+    }
 
     // gv must be fully produced (previous step's reduces on cs).
     if (t == 0) {
@@ -789,7 +805,7 @@ void HipAllreduceHalvingDoubling::run(
         launchWaitFlagGte(
             mesh_->flag(fACK_ + t * 2 + par), lastAck_[t * 2 + par], ksm);
       }
-      Seg gch = chunkOf(gv, c, nc);
+      Seg gch = sched::chunkOfA(gv, c, nc, A);
       if (gch.len > 0) {
         GA_HIP_CHECK(hipMemcpyAsync(
             peerSubInbox(peer, t, par),
@@ -801,7 +817,7 @@ void HipAllreduceHalvingDoubling::run(
       launchWriteFlag(mesh_->peerFlag(peer, fDATA_ + t * 2 + par), dseq, ksm);
 
       launchWaitFlagGte(mesh_->flag(fDATA_ + t * 2 + par), dseq, csm);
-      Seg kch = chunkOf(kp, c, nc);
+      Seg kch = sched::chunkOfA(kp, c, nc, A);
       if (kch.len > 0) {
         launchReduce2(
             work + kch.off * es,
@@ -1037,7 +1053,9 @@ void HipAllreduceDirect::run(
     }
   }
   char* user = static_cast<char*>(ptrs[0]);
-  const size_t perRank = (elements + P - 1) / P;
+  // 16-B-aligned block splits keep the fused reduction vectorized.
+  const size_t perRank = sched::alignUp(
+      (elements + P - 1) / P, std::max<size_t>(1, 16 / es));
   // Chunk the per-rank block so scatter / reduce / broadcast pipeline
   // for large payloads. Inbox layout: P sources x 2 parities x chunkCap.
   const size_t chunkCap = std::min<size_t>(
@@ -1281,9 +1299,11 @@ void HipAllgatherRing::run(
     return;
   }
   const size_t totalBytes = blockBytes * P;
+  const size_t A = std::max<size_t>(1, 16 / es);
   const int S = std::max<int>(
       1, static_cast<int>((blockBytes + inboxCap_ - 1) / inboxCap_));
-  const size_t segCapBytes = ((inElements + S - 1) / S) * es;
+  const size_t segCapBytes =
+      sched::alignUp((inElements + S - 1) / S, A) * es;
   mesh_->ensureCapacity(totalBytes, segCapBytes);
   gateStreams(callerStream, {cs_->stream(), ks_->stream()});
 
@@ -1308,7 +1328,7 @@ void HipAllgatherRing::run(
     const int i = k / S, s = k % S;
     const int par = k & 1;
     const sched::Seg send =
-        sched::segmentOf(inElements, 1, 0, s, S); // within-block segment
+        sched::segmentOfA(inElements, 1, 0, s, S, A); // within-block segment
     const size_t sendBlock = (r - i + 2 * P) % P;
     const size_t recvBlock = (r - i - 1 + 2 * P) % P;
 
@@ -1401,9 +1421,11 @@ void HipReduceScatterRing::run(
     return;
   }
   const size_t totalBytes = blockBytes * P;
+  const size_t A = std::max<size_t>(1, 16 / es);
   const int S = std::max<int>(
       1, static_cast<int>((blockBytes + inboxCap_ - 1) / inboxCap_));
-  const size_t segCapBytes = ((recvElements + S - 1) / S) * es;
+  const size_t segCapBytes =
+      sched::alignUp((recvElements + S - 1) / S, A) * es;
   mesh_->ensureCapacity(totalBytes, segCapBytes);
   gateStreams(callerStream, {cs_->stream(), ks_->stream()});
 
@@ -1428,7 +1450,7 @@ void HipReduceScatterRing::run(
   for (int k = 0; k < K; k++) {
     const int i = k / S, s = k % S;
     const int par = k & 1;
-    const sched::Seg seg = sched::segmentOf(recvElements, 1, 0, s, S);
+    const sched::Seg seg = sched::segmentOfA(recvElements, 1, 0, s, S, A);
     const size_t sendBlock = (r - i - 1 + 2 * P) % P;
     const size_t recvBlock = (r - i - 2 + 2 * P) % P;
 

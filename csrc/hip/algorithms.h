@@ -89,6 +89,7 @@ class HipAllreduceRing {
   // launch instead of hundreds of enqueues). Returns false (and may set
   // graphBroken_) when the eager path must run instead.
   bool runDeviceGraph(char* buf, size_t n, size_t es, DType dt, ReduceOp op);
+  int ringSegments(size_t partLen, size_t es) const;
   // Enqueue one ring's full schedule (reduce-scatter + allgather over
   // the element range [elemOff, elemOff+elems) with neighbor stride
   // strides_[j]) on that ring's stream pair; returns the step count.

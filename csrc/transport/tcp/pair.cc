@@ -176,6 +176,28 @@ void TcpPair::close() {
   if (fd >= 0) {
     dev_->loop().unregisterDescriptor(fd);
     ioClose();
+    // Graceful shutdown: send FIN, then drain incoming bytes until the
+    // peer's EOF. Closing a TCP socket with unread rx data raises RST,
+    // and Linux discards the peer's unread receive queue on RST — which
+    // destroys in-flight final notifications on the peer (observed as a
+    // rare "closed by peer" failure in the benchmark's final barrier).
+    (void)::shutdown(fd, SHUT_WR);
+    char drainBuf[4096];
+    const auto drainDeadline =
+        std::chrono::steady_clock::now() + std::chrono::milliseconds(200);
+    while (std::chrono::steady_clock::now() < drainDeadline) {
+      ssize_t n = ::read(fd, drainBuf, sizeof(drainBuf));
+      if (n == 0) {
+        break; // EOF: the peer closed its side too
+      }
+      if (n < 0) {
+        if (errno == EAGAIN || errno == EWOULDBLOCK) {
+          usleep(1000);
+          continue;
+        }
+        break;
+      }
+    }
     ::close(fd);
   }
 }

@@ -40,3 +40,29 @@ def test_subspans_partition(n, base):
         assert off == cur
         cur += ln
     assert cur == n
+
+
+def test_aligned_splits_partition_and_align():
+    """Aligned variants: boundaries are multiples of A elements, pieces
+    tile the range exactly, only the tail may be short."""
+    import gloo_amd as ga
+
+    for n in [1, 7, 1000, 1041667, 5_000_000, 100_000_000]:
+        for P in [1, 2, 3, 7, 8]:
+            for A in [1, 4, 8]:
+                segs = [ga._C.block_of_a(n, P, b, A) for b in range(P)]
+                total = 0
+                for i, (off, ln) in enumerate(segs):
+                    assert off % A == 0 or off == n
+                    assert off == total
+                    total += ln
+                assert total == n
+                # segments within block 0
+                S = 5
+                t2 = 0
+                for s in range(S):
+                    off, ln = ga._C.segment_of_a(n, P, 0, s, S, A)
+                    assert off % A == 0 or off >= segs[0][1]
+                    assert off == t2
+                    t2 += ln
+                assert t2 == segs[0][1]

@@ -62,11 +62,23 @@ def test_prefix_store_exercise():
     _exercise(ga.PrefixStore("p", inner), ga.PrefixStore("p", inner))
 
 
-def test_tcp_store():
+def _tcp_store_on_free_port():
+    """Create a server TcpStore, retrying past ports already bound by
+    unrelated processes on a shared CI box."""
     import random
 
-    port = random.randint(20000, 40000)
-    server = ga.TcpStore("127.0.0.1", port, is_server=True)
+    last = None
+    for _ in range(20):
+        port = random.randint(20000, 60000)
+        try:
+            return ga.TcpStore("127.0.0.1", port, is_server=True), port
+        except Exception as e:  # noqa: BLE001 - bind collision
+            last = e
+    raise last
+
+
+def test_tcp_store():
+    server, port = _tcp_store_on_free_port()
     client = ga.TcpStore("127.0.0.1", port, is_server=False)
     _exercise(server, client)
     _exercise(client, server)
@@ -83,10 +95,7 @@ def test_store_v2_append_add(mk, tmp_path):
     elif mk == "prefix":
         s = ga.PrefixStore("ns", ga.HashStore())
     else:
-        import random
-
-        port = random.randint(20000, 40000)
-        s = ga.TcpStore("127.0.0.1", port, is_server=True)
+        s, _port = _tcp_store_on_free_port()
     assert s.has_v2()
     s.append("k", b"abc")
     s.append("k", b"def")
