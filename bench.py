@@ -71,6 +71,9 @@ def main():
                          "local-reduce-first path)")
     ap.add_argument("--sweep", action="store_true",
                     help="also print the reference element sweep to stderr")
+    ap.add_argument("--uds", default="auto", choices=["auto", "0", "1"],
+                    help="unix-socket control plane (auto: CPU-only "
+                         "localhost runs)")
     args = ap.parse_args()
 
     import gloo_amd as ga
@@ -96,8 +99,13 @@ def main():
     else:
         store = ga.HashStore()
     # CPU-only fallback is single-node by construction: use the
-    # unix-socket transport (GPU runs keep TCP for the control plane).
-    use_uds = (not have_gpu) and master_addr in ("127.0.0.1", "localhost")
+    # unix-socket transport (GPU runs keep TCP for the control plane
+    # unless --uds 1).
+    if args.uds == "auto":
+        use_uds = (not have_gpu) and master_addr in (
+            "127.0.0.1", "localhost")
+    else:
+        use_uds = args.uds == "1"
     dev = ga.create_tcp_device(
         master_addr if world > 1 else "", use_uds=use_uds)
     ctx = ga.Context(rank, world)

@@ -60,6 +60,34 @@ def main(path):
                 f"group by ks.display_name order by 3 desc limit 10"):
             print(f"  {row[0][:64]:64s} n={row[1]:5d} "
                   f"tot_ms={row[2]:9.2f} avg_us={row[3]:8.1f}")
+        # Same-device "peer" copies on a timeshared box run as blit
+        # KERNELS (__amd_rocclr_copyBuffer), not SDMA memcpies: compute
+        # copy-kernel vs reduce-kernel overlap per pid too.
+        kinds = {}
+        for kid, name in c.execute(
+                f"select id, display_name from "
+                f"rocpd_info_kernel_symbol_{sfx}"):
+            if "copyBuffer" in name:
+                kinds[kid] = "copy"
+            elif "reduce" in name.lower():
+                kinds[kid] = "reduce"
+        krows = list(c.execute(
+            f"select pid, kernel_id, start, end from "
+            f"rocpd_kernel_dispatch_{sfx}"))
+        for pid in sorted(set(r[0] for r in krows)):
+            cints = [(s, e) for p, k, s, e in krows
+                     if p == pid and kinds.get(k) == "copy"]
+            rints = [(s, e) for p, k, s, e in krows
+                     if p == pid and kinds.get(k) == "reduce"]
+            if not cints or not rints:
+                continue
+            cb, cm = union_busy(cints)
+            rb, rm = union_busy(rints)
+            ov = overlap(cm, rm)
+            print(f"  pid {pid}: blit-copy busy={cb/1e6:.2f}ms "
+                  f"reduce busy={rb/1e6:.2f}ms "
+                  f"copy∩reduce={ov/1e6:.2f}ms "
+                  f"({100*ov/max(min(cb,rb),1):.0f}% of the shorter)")
         try:
             rows = list(c.execute(
                 f"select pid, start, end, size from rocpd_memory_copy_{sfx}"))
