@@ -2,7 +2,11 @@
 
 #include <fcntl.h>
 #include <openssl/err.h>
+#include <poll.h>
 #include <unistd.h>
+
+#include <algorithm>
+#include <chrono>
 
 #include "common/logging.h"
 
@@ -113,15 +117,50 @@ TlsPair::~TlsPair() {
 void TlsPair::ioHandshake(bool initiator) {
   ssl_ = SSL_new(sslCtx_);
   GA_ENFORCE(ssl_ != nullptr, "SSL_new: ", sslErrors());
-  // Handshake runs blocking; the accepted-socket path arrives nonblocking.
-  setBlocking(fd(), true);
+  // Nonblocking handshake with a deadline (reference role:
+  // gloo/transport/tcp/tls/pair.cc:44-170 SSL_do_handshake state
+  // machine): SSL_do_handshake on a nonblocking fd, polling for the
+  // direction OpenSSL asks for, bounded by the pair timeout — a peer
+  // that stalls mid-handshake can no longer hang connect() forever.
+  setBlocking(fd(), false);
   GA_ENFORCE_EQ(SSL_set_fd(ssl_, fd()), 1, "SSL_set_fd: ", sslErrors());
-  int rv = initiator ? SSL_connect(ssl_) : SSL_accept(ssl_);
-  if (rv != 1) {
+  if (initiator) {
+    SSL_set_connect_state(ssl_);
+  } else {
+    SSL_set_accept_state(ssl_);
+  }
+  auto timeout = handshakeTimeout_;
+  if (timeout.count() <= 0) {
+    timeout = std::chrono::milliseconds(30000);
+  }
+  const auto deadline = std::chrono::steady_clock::now() + timeout;
+  for (;;) {
+    int rv = SSL_do_handshake(ssl_);
+    if (rv == 1) {
+      return;
+    }
     int err = SSL_get_error(ssl_, rv);
-    GA_THROW_IO(
-        "TLS handshake failed (", initiator ? "connect" : "accept",
-        ", err=", err, "): ", sslErrors());
+    if (err != SSL_ERROR_WANT_READ && err != SSL_ERROR_WANT_WRITE) {
+      GA_THROW_IO(
+          "TLS handshake failed (", initiator ? "connect" : "accept",
+          ", err=", err, "): ", sslErrors());
+    }
+    const auto now = std::chrono::steady_clock::now();
+    if (now >= deadline) {
+      GA_THROW_IO(
+          "TLS handshake timed out (", initiator ? "connect" : "accept",
+          ")");
+    }
+    struct pollfd pfd;
+    pfd.fd = fd();
+    pfd.events = (err == SSL_ERROR_WANT_READ) ? POLLIN : POLLOUT;
+    pfd.revents = 0;
+    const int waitMs = static_cast<int>(std::min<int64_t>(
+        std::chrono::duration_cast<std::chrono::milliseconds>(
+            deadline - now)
+            .count(),
+        100));
+    (void)::poll(&pfd, 1, std::max(waitMs, 1));
   }
 }
 

@@ -51,7 +51,9 @@ class TlsDevice : public TcpDevice {
 class TlsPair : public TcpPair {
  public:
   TlsPair(TcpContext* ctx, TlsDevice* dev, int peerRank)
-      : TcpPair(ctx, dev, peerRank), sslCtx_(dev->sslCtx()) {}
+      : TcpPair(ctx, dev, peerRank),
+        sslCtx_(dev->sslCtx()),
+        handshakeTimeout_(ctx->getTimeout()) {}
   ~TlsPair() override;
 
  protected:
@@ -63,6 +65,7 @@ class TlsPair : public TcpPair {
  private:
   SSL_CTX* sslCtx_;
   SSL* ssl_{nullptr};
+  std::chrono::milliseconds handshakeTimeout_;
   // OpenSSL SSL objects are not thread-safe: the loop thread reads while
   // a user thread may be flushing writes, so every SSL_* data call is
   // serialized per pair. Lock order is ctx-mutex -> sslMu_ (writes) or
