@@ -442,6 +442,50 @@ def test_hip_allreduce_direct_multi_input():
     _two_rank_generic(fn)
 
 
+def test_hip_allreduce_ring_graph_replay_soak():
+    """Long same-shape run chain: first run eager, second captures the
+    hipGraph, 18 more replay it — cross-run seq/inbox gating must hold
+    over a long monotonic doorbell chain."""
+    def fn(ctx, rank):
+        n = 1_000_000
+        g = torch.Generator("cpu").manual_seed(rank)
+        x = _dev(torch.rand(n, generator=g))
+        ref = sum(
+            torch.rand(n, generator=torch.Generator("cpu").manual_seed(r))
+            for r in range(2))
+        algo = ga._C.HipAllreduceRing(ctx, 0)
+        y = x.clone()
+        for it in range(20):
+            y.copy_(x)
+            algo.run(y.data_ptr(), n, ga.DType.f32, ga.ReduceOp.sum)
+            assert torch.allclose(_host(y), ref, atol=1e-4), (rank, it)
+            ga.barrier(ctx, tag=981)
+
+    _two_rank_generic(fn)
+
+
+def test_hip_allreduce_ring_alternating_shapes():
+    """Shape switches: a switch forces an eager run (exact cross-run
+    gates), repeats replay the cached graph; eager and graph runs must
+    interleave with exact doorbell bookkeeping."""
+    def fn(ctx, rank):
+        algo = ga._C.HipAllreduceRing(ctx, 0)
+        shapes = [1_000_000, 700_000, 1_000_000, 1_000_000,
+                  700_000, 700_000, 1_000_000]
+        for it, n in enumerate(shapes):
+            g = torch.Generator("cpu").manual_seed(rank * 100 + it)
+            x = _dev(torch.rand(n, generator=g))
+            ref = sum(
+                torch.rand(n, generator=torch.Generator(
+                    "cpu").manual_seed(r * 100 + it))
+                for r in range(2))
+            algo.run(x.data_ptr(), n, ga.DType.f32, ga.ReduceOp.sum)
+            assert torch.allclose(_host(x), ref, atol=1e-4), (rank, it, n)
+            ga.barrier(ctx, tag=985)
+
+    _two_rank_generic(fn)
+
+
 def test_hip_p2p_bidirectional_chunked():
     """HipP2P: both ranks post a send AND a recv before flushing either
     (batch_isend_irecv pattern); 20MB payloads exercise the multi-chunk
