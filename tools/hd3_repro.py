@@ -6,9 +6,11 @@ GLOO_AMD_FLAG_DEBUG so a hang dumps every rank's doorbell flags.
 Flag layout at P=3 (T=1): 0,1=fDATA 2,3=fACK 4=fAGD 5,6=fFOLD
 7,8=fFACK 9=fPOST 10=fPACK.
 """
+import faulthandler
 import os
 import sys
 import threading
+import time
 
 os.environ.setdefault("GLOO_AMD_FLAG_DEBUG", "1")
 sys.path.insert(0, "/root/repo")
@@ -16,6 +18,7 @@ sys.path.insert(0, "/root/repo")
 import torch  # noqa: E402
 import gloo_amd as ga  # noqa: E402
 
+faulthandler.dump_traceback_later(75, repeat=True)
 torch.cuda.set_device(0)
 
 
@@ -50,17 +53,19 @@ def one_round(ri):
             errors.append(f"[round {ri} rank {rank}]\n"
                           + traceback.format_exc())
 
+    t0 = time.time()
     ths = [threading.Thread(target=worker, args=(r,), daemon=True)
            for r in range(3)]
     [t.start() for t in ths]
     [t.join(90) for t in ths]
     alive = [t.is_alive() for t in ths]
+    dt = time.time() - t0
     if errors or any(alive):
-        print(f"ROUND {ri} FAILED alive={alive}")
+        print(f"ROUND {ri} FAILED alive={alive} dt={dt:.1f}s")
         for e in errors:
             print(e)
         return False
-    print(f"round {ri} ok", flush=True)
+    print(f"round {ri} ok dt={dt:.1f}s", flush=True)
     return True
 
 
