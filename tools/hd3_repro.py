@@ -22,6 +22,10 @@ faulthandler.dump_traceback_later(75, repeat=True)
 torch.cuda.set_device(0)
 
 
+SIZE = int(os.environ.get("REPRO_SIZE", "3"))
+ENGINE = os.environ.get("REPRO_ENGINE", "hd")
+
+
 def one_round(ri):
     store = ga.HashStore()
     errors = []
@@ -29,7 +33,7 @@ def one_round(ri):
     def worker(rank):
         try:
             dev = ga.create_tcp_device()
-            ctx = ga.Context(rank, 3)
+            ctx = ga.Context(rank, SIZE)
             ctx.connect_full_mesh(store, dev)
             ctx.set_timeout(20000)
             n = 1_500_000
@@ -37,8 +41,13 @@ def one_round(ri):
             x = torch.rand(n, generator=g).cuda()
             ref = sum(
                 torch.rand(n, generator=torch.Generator("cpu").manual_seed(r))
-                for r in range(3))
-            algo = ga._C.HipAllreduceHalvingDoubling(ctx, 0)
+                for r in range(SIZE))
+            if ENGINE == "hd":
+                algo = ga._C.HipAllreduceHalvingDoubling(ctx, 0)
+            elif ENGINE == "ring":
+                algo = ga._C.HipAllreduceRing(ctx, 0)
+            else:
+                algo = ga._C.HipAllreduceDirect(ctx, 0)
             y = x.clone()
             for it in range(5):
                 y.copy_(x)
@@ -55,7 +64,7 @@ def one_round(ri):
 
     t0 = time.time()
     ths = [threading.Thread(target=worker, args=(r,), daemon=True)
-           for r in range(3)]
+           for r in range(SIZE)]
     [t.start() for t in ths]
     [t.join(90) for t in ths]
     alive = [t.is_alive() for t in ths]
