@@ -54,6 +54,19 @@ void watchdogWait(
   }
 }
 
+// Phase tracing for hang localization (GLOO_AMD_PHASE_TRACE=1): the
+// last line printed before a stall names the blocking call.
+bool phaseTrace() {
+  static bool v = getEnvFlag("GLOO_AMD_PHASE_TRACE");
+  return v;
+}
+#define GA_PHASE(msg)                                        \
+  do {                                                       \
+    if (phaseTrace()) {                                      \
+      GA_ERROR << "[phase r" << ctx_->rank << "] " << msg;   \
+    }                                                        \
+  } while (0)
+
 // Order our streams after the caller's stream (where the input tensors
 // were produced). nullptr = legacy default stream, which covers torch's
 // default current stream. The transient event is destroyed immediately:
@@ -670,12 +683,15 @@ void HipAllreduceHalvingDoubling::run(
   if (P == 1 || elements == 0) {
     return;
   }
+  GA_PHASE("hd:enter");
   gateStreams(callerStream, {cs_->stream(), ks_->stream()});
+  GA_PHASE("hd:gated");
   char* buf = static_cast<char*>(devPtr);
   const int T = std::max(1, log2P_);
   const int extras = P - pow2_;
   const bool isExtra = r >= pow2_;
   mesh_->ensureCapacity(bytes, inboxCap_);
+  GA_PHASE("hd:capacity-ok");
   // Sub-inbox layout: the 2*inboxCap inbox region split into rows x 2
   // parities; rows = T steps plus (when non-pow2) one fold row.
   const int rows = T + (extras > 0 ? 1 : 0);
@@ -883,12 +899,16 @@ void HipAllreduceHalvingDoubling::run(
       hipMemcpyAsync(buf, work, bytes, hipMemcpyDeviceToDevice, csm));
   doneEvent_->record(csm);
 
+  GA_PHASE("hd:enqueued");
   auto timeout = ctx_->getTimeout();
   watchdogWait(*doneEvent_, *mesh_, timeout, "hip_allreduce_hd (cs)");
+  GA_PHASE("hd:cs-watchdog-ok");
   initEvent_->record(ksm);
   watchdogWait(*initEvent_, *mesh_, timeout, "hip_allreduce_hd (ks)");
+  GA_PHASE("hd:ks-watchdog-ok");
   cs_->synchronize();
   ks_->synchronize();
+  GA_PHASE("hd:done");
 }
 
 // ===========================================================================

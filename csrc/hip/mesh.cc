@@ -149,6 +149,7 @@ void XgmiMesh::releasePeers() {
 }
 
 void XgmiMesh::ensureCapacity(size_t workCap, size_t inboxCap) {
+  const bool trace = getEnvFlag("GLOO_AMD_PHASE_TRACE");
   if (workCap <= workCap_ && inboxCap <= inboxCap_) {
     // Still a collective decision: all ranks compute the same sizes from
     // the same collective arguments, so either all grow or none do.
@@ -161,12 +162,18 @@ void XgmiMesh::ensureCapacity(size_t workCap, size_t inboxCap) {
   // the old allocation: hipIpcGetMemHandle on an allocation recycled
   // from freed-while-peer-mapped pages fails with invalid argument, so
   // retired buffers live until the mesh is destroyed.
+  if (trace) {
+    GA_ERROR << "[phase r" << ctx_->rank << "] mesh:grow-barrier1";
+  }
   {
     BarrierOptions bar(ctx_);
     bar.tag = ctx_->nextSlot();
     barrier(bar);
   }
   releasePeers();
+  if (trace) {
+    GA_ERROR << "[phase r" << ctx_->rank << "] mesh:grow-barrier2";
+  }
   {
     BarrierOptions bar(ctx_);
     bar.tag = ctx_->nextSlot();
@@ -174,12 +181,21 @@ void XgmiMesh::ensureCapacity(size_t workCap, size_t inboxCap) {
   }
   retired_.push_back(data_);
   data_ = nullptr;
+  if (trace) {
+    GA_ERROR << "[phase r" << ctx_->rank << "] mesh:grow-malloc";
+  }
   {
     std::lock_guard<std::mutex> lock(allocMutex());
     GA_HIP_CHECK(hipMalloc(
         reinterpret_cast<void**>(&data_), workCap_ + 2 * inboxCap_));
   }
+  if (trace) {
+    GA_ERROR << "[phase r" << ctx_->rank << "] mesh:grow-exchange";
+  }
   exchange();
+  if (trace) {
+    GA_ERROR << "[phase r" << ctx_->rank << "] mesh:grow-done";
+  }
 }
 
 int XgmiMesh::allocFlags(int count) {
