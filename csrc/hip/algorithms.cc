@@ -69,18 +69,20 @@ bool phaseTrace() {
 
 // Order our streams after the caller's stream (where the input tensors
 // were produced). nullptr = legacy default stream, which covers torch's
-// default current stream. The transient event is destroyed immediately:
-// HIP frees it once the recorded work and waits complete.
+// default current stream. The gate event must be PERSISTENT (owned by
+// the engine): destroying an event while a hipStreamWaitEvent on it is
+// still pending on a busy stream can wedge that stream permanently
+// (observed as flaky engine-restart deadlocks; re-recording a
+// persistent event is well-defined — pending waits keep the snapshot
+// they were enqueued with).
 void gateStreams(
+    HipEvent& gateEv,
     hipStream_t caller,
     std::initializer_list<hipStream_t> gated) {
-  hipEvent_t ev = nullptr;
-  GA_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
-  GA_HIP_CHECK(hipEventRecord(ev, caller));
+  gateEv.record(caller);
   for (auto s : gated) {
-    GA_HIP_CHECK(hipStreamWaitEvent(s, ev, 0));
+    gateEv.streamWait(s);
   }
-  GA_HIP_CHECK(hipEventDestroy(ev));
 }
 } // namespace
 
@@ -149,6 +151,8 @@ HipAllreduceRing::HipAllreduceRing(
     fDATA_.push_back(mesh_->allocFlags(2));
     fACK_.push_back(mesh_->allocFlags(2));
   }
+  gateEv_ = std::make_unique<HipEvent>(device_);
+  graphFork_ = std::make_unique<HipEvent>(device_);
   lastAckPerRing_.assign(R, {0, 0});
 }
 
@@ -176,19 +180,10 @@ void HipAllreduceRing::run(
   if (elements == 0 || (ctx_->size == 1 && ptrs.size() == 1)) {
     return;
   }
-  {
-    std::vector<hipStream_t> gated;
-    for (size_t j = 0; j < cs_.size(); j++) {
-      gated.push_back(cs_[j]->stream());
-      gated.push_back(ks_[j]->stream());
-    }
-    hipEvent_t ev = nullptr;
-    GA_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
-    GA_HIP_CHECK(hipEventRecord(ev, callerStream));
-    for (auto g : gated) {
-      GA_HIP_CHECK(hipStreamWaitEvent(g, ev, 0));
-    }
-    GA_HIP_CHECK(hipEventDestroy(ev));
+  gateEv_->record(callerStream);
+  for (size_t j = 0; j < cs_.size(); j++) {
+    gateEv_->streamWait(cs_[j]->stream());
+    gateEv_->streamWait(ks_[j]->stream());
   }
   char* buf = static_cast<char*>(ptrs[0]);
   if (ptrs.size() > 1) {
@@ -204,11 +199,13 @@ void HipAllreduceRing::run(
       launchReduce2(buf, buf, ptrs[i], elements, dtype, op,
                     cs_[0]->stream());
     }
-    HipEvent red(device_);
-    red.record(cs_[0]->stream());
+    // Re-record the persistent gate event: the earlier caller-gating
+    // waits keep their snapshot; destroying a transient event with
+    // pending waits is what the persistent event exists to avoid.
+    gateEv_->record(cs_[0]->stream());
     for (size_t j = 0; j < cs_.size(); j++) {
-      red.streamWait(cs_[j]->stream());
-      red.streamWait(ks_[j]->stream());
+      gateEv_->streamWait(cs_[j]->stream());
+      gateEv_->streamWait(ks_[j]->stream());
     }
   }
   if (ctx_->size > 1) {
@@ -532,13 +529,12 @@ bool HipAllreduceRing::runDeviceGraph(
     }
     bool ok = true;
     try {
-      HipEvent fork(device_);
-      fork.record(cs_[0]->stream());
+      graphFork_->record(cs_[0]->stream());
       for (int j = 0; j < R; j++) {
         if (j > 0) {
-          fork.streamWait(cs_[j]->stream());
+          graphFork_->streamWait(cs_[j]->stream());
         }
-        fork.streamWait(ks_[j]->stream());
+        graphFork_->streamWait(ks_[j]->stream());
         initEvent_[j]->record(cs_[j]->stream());
       }
       // Dry pass for per-ring K (pure shape math, no enqueue): K_j =
@@ -565,15 +561,15 @@ bool HipAllreduceRing::runDeviceGraph(
           GA_ENFORCE_EQ(K, entry.ringK[j], "graph shape math diverged");
         }
       }
-      // Join every stream back into cs_[0].
+      // Join every stream back into cs_[0] (initEvent_[j] doubles as
+      // the ks join marker; re-recording inside the capture is fine).
       for (int j = 0; j < R; j++) {
         if (j > 0) {
           doneEvent_[j]->record(cs_[j]->stream());
           doneEvent_[j]->streamWait(cs_[0]->stream());
         }
-        HipEvent join(device_);
-        join.record(ks_[j]->stream());
-        join.streamWait(cs_[0]->stream());
+        initEvent_[j]->record(ks_[j]->stream());
+        initEvent_[j]->streamWait(cs_[0]->stream());
       }
     } catch (...) {
       ok = false;
@@ -656,6 +652,7 @@ HipAllreduceHalvingDoubling::HipAllreduceHalvingDoubling(
   stepEvent_ = std::make_unique<HipEvent>(device_);
   initEvent_ = std::make_unique<HipEvent>(device_);
   doneEvent_ = std::make_unique<HipEvent>(device_);
+  gateEv_ = std::make_unique<HipEvent>(device_);
   const int T = std::max(1, log2P_);
   fDATA_ = mesh_->allocFlags(T * 2);
   fACK_ = mesh_->allocFlags(T * 2);
@@ -684,7 +681,7 @@ void HipAllreduceHalvingDoubling::run(
     return;
   }
   GA_PHASE("hd:enter");
-  gateStreams(callerStream, {cs_->stream(), ks_->stream()});
+  gateStreams(*gateEv_, callerStream, {cs_->stream(), ks_->stream()});
   GA_PHASE("hd:gated");
   char* buf = static_cast<char*>(devPtr);
   const int T = std::max(1, log2P_);
@@ -924,6 +921,7 @@ HipBroadcastOneToAll::HipBroadcastOneToAll(
   GA_HIP_CHECK(hipSetDevice(device_));
   mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, 4096);
   cs_ = pooledStream(ctx_.get(), device_, 0);
+  gateEv_ = std::make_unique<HipEvent>(device_);
   const int nf = std::min(
       {numStreams, std::max(1, ctx_->size - 1), kStreamPoolSize - 1});
   for (int i = 0; i < nf; i++) {
@@ -945,7 +943,7 @@ void HipBroadcastOneToAll::run(
     return;
   }
   mesh_->ensureCapacity(bytes, 4096);
-  gateStreams(callerStream, {cs_->stream()});
+  gateStreams(*gateEv_, callerStream, {cs_->stream()});
   char* work = mesh_->work();
   char* buf = static_cast<char*>(devPtr);
   seq_++;
@@ -1012,6 +1010,7 @@ HipAllreduceDirect::HipAllreduceDirect(
     fanout_.push_back(pooledStream(ctx_.get(), device_, 1 + i));
   }
   doneEvent_ = std::make_unique<HipEvent>(device_);
+  gateEv_ = std::make_unique<HipEvent>(device_);
   for (int i = 0; i < 4; i++) {
     chunkEvents_.push_back(std::make_unique<HipEvent>(device_));
   }
@@ -1051,7 +1050,7 @@ void HipAllreduceDirect::run(
     // Fused local reduction of the caller's pointers into ptrs[0] before
     // the wire phase (reference cuda_allreduce_ring.cc:72-120 role).
     if (ptrs.size() > 1) {
-      gateStreams(callerStream, {cs_->stream()});
+      gateStreams(*gateEv_, callerStream, {cs_->stream()});
       const int k = static_cast<int>(std::min<size_t>(ptrs.size(), 8));
       launchReduceN(
           ptrs[0], const_cast<const void* const*>(ptrs.data()), k, elements,
@@ -1111,18 +1110,10 @@ void HipAllreduceDirect::run(
         chunkOffIn(b, c);
   };
 
-  {
-    std::vector<hipStream_t> gated{cs_->stream()};
-    for (auto* st : fanout_) {
-      gated.push_back(st->stream());
-    }
-    hipEvent_t ev = nullptr;
-    GA_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
-    GA_HIP_CHECK(hipEventRecord(ev, callerStream));
-    for (auto g : gated) {
-      GA_HIP_CHECK(hipStreamWaitEvent(g, ev, 0));
-    }
-    GA_HIP_CHECK(hipEventDestroy(ev));
+  gateEv_->record(callerStream);
+  gateEv_->streamWait(cs_->stream());
+  for (auto* st : fanout_) {
+    gateEv_->streamWait(st->stream());
   }
 
   const uint64_t runSeq = ++runSeq_;
@@ -1296,6 +1287,7 @@ HipAllgatherRing::HipAllgatherRing(
   ks_ = pooledStream(ctx_.get(), device_, 1);
   initEvent_ = std::make_unique<HipEvent>(device_);
   doneEvent_ = std::make_unique<HipEvent>(device_);
+  gateEv_ = std::make_unique<HipEvent>(device_);
   fDATA_ = mesh_->allocFlags(2);
   fACK_ = mesh_->allocFlags(2);
 }
@@ -1325,7 +1317,7 @@ void HipAllgatherRing::run(
   const size_t segCapBytes =
       sched::alignUp((inElements + S - 1) / S, A) * es;
   mesh_->ensureCapacity(totalBytes, segCapBytes);
-  gateStreams(callerStream, {cs_->stream(), ks_->stream()});
+  gateStreams(*gateEv_, callerStream, {cs_->stream(), ks_->stream()});
 
   const int right = (r + 1) % P;
   const int left = (r - 1 + P) % P;
@@ -1416,6 +1408,7 @@ HipReduceScatterRing::HipReduceScatterRing(
   ks_ = pooledStream(ctx_.get(), device_, 1);
   initEvent_ = std::make_unique<HipEvent>(device_);
   doneEvent_ = std::make_unique<HipEvent>(device_);
+  gateEv_ = std::make_unique<HipEvent>(device_);
   fDATA_ = mesh_->allocFlags(2);
   fACK_ = mesh_->allocFlags(2);
 }
@@ -1447,7 +1440,7 @@ void HipReduceScatterRing::run(
   const size_t segCapBytes =
       sched::alignUp((recvElements + S - 1) / S, A) * es;
   mesh_->ensureCapacity(totalBytes, segCapBytes);
-  gateStreams(callerStream, {cs_->stream(), ks_->stream()});
+  gateStreams(*gateEv_, callerStream, {cs_->stream(), ks_->stream()});
 
   const int right = (r + 1) % P;
   const int left = (r - 1 + P) % P;
@@ -1536,6 +1529,7 @@ HipAlltoall::HipAlltoall(
   GA_HIP_CHECK(hipSetDevice(device_));
   mesh_ = std::make_unique<XgmiMesh>(ctx_, device_, 0, 4096);
   cs_ = pooledStream(ctx_.get(), device_, 0);
+  gateEv_ = std::make_unique<HipEvent>(device_);
   const int nf = std::min(
       {numStreams, std::max(1, ctx_->size - 1), kStreamPoolSize - 1});
   for (int i = 0; i < nf; i++) {
@@ -1565,18 +1559,10 @@ void HipAlltoall::run(
     return;
   }
   mesh_->ensureCapacity(blockBytes * P, 4096);
-  {
-    std::vector<hipStream_t> gated{cs_->stream()};
-    for (auto& st : fanout_) {
-      gated.push_back(st->stream());
-    }
-    hipEvent_t ev = nullptr;
-    GA_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
-    GA_HIP_CHECK(hipEventRecord(ev, callerStream));
-    for (auto g : gated) {
-      GA_HIP_CHECK(hipStreamWaitEvent(g, ev, 0));
-    }
-    GA_HIP_CHECK(hipEventDestroy(ev));
+  gateEv_->record(callerStream);
+  gateEv_->streamWait(cs_->stream());
+  for (auto& st : fanout_) {
+    gateEv_->streamWait(st->stream());
   }
   char* work = mesh_->work();
   const uint64_t seq = ++seq_;
@@ -1653,6 +1639,8 @@ HipP2P::HipP2P(std::shared_ptr<Context> ctx, int device, size_t chunkCap)
       ctx_, device_, 0, static_cast<size_t>(ctx_->size) * chunkCap_);
   ss_ = std::make_unique<HipStream>(device_);
   rs_ = std::make_unique<HipStream>(device_);
+  gateSendEv_ = std::make_unique<HipEvent>(device_);
+  gateRecvEv_ = std::make_unique<HipEvent>(device_);
   fDATA_ = mesh_->allocFlags(ctx_->size);
   fACK_ = mesh_->allocFlags(ctx_->size);
   sendChunkSeq_.assign(ctx_->size, 0);
@@ -1669,7 +1657,7 @@ void HipP2P::postSend(
   GA_HIP_CHECK(hipSetDevice(device_));
   const int r = ctx_->rank;
   GA_ENFORCE(dst != r, "p2p send to self");
-  gateStreams(callerStream, {ss_->stream()});
+  gateStreams(*gateSendEv_, callerStream, {ss_->stream()});
   const char* src = static_cast<const char*>(devPtr);
   const size_t nc = (bytes + chunkCap_ - 1) / chunkCap_;
   for (size_t c = 0; c < std::max<size_t>(nc, 1); c++) {
@@ -1702,7 +1690,7 @@ void HipP2P::postRecv(
   std::lock_guard<std::mutex> lock(mu_);
   GA_HIP_CHECK(hipSetDevice(device_));
   GA_ENFORCE(src != ctx_->rank, "p2p recv from self");
-  gateStreams(callerStream, {rs_->stream()});
+  gateStreams(*gateRecvEv_, callerStream, {rs_->stream()});
   char* dstPtr = static_cast<char*>(devPtr);
   const size_t nc = (bytes + chunkCap_ - 1) / chunkCap_;
   for (size_t c = 0; c < std::max<size_t>(nc, 1); c++) {
@@ -1759,11 +1747,13 @@ void hipAllreduceLocal(
   // Cached per-thread stream: creation costs ~300us, far more than the
   // kernels themselves for MB-scale buffers.
   static thread_local std::unique_ptr<HipStream> cached;
+  static thread_local std::unique_ptr<HipEvent> cachedEv;
   if (!cached || cached->device() != device) {
     cached = std::make_unique<HipStream>(device);
+    cachedEv = std::make_unique<HipEvent>(device);
   }
   HipStream& s = *cached;
-  gateStreams(callerStream, {s.stream()});
+  gateStreams(*cachedEv, callerStream, {s.stream()});
   const size_t es = dtypeSize(dtype);
   if (ptrs.size() <= 8) {
     // Fused: one pass reduces and broadcasts (every pointer gets the

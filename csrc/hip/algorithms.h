@@ -117,6 +117,10 @@ class HipAllreduceRing {
   std::vector<std::vector<std::unique_ptr<HipEvent>>> events_;
   std::vector<std::unique_ptr<HipEvent>> initEvent_;
   std::vector<std::unique_ptr<HipEvent>> doneEvent_;
+  // Persistent gate events: a transient event destroyed with pending
+  // stream waits can wedge a stream (see gateStreams in algorithms.cc).
+  std::unique_ptr<HipEvent> gateEv_;
+  std::unique_ptr<HipEvent> graphFork_;
   std::vector<int> fDATA_, fACK_; // flag indices (x2 parity each, per ring)
   std::vector<std::array<uint64_t, 2>> lastAckPerRing_;
   uint64_t seqBase_{0};
@@ -174,6 +178,7 @@ class HipAllreduceHalvingDoubling {
   std::unique_ptr<HipEvent> stepEvent_;
   std::unique_ptr<HipEvent> initEvent_;
   std::unique_ptr<HipEvent> doneEvent_;
+  std::unique_ptr<HipEvent> gateEv_;
   std::vector<std::unique_ptr<HipEvent>> stepEvents_;
   int fDATA_, fACK_, fAGD_; // per step t: [t*2 + parity]
   // Non-pow2 folding flags (allocated only when size != pow2_).
@@ -209,6 +214,7 @@ class HipBroadcastOneToAll {
   int root_;
   std::unique_ptr<XgmiMesh> mesh_;
   HipStream* cs_;
+  std::unique_ptr<HipEvent> gateEv_;
   std::vector<HipStream*> fanout_;
   int fBDATA_;
   int fBACK_;
@@ -252,6 +258,7 @@ class HipAllreduceDirect {
   HipStream* cs_;
   std::vector<HipStream*> fanout_;
   std::unique_ptr<HipEvent> doneEvent_;
+  std::unique_ptr<HipEvent> gateEv_;
   std::vector<std::unique_ptr<HipEvent>> chunkEvents_;
   int fRS_; // [src] scatter chunk arrived (monotonic per-chunk seq)
   int fAG_; // [src] reduced chunk arrived
@@ -314,6 +321,7 @@ class HipAllgatherRing {
   std::vector<std::unique_ptr<HipEvent>> events_;
   std::unique_ptr<HipEvent> initEvent_;
   std::unique_ptr<HipEvent> doneEvent_;
+  std::unique_ptr<HipEvent> gateEv_;
   int fDATA_, fACK_;
   uint64_t seqBase_{0};
   uint64_t lastAck_[2] = {0, 0};
@@ -345,6 +353,7 @@ class HipReduceScatterRing {
   std::vector<std::unique_ptr<HipEvent>> events_;
   std::unique_ptr<HipEvent> initEvent_;
   std::unique_ptr<HipEvent> doneEvent_;
+  std::unique_ptr<HipEvent> gateEv_;
   int fDATA_, fACK_;
   uint64_t seqBase_{0};
   uint64_t lastAck_[2] = {0, 0};
@@ -371,6 +380,7 @@ class HipAlltoall {
   int device_;
   std::unique_ptr<XgmiMesh> mesh_;
   HipStream* cs_;
+  std::unique_ptr<HipEvent> gateEv_;
   std::vector<HipStream*> fanout_;
   int fDATA_; // [src]
   int fACK_; // [src]
@@ -423,6 +433,8 @@ class HipP2P {
   // with a collective engine's schedule on a shared stream.
   std::unique_ptr<HipStream> ss_;
   std::unique_ptr<HipStream> rs_;
+  std::unique_ptr<HipEvent> gateSendEv_;
+  std::unique_ptr<HipEvent> gateRecvEv_;
   int fDATA_; // [src]: chunks sent to me by src (monotonic)
   int fACK_; // [dst]: chunks dst consumed of my sends (monotonic)
   std::vector<uint64_t> sendChunkSeq_; // per dst lane
