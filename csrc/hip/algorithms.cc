@@ -91,6 +91,10 @@ void gateStreams(
 // ===========================================================================
 
 HipAllreduceRing::~HipAllreduceRing() {
+  for (size_t j = 0; j < cs_.size(); j++) {
+    (void)hipStreamSynchronize(cs_[j]->stream());
+    (void)hipStreamSynchronize(ks_[j]->stream());
+  }
   for (auto& kv : graphs_) {
     if (kv.second.exec != nullptr) {
       (void)hipGraphExecDestroy(kv.second.exec);
@@ -908,6 +912,19 @@ void HipAllreduceHalvingDoubling::run(
   GA_PHASE("hd:done");
 }
 
+
+// Drain this engine's streams before members (mesh: flags + data pages)
+// destruct: a failed, poisoned run can leave released-but-still-
+// executing work on the POOLED streams, which outlive the engine.
+HipAllreduceHalvingDoubling::~HipAllreduceHalvingDoubling() {
+  if (cs_ != nullptr) {
+    (void)hipStreamSynchronize(cs_->stream());
+  }
+  if (ks_ != nullptr) {
+    (void)hipStreamSynchronize(ks_->stream());
+  }
+}
+
 // ===========================================================================
 // HipBroadcastOneToAll
 // ===========================================================================
@@ -988,6 +1005,15 @@ void HipBroadcastOneToAll::run(
     done.record(cs_->stream());
     watchdogWait(done, *mesh_, timeout, "hip_broadcast (recv)");
     cs_->synchronize();
+  }
+}
+
+HipBroadcastOneToAll::~HipBroadcastOneToAll() {
+  if (cs_ != nullptr) {
+    (void)hipStreamSynchronize(cs_->stream());
+  }
+  for (auto* st : fanout_) {
+    (void)hipStreamSynchronize(st->stream());
   }
 }
 
@@ -1236,6 +1262,15 @@ void HipAllreduceDirect::run(
   }
 }
 
+HipAllreduceDirect::~HipAllreduceDirect() {
+  if (cs_ != nullptr) {
+    (void)hipStreamSynchronize(cs_->stream());
+  }
+  for (auto* st : fanout_) {
+    (void)hipStreamSynchronize(st->stream());
+  }
+}
+
 // ===========================================================================
 // HipAllreduceBcube
 // ===========================================================================
@@ -1391,6 +1426,15 @@ void HipAllgatherRing::run(
   seqBase_ += K;
 }
 
+HipAllgatherRing::~HipAllgatherRing() {
+  if (cs_ != nullptr) {
+    (void)hipStreamSynchronize(cs_->stream());
+  }
+  if (ks_ != nullptr) {
+    (void)hipStreamSynchronize(ks_->stream());
+  }
+}
+
 // ===========================================================================
 // HipReduceScatterRing
 // ===========================================================================
@@ -1517,6 +1561,15 @@ void HipReduceScatterRing::run(
   seqBase_ += K;
 }
 
+HipReduceScatterRing::~HipReduceScatterRing() {
+  if (cs_ != nullptr) {
+    (void)hipStreamSynchronize(cs_->stream());
+  }
+  if (ks_ != nullptr) {
+    (void)hipStreamSynchronize(ks_->stream());
+  }
+}
+
 // ===========================================================================
 // HipAlltoall
 // ===========================================================================
@@ -1621,6 +1674,15 @@ void HipAlltoall::run(
     done.record(st->stream());
     watchdogWait(done, *mesh_, ctx_->getTimeout(), "hip_alltoall (fanout)");
     st->synchronize();
+  }
+}
+
+HipAlltoall::~HipAlltoall() {
+  if (cs_ != nullptr) {
+    (void)hipStreamSynchronize(cs_->stream());
+  }
+  for (auto* st : fanout_) {
+    (void)hipStreamSynchronize(st->stream());
   }
 }
 

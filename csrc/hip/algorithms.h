@@ -165,6 +165,7 @@ class HipAllreduceHalvingDoubling {
       DType dtype,
       ReduceOp op,
       hipStream_t callerStream = nullptr);
+  ~HipAllreduceHalvingDoubling();
 
  private:
   std::shared_ptr<Context> ctx_;
@@ -203,6 +204,7 @@ class HipBroadcastOneToAll {
       int numStreams = 4);
 
   void run(void* devPtr, size_t bytes, hipStream_t callerStream = nullptr);
+  ~HipBroadcastOneToAll();
 
   std::vector<uint64_t> debugFlags() {
     return mesh_->readFlags();
@@ -250,6 +252,7 @@ class HipAllreduceDirect {
       DType dtype,
       ReduceOp op,
       hipStream_t callerStream = nullptr);
+  ~HipAllreduceDirect();
 
  private:
   std::shared_ptr<Context> ctx_;
@@ -310,6 +313,7 @@ class HipAllgatherRing {
       size_t inElements,
       size_t es,
       hipStream_t callerStream = nullptr);
+  ~HipAllgatherRing();
 
  private:
   std::shared_ptr<Context> ctx_;
@@ -342,6 +346,7 @@ class HipReduceScatterRing {
       DType dtype,
       ReduceOp op,
       hipStream_t callerStream = nullptr);
+  ~HipReduceScatterRing();
 
  private:
   std::shared_ptr<Context> ctx_;
@@ -374,6 +379,7 @@ class HipAlltoall {
       size_t perRankElements,
       size_t es,
       hipStream_t callerStream = nullptr);
+  ~HipAlltoall();
 
  private:
   std::shared_ptr<Context> ctx_;

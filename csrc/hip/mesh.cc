@@ -29,6 +29,37 @@ void* allocFineGrained(size_t bytes) {
   return p;
 }
 
+// Flag pages are IMMORTAL: freed-and-recycled flag addresses are the
+// one thing a stale doorbell packet (CP wait-value or spin kernel left
+// on a pooled stream by a failed, poisoned run) can still reference.
+// Reusing a page keeps any such stale reference pointing at poisoned
+// values (harmless) instead of freed memory. A page is 4 KiB; a
+// process churning meshes leaks a handful at most.
+std::mutex& flagPoolMutex() {
+  static std::mutex mu;
+  return mu;
+}
+std::vector<uint64_t*>& flagPool() {
+  static auto* pool = new std::vector<uint64_t*>();
+  return *pool;
+}
+uint64_t* acquireFlagPage(size_t count) {
+  {
+    std::lock_guard<std::mutex> lock(flagPoolMutex());
+    auto& pool = flagPool();
+    if (!pool.empty()) {
+      uint64_t* p = pool.back();
+      pool.pop_back();
+      return p;
+    }
+  }
+  return static_cast<uint64_t*>(allocFineGrained(count * 8));
+}
+void releaseFlagPage(uint64_t* p) {
+  std::lock_guard<std::mutex> lock(flagPoolMutex());
+  flagPool().push_back(p);
+}
+
 struct HandleBlob {
   int32_t pid;
   int32_t device;
@@ -54,7 +85,7 @@ XgmiMesh::XgmiMesh(
     std::lock_guard<std::mutex> lock(allocMutex());
     GA_HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&data_), total));
   }
-  flags_ = static_cast<uint64_t*>(allocFineGrained(kNumFlags * 8));
+  flags_ = acquireFlagPage(kNumFlags);
   // hipMemset (sync form) completes before returning; no device-wide
   // sync here (it could block on other ranks' in-flight kernels when
   // several ranks share one GPU).
@@ -227,10 +258,10 @@ std::vector<uint64_t> XgmiMesh::readFlags(int n) {
 }
 
 XgmiMesh::~XgmiMesh() {
-  // No device-wide sync here: the owning algorithm's streams are synced
-  // by their own destructors (declared after the mesh, destroyed first);
-  // a device-wide wait could block on OTHER ranks' in-flight kernels
-  // when several ranks share one GPU.
+  // No device-wide sync here: the owning engine's destructor drains its
+  // streams before the mesh member is destroyed; a device-wide wait
+  // could block on OTHER ranks' in-flight kernels when several ranks
+  // share one GPU.
   releasePeers();
   std::lock_guard<std::mutex> lock(allocMutex());
   if (data_ != nullptr) {
@@ -240,7 +271,7 @@ XgmiMesh::~XgmiMesh() {
     (void)hipFree(p);
   }
   if (flags_ != nullptr) {
-    (void)hipFree(flags_);
+    releaseFlagPage(flags_); // recycled, never freed (see acquireFlagPage)
   }
 }
 
