@@ -75,10 +75,23 @@ bool phaseTrace() {
 // (observed as flaky engine-restart deadlocks; re-recording a
 // persistent event is well-defined — pending waits keep the snapshot
 // they were enqueued with).
+// GLOO_AMD_GATE_SYNC=1: instead of recording a gate event on the
+// caller's stream, host-synchronize it (isolation knob for legacy
+// null-stream event-record hazards in multi-threaded processes; costs
+// compute/comm overlap, so event gating stays the default).
+bool gateBySync() {
+  static bool v = getEnvFlag("GLOO_AMD_GATE_SYNC");
+  return v;
+}
+
 void gateStreams(
     HipEvent& gateEv,
     hipStream_t caller,
     std::initializer_list<hipStream_t> gated) {
+  if (gateBySync()) {
+    GA_HIP_CHECK(hipStreamSynchronize(caller));
+    return;
+  }
   gateEv.record(caller);
   for (auto s : gated) {
     gateEv.streamWait(s);
@@ -184,10 +197,14 @@ void HipAllreduceRing::run(
   if (elements == 0 || (ctx_->size == 1 && ptrs.size() == 1)) {
     return;
   }
-  gateEv_->record(callerStream);
-  for (size_t j = 0; j < cs_.size(); j++) {
-    gateEv_->streamWait(cs_[j]->stream());
-    gateEv_->streamWait(ks_[j]->stream());
+  if (gateBySync()) {
+    GA_HIP_CHECK(hipStreamSynchronize(callerStream));
+  } else {
+    gateEv_->record(callerStream);
+    for (size_t j = 0; j < cs_.size(); j++) {
+      gateEv_->streamWait(cs_[j]->stream());
+      gateEv_->streamWait(ks_[j]->stream());
+    }
   }
   char* buf = static_cast<char*>(ptrs[0]);
   if (ptrs.size() > 1) {
@@ -1136,10 +1153,14 @@ void HipAllreduceDirect::run(
         chunkOffIn(b, c);
   };
 
-  gateEv_->record(callerStream);
-  gateEv_->streamWait(cs_->stream());
-  for (auto* st : fanout_) {
-    gateEv_->streamWait(st->stream());
+  if (gateBySync()) {
+    GA_HIP_CHECK(hipStreamSynchronize(callerStream));
+  } else {
+    gateEv_->record(callerStream);
+    gateEv_->streamWait(cs_->stream());
+    for (auto* st : fanout_) {
+      gateEv_->streamWait(st->stream());
+    }
   }
 
   const uint64_t runSeq = ++runSeq_;
@@ -1612,10 +1633,14 @@ void HipAlltoall::run(
     return;
   }
   mesh_->ensureCapacity(blockBytes * P, 4096);
-  gateEv_->record(callerStream);
-  gateEv_->streamWait(cs_->stream());
-  for (auto& st : fanout_) {
-    gateEv_->streamWait(st->stream());
+  if (gateBySync()) {
+    GA_HIP_CHECK(hipStreamSynchronize(callerStream));
+  } else {
+    gateEv_->record(callerStream);
+    gateEv_->streamWait(cs_->stream());
+    for (auto& st : fanout_) {
+      gateEv_->streamWait(st->stream());
+    }
   }
   char* work = mesh_->work();
   const uint64_t seq = ++seq_;
