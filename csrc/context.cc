@@ -5,6 +5,16 @@
 
 namespace glooamd {
 
+namespace hip {
+// Defined in hip/core.cc (always linked); walks an empty pool when no
+// device work ever ran.
+void releasePooledStreams(const void* key);
+} // namespace hip
+
+Context::~Context() {
+  hip::releasePooledStreams(this);
+}
+
 Context::Context(int rank, int size, int base)
     : rank(rank), size(size), base(base) {
   GA_ENFORCE_GE(rank, 0);

@@ -19,7 +19,7 @@ namespace glooamd {
 class Context {
  public:
   Context(int rank, int size, int base = 2);
-  virtual ~Context() = default;
+  virtual ~Context(); // returns pooled HIP streams to the freelist
 
   const int rank;
   const int size;

@@ -1,5 +1,6 @@
 #include "hip/algorithms.h"
 
+#include <atomic>
 #include <chrono>
 #include <thread>
 
@@ -47,7 +48,23 @@ void watchdogWait(
         }
       }
       mesh.poisonFlags();
-      (void)hipDeviceSynchronize();
+      // Bounded drain: a wedged hardware queue can make
+      // hipDeviceSynchronize hang forever; give it a few seconds on a
+      // helper thread, then fail fast (a leaked sync thread on a
+      // broken device is acceptable in the fail-fast model — the
+      // throw is what keeps the job from hanging).
+      {
+        auto done = std::make_shared<std::atomic<bool>>(false);
+        std::thread([done] {
+          (void)hipDeviceSynchronize();
+          done->store(true);
+        }).detach();
+        const auto dl =
+            std::chrono::steady_clock::now() + std::chrono::seconds(8);
+        while (!done->load() && std::chrono::steady_clock::now() < dl) {
+          std::this_thread::sleep_for(std::chrono::milliseconds(10));
+        }
+      }
       GA_THROW_IO("hip collective timed out in ", what, " flags:", flags);
     }
     std::this_thread::yield();

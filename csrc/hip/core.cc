@@ -1,6 +1,7 @@
 #include "hip/core.h"
 
 #include <map>
+#include <vector>
 #include <memory>
 #include <tuple>
 
@@ -71,23 +72,70 @@ void HipStream::recordAndWait(HipStream& other) {
   GA_HIP_CHECK(hipStreamWaitEvent(other.stream_, event_, 0));
 }
 
+namespace {
+std::mutex& poolMutex() {
+  static std::mutex mu;
+  return mu;
+}
+// Intentionally leaked: destroying streams during static teardown (after
+// the HIP runtime unloads) is unsafe.
+std::map<std::tuple<const void*, int, int>, std::unique_ptr<HipStream>>&
+streamPool() {
+  static auto* pool = new std::map<std::tuple<const void*, int, int>,
+                                   std::unique_ptr<HipStream>>();
+  return *pool;
+}
+// Streams of destroyed contexts, reusable per (device, idx). Bounding
+// the process-lifetime stream count matters: HIP multiplexes streams
+// onto a small number of hardware queues, and once two ACTIVE streams
+// share a queue, a cross-stream event-wait packet at one queue's head
+// can block the very stream that would satisfy it (observed as flaky
+// deadlocks in engine-recreation loops once stream count grew past the
+// queue count). Reuse keeps the total at the live working set.
+std::map<std::pair<int, int>, std::vector<std::unique_ptr<HipStream>>>&
+streamFreelist() {
+  static auto* fl = new std::map<std::pair<int, int>,
+                                 std::vector<std::unique_ptr<HipStream>>>();
+  return *fl;
+}
+} // namespace
+
 HipStream* pooledStream(const void* key, int device, int idx) {
   GA_ENFORCE_GE(idx, 0);
   GA_ENFORCE_LT(idx, kStreamPoolSize, "stream pool exhausted");
-  static std::mutex mu;
-  // Intentionally leaked: destroying streams during static teardown (after
-  // the HIP runtime unloads) is unsafe.
-  static auto* pool = new std::map<std::tuple<const void*, int, int>,
-                                   std::unique_ptr<HipStream>>();
-  std::lock_guard<std::mutex> lock(mu);
+  std::lock_guard<std::mutex> lock(poolMutex());
   auto k = std::make_tuple(key, device, idx);
-  auto it = pool->find(k);
-  if (it == pool->end()) {
-    it = pool->emplace(k, std::make_unique<HipStream>(
-                              device, /*highPriority=*/idx == 0))
-             .first;
+  auto& pool = streamPool();
+  auto it = pool.find(k);
+  if (it == pool.end()) {
+    auto& fl = streamFreelist()[{device, idx}];
+    if (!fl.empty()) {
+      it = pool.emplace(k, std::move(fl.back())).first;
+      fl.pop_back();
+    } else {
+      it = pool.emplace(k, std::make_unique<HipStream>(
+                               device, /*highPriority=*/idx == 0))
+               .first;
+    }
   }
   return it->second.get();
+}
+
+void releasePooledStreams(const void* key) {
+  std::lock_guard<std::mutex> lock(poolMutex());
+  auto& pool = streamPool();
+  for (auto it = pool.begin(); it != pool.end();) {
+    if (std::get<0>(it->first) == key) {
+      // Drained before reuse; errors ignored (the context may be dying
+      // on an error path).
+      (void)hipStreamSynchronize(it->second->stream());
+      streamFreelist()[{std::get<1>(it->first), std::get<2>(it->first)}]
+          .push_back(std::move(it->second));
+      it = pool.erase(it);
+    } else {
+      ++it;
+    }
+  }
 }
 
 HipEvent::HipEvent(int device) {

@@ -77,6 +77,10 @@ class HipStream {
 // the same streams (collectives on a communicator are serialized).
 constexpr int kStreamPoolSize = 8;
 HipStream* pooledStream(const void* key, int device, int idx);
+// Return a destroyed context's pooled streams to the per-(device,idx)
+// freelist (drained first). Called from glooamd::Context's destructor
+// so the process-lifetime stream count stays at the live working set.
+void releasePooledStreams(const void* key);
 
 class HipEvent {
  public:
