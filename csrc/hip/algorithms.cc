@@ -71,6 +71,21 @@ void watchdogWait(
   }
 }
 
+
+// Bounded teardown drain: never hang a destructor on a wedged stream.
+void drainBounded(HipStream* s, int ms = 2000) {
+  if (s == nullptr) {
+    return;
+  }
+  const auto dl = std::chrono::steady_clock::now() +
+      std::chrono::milliseconds(ms);
+  while (hipStreamQuery(s->stream()) == hipErrorNotReady &&
+         std::chrono::steady_clock::now() < dl) {
+    std::this_thread::sleep_for(std::chrono::milliseconds(1));
+  }
+  (void)hipGetLastError();
+}
+
 // Phase tracing for hang localization (GLOO_AMD_PHASE_TRACE=1): the
 // last line printed before a stall names the blocking call.
 bool phaseTrace() {
@@ -122,8 +137,8 @@ void gateStreams(
 
 HipAllreduceRing::~HipAllreduceRing() {
   for (size_t j = 0; j < cs_.size(); j++) {
-    (void)hipStreamSynchronize(cs_[j]->stream());
-    (void)hipStreamSynchronize(ks_[j]->stream());
+    drainBounded(cs_[j]);
+    drainBounded(ks_[j]);
   }
   for (auto& kv : graphs_) {
     if (kv.second.exec != nullptr) {
@@ -951,12 +966,8 @@ void HipAllreduceHalvingDoubling::run(
 // destruct: a failed, poisoned run can leave released-but-still-
 // executing work on the POOLED streams, which outlive the engine.
 HipAllreduceHalvingDoubling::~HipAllreduceHalvingDoubling() {
-  if (cs_ != nullptr) {
-    (void)hipStreamSynchronize(cs_->stream());
-  }
-  if (ks_ != nullptr) {
-    (void)hipStreamSynchronize(ks_->stream());
-  }
+  drainBounded(cs_);
+  drainBounded(ks_);
 }
 
 // ===========================================================================
@@ -1043,11 +1054,9 @@ void HipBroadcastOneToAll::run(
 }
 
 HipBroadcastOneToAll::~HipBroadcastOneToAll() {
-  if (cs_ != nullptr) {
-    (void)hipStreamSynchronize(cs_->stream());
-  }
+  drainBounded(cs_);
   for (auto* st : fanout_) {
-    (void)hipStreamSynchronize(st->stream());
+    drainBounded(st);
   }
 }
 
@@ -1301,11 +1310,9 @@ void HipAllreduceDirect::run(
 }
 
 HipAllreduceDirect::~HipAllreduceDirect() {
-  if (cs_ != nullptr) {
-    (void)hipStreamSynchronize(cs_->stream());
-  }
+  drainBounded(cs_);
   for (auto* st : fanout_) {
-    (void)hipStreamSynchronize(st->stream());
+    drainBounded(st);
   }
 }
 
@@ -1465,12 +1472,8 @@ void HipAllgatherRing::run(
 }
 
 HipAllgatherRing::~HipAllgatherRing() {
-  if (cs_ != nullptr) {
-    (void)hipStreamSynchronize(cs_->stream());
-  }
-  if (ks_ != nullptr) {
-    (void)hipStreamSynchronize(ks_->stream());
-  }
+  drainBounded(cs_);
+  drainBounded(ks_);
 }
 
 // ===========================================================================
@@ -1600,12 +1603,8 @@ void HipReduceScatterRing::run(
 }
 
 HipReduceScatterRing::~HipReduceScatterRing() {
-  if (cs_ != nullptr) {
-    (void)hipStreamSynchronize(cs_->stream());
-  }
-  if (ks_ != nullptr) {
-    (void)hipStreamSynchronize(ks_->stream());
-  }
+  drainBounded(cs_);
+  drainBounded(ks_);
 }
 
 // ===========================================================================
@@ -1720,11 +1719,9 @@ void HipAlltoall::run(
 }
 
 HipAlltoall::~HipAlltoall() {
-  if (cs_ != nullptr) {
-    (void)hipStreamSynchronize(cs_->stream());
-  }
+  drainBounded(cs_);
   for (auto* st : fanout_) {
-    (void)hipStreamSynchronize(st->stream());
+    drainBounded(st);
   }
 }
 

@@ -1,6 +1,8 @@
 #include "hip/core.h"
 
+#include <chrono>
 #include <map>
+#include <thread>
 #include <vector>
 #include <memory>
 #include <tuple>
@@ -55,7 +57,16 @@ HipStream::HipStream(HipStream&& o) noexcept
 
 HipStream::~HipStream() {
   if (stream_ != nullptr) {
-    (void)hipStreamSynchronize(stream_);
+    // Bounded drain: an unbounded sync on a wedged stream would hang
+    // teardown forever; hipStreamDestroy itself defers resource release
+    // until the remaining work (if any) completes.
+    const auto dl =
+        std::chrono::steady_clock::now() + std::chrono::seconds(2);
+    while (hipStreamQuery(stream_) == hipErrorNotReady &&
+           std::chrono::steady_clock::now() < dl) {
+      std::this_thread::sleep_for(std::chrono::milliseconds(1));
+    }
+    (void)hipGetLastError();
     (void)hipStreamDestroy(stream_);
   }
   if (event_ != nullptr) {
@@ -122,15 +133,21 @@ HipStream* pooledStream(const void* key, int device, int idx) {
 }
 
 void releasePooledStreams(const void* key) {
+  // Streams that still have work (a poisoned run's tail, or a wedged
+  // queue) are QUARANTINED, not reused and not destroyed: syncing or
+  // destroying them could hang this (fail-fast) teardown path forever.
+  static auto* graveyard = new std::vector<std::unique_ptr<HipStream>>();
   std::lock_guard<std::mutex> lock(poolMutex());
   auto& pool = streamPool();
   for (auto it = pool.begin(); it != pool.end();) {
     if (std::get<0>(it->first) == key) {
-      // Drained before reuse; errors ignored (the context may be dying
-      // on an error path).
-      (void)hipStreamSynchronize(it->second->stream());
-      streamFreelist()[{std::get<1>(it->first), std::get<2>(it->first)}]
-          .push_back(std::move(it->second));
+      if (hipStreamQuery(it->second->stream()) == hipSuccess) {
+        streamFreelist()[{std::get<1>(it->first), std::get<2>(it->first)}]
+            .push_back(std::move(it->second));
+      } else {
+        (void)hipGetLastError(); // swallow hipErrorNotReady
+        graveyard->push_back(std::move(it->second));
+      }
       it = pool.erase(it);
     } else {
       ++it;
