@@ -439,3 +439,45 @@ def test_unbound_buffer_lifetime(spawn_threads):
         return True
 
     spawn_threads(2, fn)
+
+
+def test_concurrent_fat_sends_all_pairs(spawn_threads):
+    """Exercise the unlocked-writev tx drain (r02): every rank pushes an
+    8MB payload to every peer CONCURRENTLY from separate threads while
+    receiving, so several pairs of one context flush at once. With the
+    old single-lock flush this serialized; either way the data must
+    land intact (per-pair tx-busy drain, deque-stable TxOp fronts)."""
+    import threading
+
+    def fn(ctx, rank, size):
+        n = 2_000_000  # 8 MB fp32
+        sends = []
+        recvs = []
+        for peer in range(size):
+            if peer == rank:
+                continue
+            x = np.full(n, float(rank * 10 + peer), dtype=np.float32)
+            y = np.zeros(n, dtype=np.float32)
+            us = ctx.create_unbound_buffer(x.ctypes.data, x.nbytes)
+            ur = ctx.create_unbound_buffer(y.ctypes.data, y.nbytes)
+            ur.recv(peer, slot=500 + rank)  # slot keyed by SENDER
+            sends.append((us, peer, x))
+            recvs.append((ur, peer, y))
+        # concurrent posts from separate threads -> tx-busy contention
+        ths = []
+        for us, peer, _x in sends:
+            ths.append(threading.Thread(
+                target=lambda us=us, peer=peer: us.send(
+                    peer, 500 + peer)))
+        for t in ths:
+            t.start()
+        for t in ths:
+            t.join()
+        for ur, peer, y in recvs:
+            ur.wait_recv()
+            assert np.all(y == float(peer * 10 + rank)), (rank, peer)
+        for us, _p, _x in sends:
+            us.wait_send()
+        return True
+
+    spawn_threads(3, fn)
