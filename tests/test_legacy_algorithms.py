@@ -24,8 +24,7 @@ ALLREDUCE_ALGOS = [
 @pytest.mark.parametrize("size", [1, 2, 3, 4])
 @pytest.mark.parametrize("elements", [1, 100, 10_000])
 def test_legacy_allreduce(spawn_threads, algo, size, elements):
-    if algo == "allreduce_bcube" and size == 3:
-        pytest.skip("bcube needs size == base^k")
+    # (bcube now factorizes any size; HD uses binary blocks for non-pow2)
 
     def fn(ctx, rank, _):
         x = fixture(rank, elements)
@@ -181,3 +180,25 @@ def test_legacy_allreduce_hd_6(spawn_threads):
         return True
 
     spawn_threads(6, fn)
+
+
+@pytest.mark.parametrize("size", [5, 6, 7, 11])
+@pytest.mark.parametrize("elements", [1, 7, 10_000, 100_003])
+def test_legacy_hd_binary_blocks(spawn_threads, size, elements):
+    """Non-power-of-2 halving-doubling via binary blocks (reference
+    allreduce_halving_doubling.h:38-64 scheme): P decomposes into
+    power-of-2 blocks (11 -> 8+2+1) that cascade their reduce-scattered
+    segments instead of folding whole buffers into partners."""
+
+    def fn(ctx, rank, _):
+        x = fixture(rank, elements)
+        a = ga._C.create_algorithm(
+            "allreduce_halving_doubling", ctx, [x.ctypes.data], elements)
+        for it in range(3):  # repeat runs on one instance
+            np.copyto(x, fixture(rank, elements))
+            a.run()
+            expected = sum(fixture(r, elements) for r in range(size))
+            assert np.allclose(x, expected), (size, elements, it)
+        return True
+
+    spawn_threads(size, fn)
