@@ -181,9 +181,13 @@ def test_init_process_group_ddp():
     """Full stack: init_process_group('glooamd') + DDP training step in
     subprocesses."""
     worker = os.path.join(os.path.dirname(__file__), "pg_worker.py")
-    import random
+    import socket
 
-    port = random.randint(20000, 40000)
+    # reserve a free port for torch's TCPStore (closing is racy but far
+    # better than blind randints on a shared CI box)
+    with socket.socket() as _s:
+        _s.bind(("127.0.0.1", 0))
+        port = _s.getsockname()[1]
     procs = [
         subprocess.Popen(
             [sys.executable, worker, str(r), "2", str(port)],

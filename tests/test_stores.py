@@ -147,11 +147,9 @@ def test_pci_distance_helpers():
 
 def test_tcp_store_wait_timeout():
     """TcpStore.wait honors its deadline for absent keys."""
-    import random
     import time
 
-    port = random.randint(20000, 40000)
-    s = ga.TcpStore("127.0.0.1", port, is_server=True)
+    s, _port = _tcp_store_on_free_port()
     t0 = time.monotonic()
     with pytest.raises(ga.TimeoutError):
         s.wait(["never-set"], timeout_ms=300)
