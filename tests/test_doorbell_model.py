@@ -1,0 +1,279 @@
+"""Host-side model checking of the device ring allreduce's doorbell
+protocol (csrc/hip/algorithms.cc enqueueRing/runDeviceGraph).
+
+The simulator mirrors the enqueue logic op for op: per (rank, ring) a
+copy stream (ks) and a compute stream (cs); ops execute in stream
+order; a wait blocks its stream until the flag value arrives. ALL runs
+are enqueued up front with a per-rank run barrier between them (the
+real host syncs a rank's streams between runs, but OTHER ranks skew
+arbitrarily), then executed under several adversarial stream orders.
+
+Checks:
+  1. progress — every op executes under every探 explored schedule (no
+     deadlock in the dependency graph);
+  2. overwrite safety — a `put` into a peer inbox slot asserts the
+     previous token was consumed (cross-run lastAck gates / graph-mode
+     crossRunOff gates are exactly what make this true under rank skew);
+  3. token integrity — each reduce consumes exactly the expected seq.
+"""
+import math
+
+import pytest
+
+
+def align_up(v, a):
+    return v if a <= 1 else ((v + a - 1) // a) * a
+
+
+def subspan_of_a(off, length, j, parts, a):
+    per = align_up((length + parts - 1) // parts, a)
+    start = min(off + j * per, off + length)
+    end = min(start + per, off + length)
+    return start, end - start
+
+
+def ring_segments(part_len, es, P, R, inbox_cap, chunked=True):
+    A = max(1, 16 // es)
+    sub_cap = inbox_cap // R
+    per_rank = align_up((part_len + P - 1) // P, A)
+    S = max(2 if chunked else 1, (per_rank * es + sub_cap - 1) // sub_cap)
+    while align_up((per_rank + S - 1) // S, A) * es > sub_cap:
+        S += 1
+    return S
+
+
+class Sim:
+    def __init__(self):
+        self.flags = {}
+        self.streams = {}
+        self.inbox = {}  # (owner, ring, par) -> [token, consumed]
+
+    def stream(self, key):
+        return self.streams.setdefault(key, [])
+
+    def step_stream(self, ops):
+        """Execute ops from the front until blocked; return ops run."""
+        ran = 0
+        while ops:
+            op = ops[0]
+            kind = op[0]
+            if kind == "wait":
+                _, owner, fid, val = op
+                if self.flags.get((owner, fid), 0) < val:
+                    break
+            elif kind == "write":
+                _, owner, fid, val = op
+                self.flags[(owner, fid)] = max(
+                    self.flags.get((owner, fid), 0), val)
+            elif kind == "put":
+                _, owner, ring, par, token = op
+                slot = self.inbox.get((owner, ring, par))
+                assert slot is None or slot[1], (
+                    f"OVERWRITE: rank {owner} ring {ring} par {par} "
+                    f"token {slot and slot[0]} not consumed before "
+                    f"token {token} landed")
+                self.inbox[(owner, ring, par)] = [token, False]
+            elif kind == "consume":
+                _, owner, ring, par, expect = op
+                slot = self.inbox.get((owner, ring, par))
+                assert slot is not None and slot[0] == expect, (
+                    f"rank {owner} ring {ring} par {par}: got "
+                    f"{slot and slot[0]}, expected {expect}")
+                slot[1] = True
+            else:
+                raise AssertionError(kind)
+            ops.pop(0)
+            ran += 1
+        return ran
+
+    def run_to_completion(self, order=None):
+        keys = sorted(self.streams.keys())
+        if order is not None:
+            keys = sorted(keys, key=order)
+        progressed = True
+        while progressed:
+            progressed = False
+            for k in keys:
+                if self.step_stream(self.streams[k]) > 0:
+                    progressed = True
+        left = {k: len(v) for k, v in self.streams.items() if v}
+        assert not left, f"deadlock; blocked streams: {left}"
+
+
+def enqueue_run(sim, P, strides, n, es, inbox_cap, seq_base, last_ack,
+                run_idx, rel=False, ring_k=None, max_k=0):
+    """Mirror of enqueueRing for every rank/ring + the per-rank run
+    barrier the blocking run() provides."""
+    R = len(strides)
+    A = max(1, 16 // es)
+    ks_of = {}
+    for r in range(P):
+        # per-rank run barrier: every stream of rank r waits until every
+        # stream of rank r finished the previous run
+        sids = [f"cs{j}" for j in range(R)] + [f"ks{j}" for j in range(R)]
+        if run_idx > 0:
+            for sid in sids:
+                for sid2 in sids:
+                    sim.stream((r, sid)).append(
+                        ("wait", r, ("RB", sid2), run_idx))
+        for j, stride in enumerate(strides):
+            off, part_len = subspan_of_a(0, n, j, R, A)
+            if part_len == 0:
+                ks_of[(r, j)] = 0
+                continue
+            S = ring_segments(part_len, es, P, R, inbox_cap)
+            K = 2 * (P - 1) * S
+            ks_of[(r, j)] = K
+            right = (r + stride) % P
+            left = (r - stride + P) % P
+            cs = sim.stream((r, f"cs{j}"))
+            ks = sim.stream((r, f"ks{j}"))
+
+            for k in range(K):
+                par = k & 1
+                seq = seq_base + k + 1
+                if rel:
+                    offv = (k - 1) if k >= 2 else (
+                        (ring_k[j] - max_k) + k - 1)
+                    gate = seq_base + offv
+                    if gate > 0:
+                        ks.append(("wait", r, ("ACK", j, par), gate))
+                else:
+                    prev = (seq_base + k - 1) if k >= 2 else \
+                        last_ack[(r, j, par)]
+                    if prev > 0:
+                        ks.append(("wait", r, ("ACK", j, par), prev))
+                ks.append(("put", right, j, par, seq))
+                ks.append(("write", right, ("DATA", j, par), seq))
+                cs.append(("wait", r, ("DATA", j, par), seq))
+                cs.append(("consume", r, j, par, seq))
+                cs.append(("write", left, ("ACK", j, par), seq))
+        # publish this rank's run completion per stream
+        for sid in sids:
+            sim.stream((r, sid)).append(
+                ("write", r, ("RB", sid), run_idx + 1))
+    # eager/graph both maintain the same lastAck bookkeeping
+    for r in range(P):
+        for j in range(R):
+            K = ks_of[(r, j)]
+            for k in range(max(0, K - 2), K):
+                last_ack[(r, j, k & 1)] = seq_base + k + 1
+    return ks_of
+
+
+ORDERS = [
+    None,
+    lambda k: (k[0], k[1]),            # rank-major: rank 0 races ahead
+    lambda k: (-k[0], k[1]),           # last rank races ahead
+    lambda k: (k[1], k[0]),            # all ks streams first
+]
+
+
+def run_model(P, R, n_list, es=4, inbox_cap=4 << 20, graph_after=None):
+    strides = [st for st in range(1, max(P, 2))
+               if math.gcd(st, P) == 1][:R]
+    if not strides:
+        strides = [1]
+    for order in ORDERS:
+        sim = Sim()
+        seq_base = 0
+        last_ack = {}
+        R2 = len(strides)
+        for r in range(P):
+            for j in range(R2):
+                for par in (0, 1):
+                    last_ack[(r, j, par)] = 0
+        prev_n = None
+        for i, n in enumerate(n_list):
+            rel = (graph_after is not None and i >= graph_after
+                   and prev_n == n)
+            ring_k, max_k = None, 0
+            if rel:
+                A = max(1, 16 // es)
+                ring_k = []
+                for j in range(R2):
+                    _, plen = subspan_of_a(0, n, j, R2, A)
+                    K = (2 * (P - 1) *
+                         ring_segments(plen, es, P, R2, inbox_cap)
+                         if plen else 0)
+                    ring_k.append(K)
+                max_k = max(ring_k)
+            ks_of = enqueue_run(
+                sim, P, strides, n, es, inbox_cap, seq_base, last_ack,
+                run_idx=i, rel=rel, ring_k=ring_k, max_k=max_k)
+            seq_base += max(ks_of.values()) if ks_of else 0
+            prev_n = n
+        sim.run_to_completion(order)
+
+
+@pytest.mark.parametrize("P,R", [(2, 1), (3, 2), (4, 2), (8, 4)])
+def test_ring_protocol_eager_multirun(P, R):
+    """Four eager runs enqueued with arbitrary cross-rank skew: the
+    cross-run inbox gates must prevent overwrites and deadlocks."""
+    run_model(P, R, [1_000_000] * 4)
+
+
+@pytest.mark.parametrize("P,R", [(2, 1), (4, 2), (8, 4)])
+def test_ring_protocol_graph_relative(P, R):
+    """Run 1 eager, runs 2-5 graph-relative gating (incl. the per-ring
+    crossRunOff correction when a ring has fewer steps than the max)."""
+    run_model(P, R, [999_999] * 5, graph_after=1)
+
+
+@pytest.mark.parametrize("P,R", [(4, 2), (8, 4)])
+def test_ring_protocol_mixed_shapes(P, R):
+    """Shape switches force eager runs between graph replays; eager and
+    graph bookkeeping must interleave exactly."""
+    run_model(P, R, [1_000_000, 700_001, 700_001, 1_000_000, 1_000_000,
+                     700_001, 700_001], graph_after=1)
+
+
+def test_ring_protocol_small_inbox_deep_pipeline():
+    """64 KiB inbox -> dozens of segments; deep double-buffer chains."""
+    run_model(4, 2, [500_000] * 3, inbox_cap=64 * 1024)
+
+
+def _run_with_cross_run_bias(bias):
+    """3 runs, graph-relative from run 2, with the k<2 cross-run gate
+    offset biased by `bias` steps (0 = correct protocol)."""
+    P, es, cap = 4, 4, 4 << 20
+    strides = [1, 3]
+    n = 1_000_000
+    R = len(strides)
+    A = max(1, 16 // es)
+    ring_k = []
+    for j in range(R):
+        _, plen = subspan_of_a(0, n, j, R, A)
+        ring_k.append(2 * (P - 1) *
+                      ring_segments(plen, es, P, R, cap) if plen else 0)
+    max_k = max(ring_k)
+    for order in ORDERS:
+        sim = Sim()
+        last_ack = {(r, j, p): 0 for r in range(P) for j in range(R)
+                    for p in (0, 1)}
+        seq = 0
+        for i in range(3):
+            ks_of = enqueue_run(
+                sim, P, strides, n, es, cap, seq, last_ack, run_idx=i,
+                rel=(i >= 1),
+                ring_k=[k + bias for k in ring_k],  # biases the gate
+                max_k=max_k)
+            seq += max(ks_of.values())
+        sim.run_to_completion(order)
+
+
+def test_ring_protocol_detects_too_early_gate():
+    """Negative control: a cross-run gate 2 steps too permissive (the
+    class of bug the per-ring crossRunOff correction prevents when ring
+    step counts differ) must trip the overwrite assertion under at
+    least one rank-skew schedule."""
+    _run_with_cross_run_bias(0)  # correct protocol passes
+    with pytest.raises(AssertionError):
+        _run_with_cross_run_bias(-2)
+
+
+def test_ring_protocol_detects_too_strict_gate():
+    """A gate 2 steps too strict waits for acks that never come:
+    detected as a deadlock."""
+    with pytest.raises(AssertionError, match="deadlock"):
+        _run_with_cross_run_bias(+2)
