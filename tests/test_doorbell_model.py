@@ -277,3 +277,221 @@ def test_ring_protocol_detects_too_strict_gate():
     detected as a deadlock."""
     with pytest.raises(AssertionError, match="deadlock"):
         _run_with_cross_run_bias(+2)
+
+
+# ---------------------------------------------------------------------------
+# Halving-doubling model (csrc/hip/algorithms.cc HipAllreduceHalvingDoubling)
+# incl. the non-pow2 fold lanes and the allgather mirror's direct writes
+# into peer work regions.
+# ---------------------------------------------------------------------------
+
+
+class HdSim(Sim):
+    def __init__(self):
+        super().__init__()
+        self.work = {}  # (owner, t) -> [token, consumed] mirror-landing
+
+    def step_stream(self, ops):
+        ran = 0
+        while ops:
+            op = ops[0]
+            kind = op[0]
+            if kind == "wait":
+                _, owner, fid, val = op
+                if self.flags.get((owner, fid), 0) < val:
+                    break
+            elif kind == "write":
+                _, owner, fid, val = op
+                self.flags[(owner, fid)] = max(
+                    self.flags.get((owner, fid), 0), val)
+            elif kind == "put":
+                _, owner, ring, par, token = op
+                slot = self.inbox.get((owner, ring, par))
+                assert slot is None or slot[1], (
+                    f"OVERWRITE: rank {owner} slot ({ring},{par}) token "
+                    f"{slot and slot[0]} not consumed before {token}")
+                self.inbox[(owner, ring, par)] = [token, False]
+            elif kind == "consume":
+                _, owner, ring, par, expect = op
+                slot = self.inbox.get((owner, ring, par))
+                assert slot is not None and slot[0] == expect, (
+                    f"rank {owner} slot ({ring},{par}): got "
+                    f"{slot and slot[0]}, expected {expect}")
+                slot[1] = True
+            elif kind == "put_work":
+                _, owner, t, token = op
+                slot = self.work.get((owner, t))
+                assert slot is None or slot[1], (
+                    f"WORK OVERWRITE at rank {owner} step {t}: token "
+                    f"{slot and slot[0]} not staged out before {token}")
+                self.work[(owner, t)] = [token, False]
+            elif kind == "stage_out":
+                _, owner, expects = op
+                for t, token in expects:
+                    slot = self.work.get((owner, t))
+                    assert slot is not None and slot[0] == token, (
+                        f"rank {owner} stage-out step {t}: got "
+                        f"{slot and slot[0]}, expected {token}")
+                    slot[1] = True
+            else:
+                raise AssertionError(kind)
+            ops.pop(0)
+            ran += 1
+        return ran
+
+
+def hd_chunks(max_half, es, sub_bytes):
+    A = max(1, 16 // es)
+    nc = max(1, (max_half * es + sub_bytes - 1) // sub_bytes)
+    while align_up((max_half + nc - 1) // nc, A) * es > sub_bytes:
+        nc += 1
+    return nc
+
+
+def enqueue_hd_run(sim, P, n, es, inbox_cap, state, run_idx,
+                   drop_fack_gate=False):
+    """Mirror of HipAllreduceHalvingDoubling::run for every rank:
+    pow2 exchange + fold lanes for extras + mirror + stage-out."""
+    pow2 = 1
+    T = 0
+    while pow2 * 2 <= P:
+        pow2 *= 2
+        T += 1
+    T = max(1, T)
+    extras = P - pow2
+    rows = T + (1 if extras else 0)
+    sub_bytes = ((2 * inbox_cap) // (2 * rows)) // 16 * 16
+    A = max(1, 16 // es)
+    fold_elems = sub_bytes // es
+    fold_nc = (n + fold_elems - 1) // fold_elems
+
+    for r in range(P):
+        cs = sim.stream((r, "cs"))
+        ks = sim.stream((r, "ks"))
+        if run_idx > 0:  # per-rank run barrier (blocking run())
+            for sid in ("cs", "ks"):
+                for sid2 in ("cs", "ks"):
+                    sim.stream((r, sid)).append(
+                        ("wait", r, ("RB", sid2), run_idx))
+        if r >= pow2:  # extra: fold lanes + post-fold
+            partner = r - pow2
+            for c in range(fold_nc):
+                par = c & 1
+                state["foldSeq", r] = state.get(("foldSeq", r), 0) + 1
+                seq = state["foldSeq", r]
+                last = state.get(("lastFold", r, par), 0)
+                if last > 0 and not drop_fack_gate:
+                    ks.append(("wait", r, ("FACK", par), last))
+                ks.append(("put", partner, "FOLD", par, seq))
+                ks.append(("write", partner, ("FOLD", par), seq))
+                state["lastFold", r, par] = seq
+            state["postSeq", r] = state.get(("postSeq", r), 0) + 1
+            ps = state["postSeq", r]
+            cs.append(("wait", r, ("POST",), ps))
+            cs.append(("stage_out", r, [("post", ps)]))
+            cs.append(("write", partner, ("PACK",), ps))
+        else:
+            span_len = n
+            if r < extras:  # partner: fold reduces
+                extra = r + pow2
+                for c in range(fold_nc):
+                    par = c & 1
+                    state["foldSeq", r] = state.get(("foldSeq", r), 0) + 1
+                    seq = state["foldSeq", r]
+                    cs.append(("wait", r, ("FOLD", par), seq))
+                    cs.append(("consume", r, "FOLD", par, seq))
+                    cs.append(("write", extra, ("FACK", par), seq))
+            # reduce-scatter (cs records a per-step event the mirror
+            # waits on — the real evs[t] / evs[T-1]->streamWait(ksm))
+            for t in range(T):
+                peer_global = (r ^ (1 << t))
+                half = align_up((span_len + 1) // 2, A)
+                nc = hd_chunks(half, es, sub_bytes)
+                for c in range(nc):
+                    par = c & 1
+                    state["seq", r] = state.get(("seq", r), 0) + 1
+                    seq = state["seq", r]
+                    last = state.get(("lastAck", r, t, par), 0)
+                    if last > 0:
+                        ks.append(("wait", r, ("ACK", t, par), last))
+                    ks.append(("put", peer_global, t, par, seq))
+                    ks.append(("write", peer_global, ("DATA", t, par), seq))
+                    cs.append(("wait", r, ("DATA", t, par), seq))
+                    cs.append(("consume", r, t, par, seq))
+                    cs.append(("write", peer_global, ("ACK", t, par), seq))
+                    state["lastAck", r, t, par] = seq
+                cs.append(("write", r, ("EV", t), run_idx + 1))
+                span_len = half
+            # allgather mirror: one direct write into peer work per step
+            for t in range(T - 1, -1, -1):
+                peer_global = (r ^ (1 << t))
+                state["seq", r] = state.get(("seq", r), 0) + 1
+                aseq = state["seq", r]
+                if t == T - 1:
+                    ks.append(("wait", r, ("EV", T - 1), run_idx + 1))
+                else:
+                    la = state.get(("lastAgd", r, t + 1), 0)
+                    if la > 0:
+                        ks.append(("wait", r, ("AGD", t + 1), la))
+                ks.append(("put_work", peer_global, t, aseq))
+                ks.append(("write", peer_global, ("AGD", t), aseq))
+                state["lastAgd", r, t] = aseq
+            # final: wait all AGD, stage out, post-fold push
+            expects = []
+            for t in range(T):
+                la = state.get(("lastAgd", r, t), 0)
+                cs.append(("wait", r, ("AGD", t), la))
+                expects.append((t, la))
+            cs.append(("stage_out", r, expects))
+            if r < extras:
+                extra = r + pow2
+                state["postSeq", r] = state.get(("postSeq", r), 0) + 1
+                ps = state["postSeq", r]
+                if ps > 1:
+                    cs.append(("wait", r, ("PACK",), ps - 1))
+                cs.append(("put_work", extra, "post", ps))
+                cs.append(("write", extra, ("POST",), ps))
+        for sid in ("cs", "ks"):
+            sim.stream((r, sid)).append(("write", r, ("RB", sid),
+                                         run_idx + 1))
+
+
+def run_hd_model(P, n_list, es=4, inbox_cap=4 << 20):
+    for order in ORDERS:
+        sim = HdSim()
+        state = {}
+        for i, n in enumerate(n_list):
+            enqueue_hd_run(sim, P, n, es, inbox_cap, state, i)
+        sim.run_to_completion(order)
+
+
+@pytest.mark.parametrize("P", [2, 3, 4, 5, 6, 7, 8])
+def test_hd_protocol_multirun(P):
+    """Three HD runs under adversarial rank skew, incl. non-pow2 fold
+    lanes, mirror work-region writes and post-fold handshakes."""
+    run_hd_model(P, [1_500_000] * 3)
+
+
+@pytest.mark.parametrize("P", [3, 6])
+def test_hd_protocol_small_inbox(P):
+    """Chunked fold + sub-chunked halves (deep fold/step pipelines)."""
+    run_hd_model(P, [800_000] * 3, inbox_cap=256 * 1024)
+
+
+def test_hd_protocol_detects_missing_fold_ack_gate():
+    """Negative control: without the fFACK slot-reuse gate the extra
+    rank's chunk c overwrites chunk c-2's fold slot before the partner
+    consumed it (needs a chunked fold: small inbox)."""
+    P, es, cap = 3, 4, 256 * 1024
+    failed = False
+    for order in ORDERS:
+        sim = HdSim()
+        state = {}
+        for i in range(2):
+            enqueue_hd_run(sim, P, 800_000, es, cap, state, i,
+                           drop_fack_gate=True)
+        try:
+            sim.run_to_completion(order)
+        except AssertionError:
+            failed = True
+    assert failed, "checker failed to catch the dropped fFACK gate"
