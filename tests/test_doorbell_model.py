@@ -495,3 +495,121 @@ def test_hd_protocol_detects_missing_fold_ack_gate():
         except AssertionError:
             failed = True
     assert failed, "checker failed to catch the dropped fFACK gate"
+
+
+# ---------------------------------------------------------------------------
+# Direct (one-shot) allreduce model (HipAllreduceDirect): scatter /
+# fused-reduce / broadcast chunk pipeline with fRS/fACK/fAG/fDONE and
+# the cross-run fDONE gate protecting peer work regions.
+# ---------------------------------------------------------------------------
+
+
+def enqueue_direct_run(sim, P, C, state, run_idx, nf=7,
+                       drop_ack_gate=False, drop_ce_gate=False):
+    run_seq = run_idx + 1
+    base = state.get("chunkSeqBase", 0)
+
+    def cseq(c):
+        return base + c + 1
+
+    for r in range(P):
+        sids = ["cs"] + [f"f{i}" for i in range(min(nf, P - 1))]
+        if run_idx > 0:
+            for sid in sids:
+                for sid2 in sids:
+                    sim.stream((r, sid)).append(
+                        ("wait", r, ("RB", sid2), run_idx))
+        cs = sim.stream((r, "cs"))
+        fan = [sim.stream((r, f"f{i}"))
+               for i in range(min(nf, P - 1))]
+        # peers must have copied out the previous run before we write
+        # their work regions again
+        for j in range(1, P):
+            d = (r + j) % P
+            st = fan[(j - 1) % len(fan)]
+            if run_seq > 1:
+                st.append(("wait", r, ("DONE", d), run_seq - 1))
+        for c in range(C):
+            par = c & 1
+            for j in range(1, P):
+                d = (r + j) % P
+                st = fan[(j - 1) % len(fan)]
+                gate = (cseq(c - 2) if c >= 2
+                        else state.get(("lastAck", r, d, par), 0))
+                if gate > 0 and not drop_ack_gate:
+                    st.append(("wait", r, ("ACK", d), gate))
+                st.append(("put", d, ("in", r), par, cseq(c)))
+                st.append(("write", d, ("RS", r), cseq(c)))
+            for src in range(P):
+                if src != r:
+                    cs.append(("wait", r, ("RS", src), cseq(c)))
+            for src in range(P):
+                if src != r:
+                    cs.append(("consume", r, ("in", src), par, cseq(c)))
+            for src in range(P):
+                if src != r:
+                    cs.append(("write", src, ("ACK", r), cseq(c)))
+            cs.append(("write", r, ("CE", c % 4), cseq(c)))
+            for j in range(1, P):
+                d = (r + j) % P
+                st = fan[(j - 1) % len(fan)]
+                if not drop_ce_gate:
+                    st.append(("wait", r, ("CE", c % 4), cseq(c)))
+                st.append(("put_work", d, (r, c), cseq(c)))
+                st.append(("write", d, ("AG", r), cseq(c)))
+        expects = []
+        for src in range(P):
+            if src != r:
+                cs.append(("wait", r, ("AG", src), cseq(C - 1)))
+                for c in range(C):
+                    expects.append(((src, c), cseq(c)))
+        cs.append(("stage_out", r, expects))
+        for src in range(P):
+            if src != r:
+                cs.append(("write", src, ("DONE", r), run_seq))
+        for sid in sids:
+            sim.stream((r, sid)).append(("write", r, ("RB", sid),
+                                         run_idx + 1))
+    for r in range(P):
+        for d in range(P):
+            for c in range(max(0, C - 2), C):
+                state["lastAck", r, d, c & 1] = cseq(c)
+    state["chunkSeqBase"] = base + C
+
+
+def run_direct_model(P, C, runs=3, nf=7, drop_ack_gate=False,
+                     drop_ce_gate=False):
+    for order in ORDERS:
+        sim = HdSim()
+        state = {}
+        for i in range(runs):
+            enqueue_direct_run(sim, P, C, state, i, nf=nf,
+                               drop_ack_gate=drop_ack_gate,
+                               drop_ce_gate=drop_ce_gate)
+        sim.run_to_completion(order)
+
+
+@pytest.mark.parametrize("P,C", [(2, 1), (2, 3), (4, 2), (8, 4), (8, 1)])
+def test_direct_protocol_multirun(P, C):
+    """Chunk-pipelined direct allreduce, 3 runs under rank skew:
+    double-buffered inbox slots, chunkEvent broadcast gating, and the
+    cross-run fDONE work-region gate."""
+    run_direct_model(P, C)
+
+
+def test_direct_protocol_fewer_fanout_streams():
+    run_direct_model(8, 3, nf=3)
+
+
+def test_direct_protocol_defense_in_depth_and_teeth():
+    """Model FINDING: with the per-chunk interleaved enqueue (scatter,
+    reduce-gated broadcast per fanout stream), the pipeline is
+    self-throttling — dropping the fACK slot-reuse gate alone cannot
+    produce an overwrite (the gate is defense in depth, load-bearing
+    only if the enqueue shape changes). Dropping BOTH the chunkEvent
+    broadcast gate and the ack gate breaks the throttle and the checker
+    catches the overwrite — proving the model has teeth."""
+    run_direct_model(4, 4, runs=2, drop_ack_gate=True)  # belt holds
+    with pytest.raises(AssertionError):
+        run_direct_model(4, 4, runs=2, drop_ack_gate=True,
+                         drop_ce_gate=True)
