@@ -613,3 +613,64 @@ def test_direct_protocol_defense_in_depth_and_teeth():
     with pytest.raises(AssertionError):
         run_direct_model(4, 4, runs=2, drop_ack_gate=True,
                          drop_ce_gate=True)
+
+
+# ---------------------------------------------------------------------------
+# HipP2P lane model: per-(src->dst) double-buffered chunk lanes with
+# post/flush split (posts enqueue, flush is the only sync point).
+# ---------------------------------------------------------------------------
+
+
+def enqueue_p2p_message(sim, state, src, dst, nchunks):
+    ss = sim.stream((src, "ss"))
+    rs = sim.stream((dst, "rs"))
+    for _ in range(nchunks):
+        state["s", src, dst] = state.get(("s", src, dst), 0) + 1
+        seq = state["s", src, dst]
+        par = seq & 1
+        if seq > 2:
+            ss.append(("wait", src, ("ACK", dst), seq - 2))
+        ss.append(("put", dst, ("lane", src), par, seq))
+        ss.append(("write", dst, ("DATA", src), seq))
+    for _ in range(nchunks):
+        state["r", dst, src] = state.get(("r", dst, src), 0) + 1
+        seq = state["r", dst, src]
+        par = seq & 1
+        rs.append(("wait", dst, ("DATA", src), seq))
+        rs.append(("consume", dst, ("lane", src), par, seq))
+        rs.append(("write", src, ("ACK", dst), seq))
+
+
+@pytest.mark.parametrize("nchunks", [1, 2, 5, 9])
+def test_p2p_lane_protocol(nchunks):
+    """Bidirectional multi-message exchange + a gather-shaped fan-in:
+    lane seqs are per-direction, so concurrent traffic cannot cross."""
+    for order in ORDERS:
+        sim = Sim()
+        state = {}
+        # bidirectional pair traffic, several messages
+        for _ in range(3):
+            enqueue_p2p_message(sim, state, 0, 1, nchunks)
+            enqueue_p2p_message(sim, state, 1, 0, nchunks)
+        # gather fan-in to rank 0 from 3 senders
+        for src in (1, 2, 3):
+            enqueue_p2p_message(sim, state, src, 0, nchunks)
+        sim.run_to_completion(order)
+
+
+def test_p2p_lane_detects_missing_ack():
+    """Negative control: without the seq-2 ack gate a >2-chunk message
+    overwrites the un-consumed lane slot."""
+    sim = Sim()
+    state = {}
+    ss = sim.stream((0, "ss"))
+    rs = sim.stream((1, "rs"))
+    for seq in (1, 2, 3):
+        ss.append(("put", 1, ("lane", 0), seq & 1, seq))
+        ss.append(("write", 1, ("DATA", 0), seq))
+    for seq in (1, 2, 3):
+        rs.append(("wait", 1, ("DATA", 0), seq))
+        rs.append(("consume", 1, ("lane", 0), seq & 1, seq))
+        rs.append(("write", 0, ("ACK", 1), seq))
+    with pytest.raises(AssertionError):
+        sim.run_to_completion(lambda k: 0 if k[1] == "ss" else 1)
