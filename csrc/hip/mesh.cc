@@ -29,35 +29,25 @@ void* allocFineGrained(size_t bytes) {
   return p;
 }
 
-// Flag pages are IMMORTAL: freed-and-recycled flag addresses are the
-// one thing a stale doorbell packet (CP wait-value or spin kernel left
-// on a pooled stream by a failed, poisoned run) can still reference.
-// Reusing a page keeps any such stale reference pointing at poisoned
-// values (harmless) instead of freed memory. A page is 4 KiB; a
-// process churning meshes leaks a handful at most.
+// Flag pages are IMMORTAL AND NEVER REUSED: a stale doorbell packet
+// (CP wait-value, spin kernel, or an unexecuted flag write left on a
+// quarantined stream of a failed, poisoned run — possibly in a PEER
+// process via the IPC mapping) may still reference the page long after
+// its mesh died. Freeing would make that packet touch freed memory;
+// recycling would let a late stale WRITE corrupt an unrelated new
+// mesh's flags. Leaking is strictly safe: a page is 4 KiB per mesh
+// lifetime (a full test-suite run leaks a couple of MB).
 std::mutex& flagPoolMutex() {
   static std::mutex mu;
   return mu;
 }
-std::vector<uint64_t*>& flagPool() {
-  static auto* pool = new std::vector<uint64_t*>();
-  return *pool;
-}
 uint64_t* acquireFlagPage(size_t count) {
-  {
-    std::lock_guard<std::mutex> lock(flagPoolMutex());
-    auto& pool = flagPool();
-    if (!pool.empty()) {
-      uint64_t* p = pool.back();
-      pool.pop_back();
-      return p;
-    }
-  }
   return static_cast<uint64_t*>(allocFineGrained(count * 8));
 }
 void releaseFlagPage(uint64_t* p) {
+  static auto* graveyard = new std::vector<uint64_t*>();
   std::lock_guard<std::mutex> lock(flagPoolMutex());
-  flagPool().push_back(p);
+  graveyard->push_back(p);
 }
 
 struct HandleBlob {
