@@ -674,3 +674,18 @@ def test_p2p_lane_detects_missing_ack():
         rs.append(("write", 0, ("ACK", 1), seq))
     with pytest.raises(AssertionError):
         sim.run_to_completion(lambda k: 0 if k[1] == "ss" else 1)
+
+
+def test_model_math_matches_cpp():
+    """The model's split-math mirror must equal the C++ schedule.h
+    aligned variants the engines actually use (keeps the model honest
+    if the C++ math ever changes)."""
+    import gloo_amd as ga
+
+    for n in [1, 7, 999_999, 1_000_000, 5_000_000, 100_000_000]:
+        for parts in [1, 2, 3, 4, 8]:
+            for A in [1, 2, 4, 8]:
+                for j in range(parts):
+                    assert subspan_of_a(0, n, j, parts, A) == \
+                        ga._C.subspan_of_a(0, n, j, parts, A), \
+                        (n, parts, A, j)
